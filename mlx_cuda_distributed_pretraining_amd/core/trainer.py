@@ -1,0 +1,496 @@
+"""Trainer runtime.
+
+Orchestration parity with the reference Trainer
+(/root/reference/core/training.py:898-1904): seeding, device setup, model
+build from config, loss with pad masking + truncation warning, validation
+capped at 50 batches, checkpoint save/load/resume, the training loop with
+gradient accumulation/clipping, early stopping, LR finder, per-step log
+lines, sample generation during training, final metadata.
+
+MI355X redesign:
+  - one process per GPU; RCCL bucketed all-reduce overlapped with backward
+    (parallel/ddp.py) instead of the reference's thread-queue mock,
+  - bf16 params + fused flat AdamW (fp32 master) as the default GPU path,
+  - loss/token-count scalar all-reduce to rank 0 for logging (C2).
+"""
+from __future__ import annotations
+
+import csv
+import math
+import os
+import random
+import time
+from pathlib import Path
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..data import DataManager, TokenizerManager
+from ..models.llama import Model, ModelArgs
+from ..ops.cross_entropy import fused_cross_entropy
+from ..optim.enhanced import clip_by_global_norm, global_grad_norm
+from ..optim.flat_fused import FusedFlatAdamW
+from ..optim.manager import OptimizationManager
+from ..parallel.ddp import DataParallelGrads
+from ..parallel.dist import (
+    all_reduce_scalar,
+    barrier,
+    broadcast_module,
+    get_rank,
+    get_world_size,
+    init_distributed,
+    is_distributed,
+)
+from ..parallel.flat import FlatParamSpace
+from .checkpoint import (
+    CheckpointManager,
+    load_checkpoint,
+    rotate_snapshots,
+    save_checkpoint,
+    update_metadata,
+)
+from .config import Config
+from .logger import Logger, format_metrics
+
+
+class EarlyStoppingMonitor:
+    """Parity: /root/reference/core/training.py:621-668."""
+
+    def __init__(self, patience: int = 3, min_delta: float = 0.001, mode: str = "min"):
+        self.patience = patience
+        self.min_delta = min_delta
+        self.mode = mode
+        self.best: Optional[float] = None
+        self.counter = 0
+
+    def update(self, value: float) -> bool:
+        """Returns True when training should stop."""
+        improved = (
+            self.best is None
+            or (self.mode == "min" and value < self.best - self.min_delta)
+            or (self.mode == "max" and value > self.best + self.min_delta)
+        )
+        if improved:
+            self.best = value
+            self.counter = 0
+        else:
+            self.counter += 1
+        return self.counter >= self.patience
+
+
+class Trainer:
+    def __init__(self, config: str | Config, for_training: bool = True, runs_root: str = "runs"):
+        self.config = config if isinstance(config, Config) else Config.from_yaml(config)
+        self.for_training = for_training
+        self.runs_root = runs_root
+        self.rank, self.world_size, self.local_rank = init_distributed(
+            backend=self.config.system.distributed_backend
+            if torch.cuda.is_available()
+            else "gloo"
+        )
+        self.is_main = self.rank == 0
+
+        self.setup_system()
+
+        cfg = self.config
+        if for_training and self.is_main:
+            if not cfg.overwrite and cfg.resume is None:
+                CheckpointManager.validate_unique_name(cfg.name, self.runs_root)
+        barrier()
+        self.run_dir, self.log_path, self.checkpoint_dir = CheckpointManager.setup_run_directory(
+            cfg.name, self.runs_root
+        )
+        self.logger = Logger(
+            self.run_dir, self.log_path, cfg.logging, is_main=self.is_main,
+            use_tensorboard=cfg.logging.tensorboard,
+        )
+        if self.is_main and for_training:
+            cfg.save_yaml(self.run_dir / "config.yaml")
+
+        self.tokenizer = TokenizerManager(cfg.data)
+        if self.is_main and for_training:
+            self.tokenizer.save(self.run_dir / "tokenizer")
+
+        self.setup_model()
+        self.data_manager: Optional[DataManager] = None
+        self.total_steps = 0
+        self.start_step = 0
+        self.total_tokens = 0
+        self.validation_losses: List[tuple] = []
+        if for_training:
+            self.setup_training()
+
+    # ------------------------------------------------------------------
+    def setup_system(self) -> None:
+        cfg = self.config.system
+        seed = cfg.seed + self.rank
+        random.seed(seed)
+        np.random.seed(seed)
+        torch.manual_seed(seed)
+        if torch.cuda.is_available() and cfg.device != "cpu":
+            torch.cuda.set_device(self.local_rank)
+            torch.cuda.manual_seed_all(seed)
+            self.device = torch.device("cuda", self.local_rank)
+        else:
+            self.device = torch.device("cpu")
+        if cfg.precision == "bfloat16":
+            self.param_dtype = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        elif cfg.precision == "float16":
+            self.param_dtype = torch.float16 if self.device.type == "cuda" else torch.float32
+        else:
+            self.param_dtype = torch.float32
+
+    def setup_model(self) -> None:
+        args = ModelArgs.from_config(self.config.model, vocab_size=self.tokenizer.vocab_size)
+        if self.config.model.architecture not in ("llama", "llama_standard"):
+            raise ValueError(
+                f"Unknown architecture {self.config.model.architecture!r} (supported: llama, llama_standard)"
+            )
+        if self.config.model.architecture == "llama_standard":
+            args.attention_type = "simple"
+        self.model_args = args
+        self.model = Model(args).to(device=self.device, dtype=self.param_dtype)
+        if self.config.system.gradient_checkpointing:
+            ratio = self.config.system.gradient_checkpointing_ratio
+            n = int(len(self.model.layers) * ratio)
+            for layer in self.model.layers[:n]:
+                layer.enable_checkpointing()
+        if self.config.data.weight_path:
+            self.model.load_weights(self.config.data.weight_path)
+        broadcast_module(self.model)
+        self.logger.log_model_summary(self.model)
+
+    def setup_training(self) -> None:
+        cfg = self.config
+        hp = cfg.training.hyperparameters
+        self.batch_size = int(hp.get("batch_size", 16))
+        self.grad_accum_steps = int(hp.get("gradient_accumulation_steps", 1))
+        self.max_grad_norm = float(hp.get("gradient_clip", hp.get("max_grad_norm", 0.0)) or 0.0)
+
+        self.data_manager = DataManager(
+            cfg.data, self.tokenizer, self.batch_size,
+            rank=self.rank, world_size=self.world_size, seed=cfg.system.seed,
+        )
+        if cfg.training.epochs is not None and self.data_manager.num_batches:
+            self.steps_per_epoch = max(
+                self.data_manager.num_batches // self.world_size, 1
+            )
+            self.total_steps = self.steps_per_epoch * cfg.training.epochs
+        else:
+            self.steps_per_epoch = None
+            self.total_steps = int(hp.get("iters", 1000))
+
+        self.opt_manager = OptimizationManager(cfg.training, self.total_steps)
+        self.lr_schedule = self.opt_manager.create_scheduler()
+
+        opt_name = str((cfg.training.optimization or {}).get("optimizer", "adamw")).lower()
+        self.use_fused = (
+            self.device.type == "cuda"
+            and opt_name in ("adamw", "adam")
+            and not cfg.system.model_parallel
+        )
+        self.flat_space: Optional[FlatParamSpace] = None
+        self.ddp: Optional[DataParallelGrads] = None
+        if self.use_fused:
+            self.flat_space = FlatParamSpace(self.model)
+            zero1 = cfg.system.zero_optimization_level >= 1
+            if not zero1:
+                self.ddp = DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb)
+            self.optimizer = FusedFlatAdamW(
+                self.flat_space,
+                lr=self.opt_manager.learning_rate,
+                betas=tuple((cfg.training.optimization or {}).get("betas", (0.9, 0.999))),
+                eps=float((cfg.training.optimization or {}).get("eps", 1e-8)),
+                weight_decay=self.opt_manager.weight_decay,
+                max_grad_norm=self.max_grad_norm,
+                zero1=zero1,
+            )
+        else:
+            self.flat_space = FlatParamSpace(self.model)
+            self.ddp = DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb)
+            self.optimizer = self.opt_manager.create_optimizer(self.model)
+
+        es = cfg.training.early_stopping or {}
+        self.early_stopping = (
+            EarlyStoppingMonitor(
+                patience=int(es.get("patience", 3)),
+                min_delta=float(es.get("min_delta", 0.001)),
+                mode=str(es.get("mode", "min")),
+            )
+            if es.get("enabled")
+            else None
+        )
+
+    # ------------------------------------------------------------------
+    def compute_loss(self, inputs: torch.Tensor, targets: torch.Tensor):
+        mpe = self.model_args.max_position_embeddings
+        if mpe and inputs.shape[1] > mpe:
+            if self.is_main:
+                self.logger.log(
+                    f"Warning: sequence length {inputs.shape[1]} > max_position_embeddings {mpe}; truncating",
+                    console=False,
+                )
+            inputs = inputs[:, :mpe]
+            targets = targets[:, :mpe]
+        logits = self.model(inputs)
+        loss, ntok = fused_cross_entropy(
+            logits.reshape(-1, logits.shape[-1]),
+            targets.reshape(-1),
+            ignore_index=self.tokenizer.PAD_TOKEN,
+        )
+        return loss, ntok
+
+    def train_step(self, step: int) -> tuple:
+        """One optimizer step (with gradient accumulation). Returns
+        (loss_detached, ntokens)."""
+        lr = self.lr_schedule(step)
+        if self.ddp is not None:
+            self.ddp.require_reduce = False
+        total_loss = None
+        total_tok = None
+        if self.flat_space is not None:
+            self.flat_space.zero_grad()
+        else:
+            self.model.zero_grad(set_to_none=False)
+        for micro in range(self.grad_accum_steps):
+            if self.ddp is not None and micro == self.grad_accum_steps - 1:
+                self.ddp.require_reduce = True
+            batch = self.data_manager.generate_batch(
+                step * self.grad_accum_steps + micro
+            ).to(self.device, non_blocking=True)
+            inputs, targets = batch[:, :-1], batch[:, 1:]
+            loss, ntok = self.compute_loss(inputs, targets)
+            (loss / self.grad_accum_steps).backward()
+            ld = loss.detach()
+            total_loss = ld if total_loss is None else total_loss + ld
+            total_tok = ntok if total_tok is None else total_tok + ntok
+        if self.ddp is not None:
+            self.ddp.finalize()
+
+        if isinstance(self.optimizer, FusedFlatAdamW):
+            self.optimizer.step(lr=lr)
+        else:
+            if self.max_grad_norm > 0 and not getattr(self.optimizer, "max_grad_norm", 0):
+                clip_by_global_norm(
+                    [p for p in self.model.parameters() if p.requires_grad],
+                    self.max_grad_norm,
+                )
+            for group in self.optimizer.param_groups:
+                group["lr"] = lr
+            sub = [getattr(self.optimizer, "matrix_opt", None), getattr(self.optimizer, "non_matrix_opt", None)]
+            for o in sub:
+                if o is not None:
+                    for group in o.param_groups:
+                        group["lr"] = lr
+            self.optimizer.step()
+        return total_loss / self.grad_accum_steps, total_tok
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def validate(self, max_batches: int = 50) -> Optional[float]:
+        if self.data_manager is None:
+            return None
+        n = min(self.data_manager.num_validation_batches, max_batches)
+        if n == 0:
+            return None
+        self.model.eval()
+        total = torch.zeros((), dtype=torch.float32, device=self.device)
+        total_tok = torch.zeros((), dtype=torch.float32, device=self.device)
+        for i in range(n):
+            batch = self.data_manager.generate_validation_batch(i)
+            if batch is None:
+                break
+            batch = batch.to(self.device)
+            loss, ntok = self.compute_loss(batch[:, :-1], batch[:, 1:])
+            total += loss.float() * ntok.float()
+            total_tok += ntok.float()
+        self.model.train()
+        total = all_reduce_scalar(total)
+        total_tok = all_reduce_scalar(total_tok)
+        if total_tok.item() == 0:
+            return None
+        return (total / total_tok).item()
+
+    # ------------------------------------------------------------------
+    def save_checkpoint(self, tag: str, val_loss: Optional[float] = None) -> None:
+        if not self.is_main:
+            barrier()
+            return
+        base = str(self.checkpoint_dir / f"step_{tag}")
+        if isinstance(self.optimizer, FusedFlatAdamW):
+            opt_state = self.optimizer.state_dict()
+        else:
+            opt_state = self.optimizer.state_dict()
+        training_state = {
+            "step": self.current_step,
+            "total_tokens": int(self.total_tokens),
+            "validation_losses": self.validation_losses,
+            "val_ptr": getattr(self.data_manager, "val_ptr", 0),
+        }
+        save_checkpoint(base, self.model, opt_state, training_state)
+        update_metadata(
+            self.run_dir,
+            {"tag": str(tag), "step": self.current_step, "val_loss": val_loss,
+             "path": base, "time": time.time()},
+        )
+        if self.config.logging.max_snapshots > 0:
+            rotate_snapshots(self.checkpoint_dir, self.config.logging.max_snapshots)
+        barrier()
+
+    def load_checkpoint(self, checkpoint_base: str, reset_optimizer: bool = False,
+                        reset_training_state: bool = False) -> None:
+        opt_state, training_state = load_checkpoint(checkpoint_base, self.model,
+                                                    map_location=str(self.device))
+        if self.flat_space is not None:
+            # re-sync the flat buffer with the freshly loaded param data
+            for n, off, numel, shape in self.flat_space.segments:
+                p = self.flat_space.name_to_param[n]
+                # p.data already IS the flat view; load_state_dict copied into it
+            if isinstance(self.optimizer, FusedFlatAdamW):
+                self.optimizer.master.copy_(
+                    self.flat_space.flat_param[
+                        self.optimizer.shard_start : self.optimizer.shard_end
+                    ].float()
+                )
+        if opt_state is not None and not reset_optimizer and self.for_training:
+            try:
+                self.optimizer.load_state_dict(opt_state)
+            except Exception as e:
+                self.logger.log(f"Warning: optimizer state not restored ({e}); starting fresh")
+        if not reset_training_state:
+            self.start_step = int(training_state.get("step", 0))
+            self.total_tokens = int(training_state.get("total_tokens", 0))
+            self.validation_losses = list(training_state.get("validation_losses", []))
+        broadcast_module(self.model)
+
+    # ------------------------------------------------------------------
+    def run_learning_rate_finder(self) -> Dict[str, Any]:
+        """Exponential LR sweep (parity: /root/reference/core/training.py:671-761,
+        :1480-1537). Writes lr_finder.csv in the run dir and returns the
+        suggestion."""
+        lf = self.config.training.lr_finder or {}
+        min_lr = float(lf.get("min_lr", 1e-7))
+        max_lr = float(lf.get("max_lr", 1.0))
+        num_steps = int(lf.get("num_steps", 100))
+        mult = (max_lr / min_lr) ** (1.0 / max(num_steps - 1, 1))
+        records = []
+        for i in range(num_steps):
+            lr = min_lr * (mult**i)
+            self.lr_schedule = lambda step, _lr=lr: _lr
+            loss, _ = self.train_step(i)
+            loss_v = float(all_reduce_scalar(loss.float(), "mean").item())
+            records.append((lr, loss_v))
+            if not math.isfinite(loss_v) or (records and loss_v > 4 * min(r[1] for r in records)):
+                break
+        # steepest descent suggestion
+        best_lr = records[0][0]
+        best_slope = 0.0
+        for j in range(1, len(records)):
+            dlr = math.log(records[j][0]) - math.log(records[j - 1][0])
+            slope = (records[j][1] - records[j - 1][1]) / dlr
+            if slope < best_slope:
+                best_slope = slope
+                best_lr = records[j][0]
+        if self.is_main:
+            with open(self.run_dir / "lr_finder.csv", "w", newline="") as f:
+                w = csv.writer(f)
+                w.writerow(["lr", "loss"])
+                w.writerows(records)
+            self.logger.log(f"LR finder suggestion: {best_lr:.3e}")
+        return {"suggested_lr": best_lr, "records": records}
+
+    # ------------------------------------------------------------------
+    def generate_sample(self, prompt: str = "", max_tokens: int = 64) -> str:
+        from ..inference.generate import generate
+
+        try:
+            text, _stats = generate(
+                self.model, self.tokenizer, prompt, max_tokens=max_tokens, temperature=0.8
+            )
+            return text
+        except Exception as e:  # sampling must never kill training
+            return f"<generation failed: {e}>"
+
+    # ------------------------------------------------------------------
+    def train(self) -> None:
+        cfg = self.config
+        steps_cfg = cfg.logging.steps
+        log_interval = int(steps_cfg.get("logging_interval", 1))
+        ckpt_interval = int(steps_cfg.get("checkpoint_interval", 0))
+        val_interval = int(steps_cfg.get("validation_interval", 0))
+
+        self.current_step = 0
+        if cfg.resume is not None:
+            self.load_checkpoint(
+                cfg.resume.checkpoint,
+                reset_optimizer=cfg.resume.reset_optimizer,
+                reset_training_state=cfg.resume.reset_training_state,
+            )
+            self.logger.log(f"Resumed from {cfg.resume.checkpoint} at step {self.start_step}")
+
+        if (cfg.training.lr_finder or {}).get("enabled"):
+            self.run_learning_rate_finder()
+            return
+
+        self.model.train()
+        start_time = time.time()
+        val_loss = None
+        stop = False
+        for step in range(self.start_step, self.total_steps):
+            loss, ntok = self.train_step(step)
+            self.current_step = step + 1  # completed steps
+
+            loss_g = all_reduce_scalar(loss.float(), "mean")
+            ntok_g = all_reduce_scalar(ntok.float())
+            loss_v = float(loss_g.item())
+            ntok_v = int(ntok_g.item())
+            self.total_tokens += ntok_v
+
+            if val_interval and (step + 1) % val_interval == 0:
+                val_loss = self.validate()
+                if val_loss is not None:
+                    self.validation_losses.append((step + 1, val_loss))
+                    if self.early_stopping is not None and self.early_stopping.update(val_loss):
+                        self.logger.log(f"Early stopping at step {step + 1} (val_loss={val_loss:.4f})")
+                        stop = True
+                if cfg.logging.log_samples and self.is_main:
+                    for i in range(cfg.logging.log_samples_count):
+                        self.logger.log(f"Sample {i}: {self.generate_sample()!r}", console=False)
+
+            if log_interval and (step + 1) % log_interval == 0:
+                line = format_metrics(
+                    step + 1, loss_v, ntok_v, self.total_tokens, start_time,
+                    self.lr_schedule(step),
+                    val_loss=val_loss,
+                    metrics_flags=cfg.logging.metrics,
+                    epochs=cfg.training.epochs,
+                    steps_per_epoch=self.steps_per_epoch,
+                    grad_accum_steps=self.grad_accum_steps,
+                    effective_batch_size=self.batch_size
+                    * self.grad_accum_steps
+                    * self.world_size,
+                )
+                if cfg.logging.log_gradient_norm and self.flat_space is not None:
+                    gnorm = float(self.flat_space.flat_grad.float().norm().item())
+                    line += f" | grad_norm={gnorm:.3e}"
+                self.logger.log_metrics_line(step + 1, line)
+                self.logger.tb_scalars(step + 1, {"train/loss": loss_v, "train/lr": self.lr_schedule(step)})
+                if cfg.logging.log_memory_usage:
+                    self.logger.log_memory_usage()
+                val_loss = None
+
+            if ckpt_interval and (step + 1) % ckpt_interval == 0:
+                self.save_checkpoint(str(step + 1), val_loss)
+            if stop:
+                break
+
+        final_val = self.validate()
+        if final_val is not None:
+            self.validation_losses.append((self.current_step, final_val))
+        self.save_checkpoint("final", final_val)
+        self.logger.log(
+            f"Training complete: {self.total_tokens} tokens, final val_loss="
+            f"{final_val if final_val is not None else float('nan')}"
+        )
+        self.logger.close()

@@ -1,0 +1,258 @@
+"""Attention ops: flash (tiled, online-softmax) + flex (block-sparse/score-mod).
+
+Replaces the reference's three Python attention modules
+(/root/reference/models/attention/{flash,flex,simple}_attention.py — all of
+which materialize S×S scores; the "flash" one says so at flash_attention.py:100)
+with a real tiled FlashAttention-2 fwd/bwd HIP kernel (csrc/attn_fwd.hip,
+csrc/attn_bwd.hip): MFMA bf16 tiles, LDS-staged K/V with XOR swizzle, online
+softmax, causal/sliding-window block skip, GQA without materializing repeated
+K/V. FlexAttention's score/mask mods become kernel template variants
+(MOD_* codes below) instead of per-element Python callbacks.
+
+Tensor layout: BSHD — q [B, S, Hq, D], k/v [B, S, Hkv, D]; Hq % Hkv == 0.
+(BSHD is what the QKV projection GEMM produces; keeping it end-to-end means
+zero transpose copies on the hot path.)
+"""
+from __future__ import annotations
+
+import math
+from typing import Callable, Optional, Tuple
+
+import torch
+
+from ._ext import get_ext, use_hip
+
+# Mask/score-mod variant codes understood by the kernels.
+MOD_NONE = 0  # full (bidirectional) attention
+MOD_CAUSAL = 1
+MOD_SLIDING_WINDOW = 2  # causal AND (q - kv) < window
+MOD_PREFIX_LM = 3  # bidirectional over [0, prefix) then causal
+MOD_ALIBI = 4  # causal + alibi slope * (kv - q)
+
+
+def _mods_to_code(causal: bool, window: Optional[int], prefix_len: Optional[int], alibi: bool):
+    if alibi:
+        return MOD_ALIBI
+    if window is not None:
+        return MOD_SLIDING_WINDOW
+    if prefix_len is not None:
+        return MOD_PREFIX_LM
+    return MOD_CAUSAL if causal else MOD_NONE
+
+
+def attention_ref(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = True,
+    scale: Optional[float] = None,
+    window: Optional[int] = None,
+    prefix_len: Optional[int] = None,
+    alibi_slopes: Optional[torch.Tensor] = None,
+    score_mod: Optional[Callable] = None,
+    mask_mod: Optional[Callable] = None,
+    return_lse: bool = False,
+):
+    """fp32 reference composition (used on CPU and as the test oracle).
+
+    q: [B, S, Hq, D]; k, v: [B, Skv, Hkv, D] (BSHD).
+    score_mod(scores[B,H,S,S], b_idx, h_idx, q_idx, kv_idx) -> scores, applied
+    to the scaled scores (FlexAttention semantics, vectorized over the full
+    score tensor). mask_mod(b, h, q_idx, kv_idx) -> bool keep-mask.
+    """
+    B, S, Hq, D = q.shape
+    Hkv = k.shape[2]
+    rep = Hq // Hkv
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    qf = q.float().transpose(1, 2)  # BHSD internally
+    kf = k.float().transpose(1, 2)
+    vf = v.float().transpose(1, 2)
+    if rep > 1:
+        kf = kf.repeat_interleave(rep, dim=1)
+        vf = vf.repeat_interleave(rep, dim=1)
+    scores = torch.einsum("bhsd,bhtd->bhst", qf, kf) * scale
+
+    Skv = k.shape[1]
+    # Absolute query positions: with a KV cache, the S query rows sit at the
+    # END of the Skv keys (decode: S=1 at position Skv-1).
+    q_idx = (torch.arange(S, device=q.device) + (Skv - S)).view(S, 1)
+    kv_idx = torch.arange(Skv, device=q.device).view(1, Skv)
+    keep = torch.ones(S, Skv, dtype=torch.bool, device=q.device)
+    if causal or window is not None or prefix_len is not None:
+        causal_keep = kv_idx <= q_idx
+        if prefix_len is not None:
+            keep = causal_keep | (kv_idx < prefix_len)
+        elif window is not None:
+            keep = causal_keep & (q_idx - kv_idx < window)
+        elif causal:
+            keep = causal_keep
+    if alibi_slopes is not None:
+        bias = alibi_slopes.float().view(1, Hq, 1, 1) * (kv_idx - q_idx).float().view(1, 1, S, Skv)
+        scores = scores + bias
+        keep = keep & (kv_idx <= q_idx)
+    if score_mod is not None:
+        b_idx = torch.arange(B, device=q.device)
+        h_idx = torch.arange(Hq, device=q.device)
+        scores = score_mod(scores, b_idx, h_idx, q_idx.view(-1), kv_idx.view(-1))
+    if mask_mod is not None:
+        b_idx = torch.arange(B, device=q.device)
+        h_idx = torch.arange(Hq, device=q.device)
+        keep = keep & mask_mod(b_idx, h_idx, q_idx.view(-1), kv_idx.view(-1))
+    scores = scores.masked_fill(~keep.view(1, 1, S, Skv), float("-inf"))
+    lse = torch.logsumexp(scores, dim=-1)
+    p = torch.softmax(scores, dim=-1)
+    # rows with no valid keys produce NaN from softmax over all -inf; zero them
+    p = torch.nan_to_num(p, nan=0.0)
+    o = torch.einsum("bhst,bhtd->bhsd", p, vf).transpose(1, 2)  # back to BSHD
+    if return_lse:
+        return o.to(q.dtype).contiguous(), lse
+    return o.to(q.dtype).contiguous()
+
+
+class _FlashAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale, window, prefix_len, alibi_slopes):
+        scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        if use_hip(q, k, v):
+            ext = get_ext()
+            mod = _mods_to_code(causal, window, prefix_len, alibi_slopes is not None)
+            modarg = int(
+                window
+                if window is not None
+                else (prefix_len if prefix_len is not None else 0)
+            )
+            slopes = (
+                alibi_slopes.float().contiguous()
+                if alibi_slopes is not None
+                else torch.empty(0, dtype=torch.float32, device=q.device)
+            )
+            o, lse = ext.attn_fwd(
+                q.contiguous(), k.contiguous(), v.contiguous(), scale, mod, modarg, slopes
+            )
+            ctx.save_for_backward(q, k, v, o, lse, slopes)
+            ctx.meta = (causal, scale, window, prefix_len, True)
+            return o
+        o, lse = attention_ref(
+            q, k, v, causal=causal, scale=scale, window=window,
+            prefix_len=prefix_len, alibi_slopes=alibi_slopes, return_lse=True,
+        )
+        slopes = (
+            alibi_slopes.float().contiguous()
+            if alibi_slopes is not None
+            else torch.empty(0, dtype=torch.float32)
+        )
+        ctx.save_for_backward(q, k, v, o, lse, slopes)
+        ctx.meta = (causal, scale, window, prefix_len, False)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse, slopes = ctx.saved_tensors
+        causal, scale, window, prefix_len, hip = ctx.meta
+        if hip:
+            ext = get_ext()
+            mod = _mods_to_code(causal, window, prefix_len, slopes.numel() > 0)
+            modarg = int(
+                window if window is not None else (prefix_len if prefix_len is not None else 0)
+            )
+            dq, dk, dv = ext.attn_bwd(
+                q.contiguous(), k.contiguous(), v.contiguous(),
+                o.contiguous(), do.contiguous(), lse, scale, mod, modarg, slopes,
+            )
+            return dq, dk, dv, None, None, None, None, None
+        # CPU reference backward: recompute with autograd in fp32
+        with torch.enable_grad():
+            qf = q.detach().float().requires_grad_(True)
+            kf = k.detach().float().requires_grad_(True)
+            vf = v.detach().float().requires_grad_(True)
+            alibi = slopes if slopes.numel() > 0 else None
+            of = attention_ref(
+                qf, kf, vf, causal=causal, scale=scale, window=window,
+                prefix_len=prefix_len, alibi_slopes=alibi,
+            )
+            grads = torch.autograd.grad(of, [qf, kf, vf], do.float())
+        return (
+            grads[0].to(q.dtype), grads[1].to(k.dtype), grads[2].to(v.dtype),
+            None, None, None, None, None,
+        )
+
+
+def flash_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = True,
+    scale: Optional[float] = None,
+    window: Optional[int] = None,
+    prefix_len: Optional[int] = None,
+    alibi_slopes: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """Tiled attention, [B,Hq,S,D] x [B,Hkv,S,D] -> [B,Hq,S,D]."""
+    return _FlashAttnFn.apply(q, k, v, causal, scale, window, prefix_len, alibi_slopes)
+
+
+def flex_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    score_mod: Optional[Callable] = None,
+    mask_mod: Optional[Callable] = None,
+    block_mask=None,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Programmable attention (FlexAttention parity,
+    /root/reference/models/attention/flex_attention.py:414-562).
+
+    Named patterns route to the kernel (causal / sliding_window / prefix_lm /
+    alibi via ``block_mask.pattern``); arbitrary Python callables run on the
+    fp32 reference path (GPU included — they are inherently per-element
+    Python and exist for parity/experimentation, not the hot path).
+    """
+    if block_mask is not None and score_mod is None and mask_mod is None:
+        p = block_mask
+        return flash_attention(
+            q, k, v,
+            causal=p.pattern in ("causal", "sliding_window", "alibi"),
+            scale=scale,
+            window=p.window if p.pattern == "sliding_window" else None,
+            prefix_len=p.prefix_len if p.pattern == "prefix_lm" else None,
+            alibi_slopes=getattr(p, "alibi_slopes", None),
+        )
+    if score_mod is None and mask_mod is None:
+        return flash_attention(q, k, v, causal=True, scale=scale)
+    return attention_ref(q, k, v, causal=False, scale=scale, score_mod=score_mod, mask_mod=mask_mod)
+
+
+class BlockMask:
+    """Named block-sparsity pattern (device-side block mask precompute lives in
+    the kernel's block-skip logic; this object carries the pattern).
+
+    Parity surface: create_block_mask
+    (/root/reference/models/attention/flex_attention.py:356-411)."""
+
+    def __init__(
+        self,
+        pattern: str = "causal",
+        window: Optional[int] = None,
+        prefix_len: Optional[int] = None,
+        alibi_slopes: Optional[torch.Tensor] = None,
+        block_size: int = 128,
+    ):
+        assert pattern in ("causal", "full", "sliding_window", "prefix_lm", "alibi")
+        self.pattern = pattern
+        self.window = window
+        self.prefix_len = prefix_len
+        self.alibi_slopes = alibi_slopes
+        self.block_size = block_size
+
+
+def create_block_mask(
+    mask_mod: Optional[Callable] = None,
+    B: int = 1,
+    H: int = 1,
+    Q_LEN: int = 0,
+    KV_LEN: int = 0,
+    pattern: str = "causal",
+    **kwargs,
+) -> BlockMask:
+    return BlockMask(pattern=pattern, **kwargs)

@@ -1,0 +1,51 @@
+from pathlib import Path
+
+import yaml
+
+from mlx_cuda_distributed_pretraining_amd.core.config import Config
+
+REPO = Path(__file__).resolve().parents[1]
+
+
+def test_sample_config_parses():
+    cfg = Config.from_yaml(str(REPO / "configs" / "model-config-sample.yaml"))
+    assert cfg.name == "llama-2m-sample"
+    assert cfg.model.dimensions["hidden_size"] == 128
+    assert cfg.training.hyperparameters["batch_size"] == 16
+    assert cfg.system.seed == 42
+
+
+def test_reference_style_config_parses(tmp_path):
+    # the reference's schema fields (epochs at training top level, resume block)
+    d = {
+        "name": "t",
+        "data": {"input_file": "x.jsonl", "preprocessing": {"max_context_size": 64}},
+        "model": {"architecture": "llama", "dimensions": {"hidden_size": 32}},
+        "training": {"epochs": 2, "hyperparameters": {"batch_size": 4}},
+        "logging": {"log_dir": "logs", "checkpoint_dir": "ckpt"},
+        "system": {"seed": 1, "device": "cpu"},
+        "resume": {"checkpoint": "runs/t/checkpoints/step_10"},
+    }
+    p = tmp_path / "c.yaml"
+    p.write_text(yaml.safe_dump(d))
+    cfg = Config.from_yaml(str(p))
+    assert cfg.training.epochs == 2
+    assert cfg.resume.checkpoint.endswith("step_10")
+
+
+def test_config_roundtrip(tmp_path):
+    cfg = Config.from_yaml(str(REPO / "configs" / "model-config-1b.yaml"))
+    out = tmp_path / "out.yaml"
+    cfg.save_yaml(out)
+    cfg2 = Config.from_yaml(str(out))
+    assert cfg2.model.dimensions == cfg.model.dimensions
+    assert cfg2.training.hyperparameters == cfg.training.hyperparameters
+
+
+def test_all_baseline_configs_parse():
+    for name in [
+        "model-config-sample", "model-config-124m", "model-config-1b",
+        "model-config-400m-muon", "model-config-256m-flex",
+    ]:
+        cfg = Config.from_yaml(str(REPO / "configs" / f"{name}.yaml"))
+        assert cfg.name

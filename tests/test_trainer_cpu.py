@@ -1,0 +1,130 @@
+"""End-to-end trainer on the 2M sample config (CPU)."""
+import json
+import re
+from pathlib import Path
+
+import torch
+
+from mlx_cuda_distributed_pretraining_amd.core.config import Config
+from mlx_cuda_distributed_pretraining_amd.core.trainer import EarlyStoppingMonitor, Trainer
+
+REPO = Path(__file__).resolve().parents[1]
+
+
+def small_cfg(tmp_runs, **overrides):
+    cfg = Config.from_yaml(str(REPO / "configs" / "model-config-sample.yaml"))
+    cfg.model.dimensions = {"hidden_size": 32, "intermediate_size": 64, "num_layers": 2}
+    cfg.model.attention = {"num_heads": 2, "num_kv_heads": None, "head_dim": None,
+                           "max_position_embeddings": None}
+    cfg.data.preprocessing["max_context_size"] = 32
+    cfg.training.hyperparameters["batch_size"] = 4
+    cfg.training.hyperparameters["iters"] = 12
+    cfg.logging.steps = {"logging_interval": 4, "checkpoint_interval": 0,
+                         "validation_interval": 6}
+    for k, v in overrides.items():
+        setattr(cfg, k, v)
+    return cfg
+
+
+def test_end_to_end_training_run(tmp_runs):
+    cfg = small_cfg(tmp_runs)
+    trainer = Trainer(cfg, runs_root=tmp_runs)
+    trainer.train()
+    run_dir = Path(tmp_runs) / cfg.name
+    assert (run_dir / "config.yaml").exists()
+    log = (run_dir / "log.txt").read_text()
+    # log-line contract (reference format: Step N: loss=... | ppl=... | ...)
+    m = re.search(r"Step \d+: .*loss=\d\.\d+e[+-]\d+ \| ppl=[\d.]+ \| .*tok/s=[\d.]+K", log)
+    assert m, f"log line format mismatch:\n{log}"
+    assert (run_dir / "checkpoints" / "step_final_model.safetensors").exists()
+    meta = json.loads((run_dir / "metadata.json").read_text())
+    assert meta["checkpoints"]
+
+
+def test_loss_decreases_over_run(tmp_runs):
+    cfg = small_cfg(tmp_runs)
+    cfg.training.hyperparameters["iters"] = 30
+    cfg.training.hyperparameters["learning_rate"] = 1e-2
+    cfg.training.scheduler = {"type": "constant"}
+    trainer = Trainer(cfg, runs_root=tmp_runs)
+    losses = []
+    for _ in range(30):
+        trainer.current_step = 0
+        # same batch every time (step=0): the model must memorize it, so the
+        # loss must fall well below the uniform-entropy floor.
+        loss, _ = trainer.train_step(0)
+        losses.append(float(loss))
+    assert sum(losses[-5:]) / 5 < sum(losses[:5]) / 5 * 0.9
+
+
+def test_checkpoint_resume_bitexact(tmp_runs):
+    cfg = small_cfg(tmp_runs)
+    cfg.training.hyperparameters["iters"] = 6
+    cfg.logging.steps["checkpoint_interval"] = 3
+    cfg.logging.steps["validation_interval"] = 0
+    trainer = Trainer(cfg, runs_root=tmp_runs)
+    trainer.train()
+    final_sd = {k: v.clone() for k, v in trainer.model.state_dict().items()}
+
+    # resume from step 3, retrain to 6, expect the same data path
+    cfg2 = small_cfg(tmp_runs)
+    cfg2.name = cfg.name
+    cfg2.overwrite = True
+    cfg2.training.hyperparameters["iters"] = 6
+    cfg2.logging.steps["checkpoint_interval"] = 0
+    cfg2.logging.steps["validation_interval"] = 0
+    from mlx_cuda_distributed_pretraining_amd.core.config import ResumeConfig
+
+    ckpt_base = str(Path(tmp_runs) / cfg.name / "checkpoints" / "step_3")
+    cfg2.resume = ResumeConfig(checkpoint=ckpt_base)
+    trainer2 = Trainer(cfg2, runs_root=tmp_runs)
+    trainer2.train()
+    for k, v in trainer2.model.state_dict().items():
+        assert torch.allclose(v, final_sd[k], atol=1e-5), f"mismatch after resume: {k}"
+
+
+def test_early_stopping_monitor():
+    es = EarlyStoppingMonitor(patience=2, min_delta=0.01)
+    assert not es.update(1.0)
+    assert not es.update(0.5)   # improvement
+    assert not es.update(0.5)   # no improvement (1)
+    assert es.update(0.5)       # no improvement (2) -> stop
+
+
+def test_lr_finder(tmp_runs):
+    cfg = small_cfg(tmp_runs)
+    cfg.training.lr_finder = {"enabled": True, "min_lr": 1e-6, "max_lr": 1e-1, "num_steps": 8}
+    trainer = Trainer(cfg, runs_root=tmp_runs)
+    trainer.train()
+    csv_path = Path(tmp_runs) / cfg.name / "lr_finder.csv"
+    assert csv_path.exists()
+    lines = csv_path.read_text().strip().splitlines()
+    assert len(lines) >= 3
+
+
+def test_grad_accumulation_equivalence(tmp_runs):
+    # accum=2 with bs=2 should equal accum=1 with the same data... we check it runs
+    # and produces finite loss (exact equality needs identical batch composition).
+    cfg = small_cfg(tmp_runs)
+    cfg.name = "accum-test"
+    cfg.training.hyperparameters["gradient_accumulation_steps"] = 2
+    trainer = Trainer(cfg, runs_root=tmp_runs)
+    loss, ntok = trainer.train_step(0)
+    assert torch.isfinite(loss)
+    assert ntok.item() > 0
+
+
+def test_epochs_mode(tmp_runs):
+    import json as _json
+
+    data_file = Path(tmp_runs).parent / "train.jsonl"
+    data_file.parent.mkdir(parents=True, exist_ok=True)
+    data_file.write_text("\n".join(_json.dumps({"text": "hello world " * 5}) for _ in range(16)))
+    cfg = small_cfg(tmp_runs)
+    cfg.name = "epochs-test"
+    cfg.data.synthetic = False
+    cfg.data.input_file = str(data_file)
+    cfg.training.epochs = 2
+    trainer = Trainer(cfg, runs_root=tmp_runs)
+    assert trainer.total_steps == trainer.steps_per_epoch * 2
+    trainer.train()
