@@ -1,0 +1,459 @@
+// Tiled FlashAttention-2 backward (kernel K1 bwd).
+//
+// Deterministic three-kernel design (no atomics):
+//   preprocess: Drow[b,h,s] = sum_d dO*O
+//   dQ kernel : block per q-tile; recomputes P from (Q,K,lse); accumulates
+//               dQ = (P∘(dP−Drow))·scale @ K in registers
+//   dK kernel : block per kv-tile; loops the GQA group's q heads; accumulates
+//               dK = dS^T @ Q in registers
+//   dV kernel : block per kv-tile; accumulates dV = P^T @ dO in registers
+// Splitting dK/dV keeps each kernel under the 256-VGPR budget at D=128
+// (merged, the two 32x128 fp32 accumulators alone are 128 VGPRs).
+// All use the same MFMA layout + acc_to_afrag transform as the forward.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "attn_common.h"
+
+namespace {
+
+constexpr int KVB = 32;
+constexpr int NW = 4;
+constexpr int TPB = NW * WAVE;
+constexpr int QPB = 32 * NW;   // q rows per block (dQ kernel)
+constexpr int KPB = 32 * NW;   // kv rows per block (dK/dV kernels)
+
+// ---------------- preprocess: Drow = rowsum(dO * O) ----------------
+__global__ void bwd_preprocess_kernel(const __hip_bfloat16* __restrict__ dout,
+                                      const __hip_bfloat16* __restrict__ o,
+                                      float* __restrict__ drow,
+                                      long rows, int Hq, int D) {
+  // one wave per (b,s,h) row; rows-major layout is BSHD so row index = (b*S+s)*Hq+h
+  const long row = blockIdx.x * (long)(TPB / WAVE) + threadIdx.x / WAVE;
+  if (row >= rows) return;
+  const int lane = threadIdx.x % WAVE;
+  const __hip_bfloat16* dp = dout + row * D;
+  const __hip_bfloat16* op = o + row * D;
+  float acc = 0.f;
+  for (int i = lane; i < D; i += WAVE) acc += to_f32(dp[i]) * to_f32(op[i]);
+  acc = wave_reduce_sum(acc);
+  // drow stored in BSHD-row order: index (b*S+s)*Hq+h, matching how the
+  // dq/dkv kernels read it.
+  if (lane == 0) drow[row] = acc;
+}
+
+// q-range of kv-tile blocks per mod (in q-row space)
+template <int MOD>
+__device__ __forceinline__ void q_range_for_kv(int kv0, int kpb, int q_off, int Sq,
+                                               int modarg, int& q_lo, int& q_hi) {
+  q_lo = 0;
+  q_hi = Sq;
+  if constexpr (MOD == MOD_CAUSAL || MOD == MOD_ALIBI) {
+    q_lo = max(0, kv0 - q_off) & ~31;
+  } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+    q_lo = max(0, kv0 - q_off) & ~31;
+    q_hi = min(Sq, kv0 + kpb - 1 + modarg - q_off + 1);
+  } else if constexpr (MOD == MOD_PREFIX_LM) {
+    if (kv0 >= modarg) q_lo = max(0, kv0 - q_off) & ~31;
+  }
+}
+
+// ---------------- dQ kernel ----------------
+template <int D, int MOD>
+__global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    __hip_bfloat16* __restrict__ dq, const float* __restrict__ slopes,
+    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg) {
+  constexpr int DBLK = D / 16;
+  constexpr int DCOL = D / 32;
+  constexpr int KSTR = D + 8;
+  constexpr int VSTR = KVB + 8;
+
+  // K row-major + K^T + V row-major
+  __shared__ __hip_bfloat16 smem[KVB * KSTR + D * VSTR + KVB * KSTR];
+  __hip_bfloat16* k_lds = smem;
+  __hip_bfloat16* kt_lds = smem + KVB * KSTR;
+  __hip_bfloat16* v_lds = kt_lds + D * VSTR;
+
+  const int b = blockIdx.z, hq = blockIdx.y, qtile = blockIdx.x;
+  const int hkv = hq / (Hq / Hkv);
+  const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
+  const int lq = lane & 31, hi = lane >> 5;
+  const int q0w = qtile * QPB + wave * 32;
+  const int qrow = q0w + lq;
+  const bool q_valid = qrow < Sq;
+  const int q_off = Skv - Sq;
+  const int q_pos = qrow + q_off;
+  const float slope = (MOD == MOD_ALIBI) ? slopes[hq] : 0.f;
+
+  // Q and dO fragments (B-operand layout: lane holds row q=lq, 8 d values)
+  bf16x8 qf[DBLK], dof[DBLK];
+  {
+    const long base = (((long)b * Sq + (q_valid ? qrow : 0)) * Hq + hq) * (long)D + hi * 8;
+#pragma unroll
+    for (int dblk = 0; dblk < DBLK; ++dblk) {
+      Bf16x8U uq, ud;
+      *reinterpret_cast<uint4*>(uq.s) =
+          q_valid ? *reinterpret_cast<const uint4*>(q + base + dblk * 16) : uint4{0, 0, 0, 0};
+      *reinterpret_cast<uint4*>(ud.s) =
+          q_valid ? *reinterpret_cast<const uint4*>(dout + base + dblk * 16) : uint4{0, 0, 0, 0};
+      qf[dblk] = uq.v;
+      dof[dblk] = ud.v;
+    }
+  }
+  const float Lq = q_valid ? lse[((long)b * Hq + hq) * Sq + qrow] : INFINITY;
+  const float Dq = q_valid ? drow[((long)b * Sq + qrow) * Hq + hq] : 0.f;
+
+  // kv range (same as forward)
+  const int blk_qpos_lo = qtile * QPB + q_off;
+  const int blk_qpos_hi = blk_qpos_lo + QPB - 1;
+  int kv_lo = 0, kv_hi = Skv;
+  if constexpr (MOD == MOD_CAUSAL || MOD == MOD_ALIBI) {
+    kv_hi = min(Skv, blk_qpos_hi + 1);
+  } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+    kv_hi = min(Skv, blk_qpos_hi + 1);
+    kv_lo = max(0, blk_qpos_lo - modarg + 1) & ~(KVB - 1);
+  } else if constexpr (MOD == MOD_PREFIX_LM) {
+    kv_hi = min(Skv, max(blk_qpos_hi + 1, modarg));
+  }
+
+  float dq_acc[DCOL][16];
+#pragma unroll
+  for (int dc = 0; dc < DCOL; ++dc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[dc][r] = 0.f;
+
+  for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
+    {  // stage K (row-major + transposed) and V (row-major)
+      constexpr int U4ROW = D / 8;
+      constexpr int TOT = KVB * U4ROW;
+      for (int u = tid; u < TOT; u += TPB) {
+        const int row = u / U4ROW;
+        const int d0 = (u % U4ROW) * 8;
+        const bool valid = kv0 + row < Skv;
+        const long src = (((long)b * Skv + (valid ? kv0 + row : 0)) * Hkv + hkv) * (long)D + d0;
+        Bf16x8U ku, vu;
+        *reinterpret_cast<uint4*>(ku.s) =
+            valid ? *reinterpret_cast<const uint4*>(k + src) : uint4{0, 0, 0, 0};
+        *reinterpret_cast<uint4*>(vu.s) =
+            valid ? *reinterpret_cast<const uint4*>(v + src) : uint4{0, 0, 0, 0};
+        *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(ku.s);
+        *reinterpret_cast<uint4*>(v_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(vu.s);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kt_lds[(d0 + j) * VSTR + row] = ku.h[j];
+      }
+    }
+    __syncthreads();
+
+    // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
+    f32x16 st = {}, dpt = {};
+#pragma unroll
+    for (int dblk = 0; dblk < DBLK; ++dblk) {
+      Bf16x8U kf, vf;
+      *reinterpret_cast<uint4*>(kf.s) =
+          *reinterpret_cast<const uint4*>(k_lds + lq * KSTR + dblk * 16 + hi * 8);
+      *reinterpret_cast<uint4*>(vf.s) =
+          *reinterpret_cast<const uint4*>(v_lds + lq * KSTR + dblk * 16 + hi * 8);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
+      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf.v, dof[dblk], dpt, 0, 0, 0);
+    }
+
+    float ds[16];
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int k_pos = kv0 + acc_row(reg, hi);
+      const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+      float s = st[reg] * scale;
+      if constexpr (MOD == MOD_ALIBI) s += slope * (k_pos - q_pos);
+      const float p = keep ? __expf(s - Lq) : 0.f;
+      ds[reg] = p * (dpt[reg] - Dq) * scale;
+    }
+
+    bf16x8 da0, da1;
+    acc_to_afrag(ds, da0, da1);  // -> dS[32q x 16k] A-fragments
+#pragma unroll
+    for (int dc = 0; dc < DCOL; ++dc) {
+      f32x16 acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[r] = dq_acc[dc][r];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        Bf16x8U kb;  // B = K[16k x 32d] from K^T image
+        *reinterpret_cast<uint4*>(kb.s) = *reinterpret_cast<const uint4*>(
+            kt_lds + (dc * 32 + lq) * VSTR + ks * 16 + hi * 8);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb.v, acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
+    }
+    __syncthreads();
+  }
+
+  // store dq: element (r=q_local, c=d_local)
+  const int dl = lane & 31;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int r = acc_row(reg, hi);
+    const int q_r = q0w + r;
+    if (q_r >= Sq) continue;
+    __hip_bfloat16* dqr = dq + (((long)b * Sq + q_r) * Hq + hq) * (long)D + dl;
+#pragma unroll
+    for (int dc = 0; dc < DCOL; ++dc) dqr[dc * 32] = __float2bfloat16(dq_acc[dc][reg]);
+  }
+}
+
+// ---------------- dK / dV kernels (block per kv-tile, loop GQA group) -------
+// TEMPLATE WANT_DK: true -> compute dK (needs dP, V frags); false -> dV.
+template <int D, int MOD, bool WANT_DK>
+__global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
+    const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
+    const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ drow,
+    __hip_bfloat16* __restrict__ dkv_out, const float* __restrict__ slopes,
+    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg) {
+  constexpr int DBLK = D / 16;
+  constexpr int DCOL = D / 32;
+  constexpr int KSTR = D + 8;
+  constexpr int VSTR = 32 + 8;
+
+  // q-tile images: row-major Q [32][KSTR]; transposed X^T [D][VSTR] where
+  // X = Q (dK) or dO (dV); plus dO row-major for dP (dK only); lse/drow [32].
+  __shared__ __hip_bfloat16 smem[32 * KSTR + D * VSTR + 32 * KSTR];
+  __shared__ float stats_lds[2][32];
+  __hip_bfloat16* q_lds = smem;                 // Q row-major (A-frags)
+  __hip_bfloat16* xt_lds = smem + 32 * KSTR;    // Q^T (dK) / dO^T (dV)
+  __hip_bfloat16* do_lds = xt_lds + D * VSTR;   // dO row-major (dK only)
+
+  const int b = blockIdx.z, hkv = blockIdx.y, kvtile = blockIdx.x;
+  const int group = Hq / Hkv;
+  const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
+  const int lk = lane & 31, hi = lane >> 5;
+  const int kv0w = kvtile * KPB + wave * 32;  // this wave's kv rows
+  const int krow = kv0w + lk;
+  const bool k_valid = krow < Skv;
+  const int q_off = Skv - Sq;
+
+  // K (and V for dK) fragments: lane holds row k=lk, 8 d values
+  bf16x8 kf[DBLK], vf[WANT_DK ? DBLK : 1];
+  {
+    const long base = (((long)b * Skv + (k_valid ? krow : 0)) * Hkv + hkv) * (long)D + hi * 8;
+#pragma unroll
+    for (int dblk = 0; dblk < DBLK; ++dblk) {
+      Bf16x8U ku;
+      *reinterpret_cast<uint4*>(ku.s) =
+          k_valid ? *reinterpret_cast<const uint4*>(k + base + dblk * 16) : uint4{0, 0, 0, 0};
+      kf[dblk] = ku.v;
+      if constexpr (WANT_DK) {
+        Bf16x8U vu;
+        *reinterpret_cast<uint4*>(vu.s) =
+            k_valid ? *reinterpret_cast<const uint4*>(v + base + dblk * 16) : uint4{0, 0, 0, 0};
+        vf[dblk] = vu.v;
+      }
+    }
+  }
+
+  float acc_out[DCOL][16];
+#pragma unroll
+  for (int dc = 0; dc < DCOL; ++dc)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) acc_out[dc][r] = 0.f;
+
+  int q_lo, q_hi;
+  q_range_for_kv<MOD>(kvtile * KPB, KPB, q_off, Sq, modarg, q_lo, q_hi);
+
+  for (int g = 0; g < group; ++g) {
+    const int hq = hkv * group + g;
+    const float slope = (MOD == MOD_ALIBI) ? slopes[hq] : 0.f;
+    for (int q0 = q_lo; q0 < q_hi; q0 += 32) {
+      {  // stage q-tile: Q rm (+ dO rm for dK), X^T, lse/drow
+        constexpr int U4ROW = D / 8;
+        constexpr int TOT = 32 * U4ROW;
+        for (int u = tid; u < TOT; u += TPB) {
+          const int row = u / U4ROW;
+          const int d0 = (u % U4ROW) * 8;
+          const bool valid = q0 + row < Sq;
+          const long src = (((long)b * Sq + (valid ? q0 + row : 0)) * Hq + hq) * (long)D + d0;
+          Bf16x8U qu, du;
+          *reinterpret_cast<uint4*>(qu.s) =
+              valid ? *reinterpret_cast<const uint4*>(q + src) : uint4{0, 0, 0, 0};
+          *reinterpret_cast<uint4*>(du.s) =
+              valid ? *reinterpret_cast<const uint4*>(dout + src) : uint4{0, 0, 0, 0};
+          *reinterpret_cast<uint4*>(q_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(qu.s);
+          if constexpr (WANT_DK) {
+            *reinterpret_cast<uint4*>(do_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(du.s);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) xt_lds[(d0 + j) * VSTR + row] = qu.h[j];
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) xt_lds[(d0 + j) * VSTR + row] = du.h[j];
+          }
+        }
+        if (tid < 32) {
+          const int qr = q0 + tid;
+          stats_lds[0][tid] = (qr < Sq) ? lse[((long)b * Hq + hq) * Sq + qr] : INFINITY;
+          stats_lds[1][tid] = (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + hq] : 0.f;
+        }
+      }
+      __syncthreads();
+
+      // S = mfma(Q, K^T): A=Q rm frags from LDS, B=K^T = register kf
+      // element (r=q_local, c=k_local)
+      f32x16 s_acc = {}, dp_acc = {};
+      // A-operand row index is lane&31 (= q_local for Q/dO); B-operand column
+      // index is also lane&31 (= k_local for the register K^T/V^T fragments).
+#pragma unroll
+      for (int dblk = 0; dblk < DBLK; ++dblk) {
+        Bf16x8U qa;
+        *reinterpret_cast<uint4*>(qa.s) =
+            *reinterpret_cast<const uint4*>(q_lds + lk * KSTR + dblk * 16 + hi * 8);
+        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa.v, kf[dblk], s_acc, 0, 0, 0);
+        if constexpr (WANT_DK) {
+          Bf16x8U da;
+          *reinterpret_cast<uint4*>(da.s) =
+              *reinterpret_cast<const uint4*>(do_lds + lk * KSTR + dblk * 16 + hi * 8);
+          dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da.v, vf[dblk], dp_acc, 0, 0, 0);
+        }
+      }
+
+      float pv[16];
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int q_r = q0 + acc_row(reg, hi);
+        const int q_pos = q_r + q_off;
+        const int k_pos = krow;
+        const bool keep = k_valid && q_r < Sq && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+        float s = s_acc[reg] * scale;
+        if constexpr (MOD == MOD_ALIBI) s += slope * (k_pos - q_pos);
+        const float p = keep ? __expf(s - stats_lds[0][acc_row(reg, hi)]) : 0.f;
+        if constexpr (WANT_DK) {
+          pv[reg] = p * (dp_acc[reg] - stats_lds[1][acc_row(reg, hi)]) * scale;  // dS
+        } else {
+          pv[reg] = p;
+        }
+      }
+
+      // transform: acc holds M[r=q][c=k]; A-frags of M^T = P^T (dV) or dS^T (dK)
+      bf16x8 a0, a1;
+      acc_to_afrag(pv, a0, a1);
+#pragma unroll
+      for (int dc = 0; dc < DCOL; ++dc) {
+        f32x16 acc;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[r] = acc_out[dc][r];
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          Bf16x8U xb;  // B = X[16q x 32d] from X^T image (X = Q for dK, dO for dV)
+          *reinterpret_cast<uint4*>(xb.s) = *reinterpret_cast<const uint4*>(
+              xt_lds + (dc * 32 + lk) * VSTR + ks * 16 + hi * 8);
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb.v, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc_out[dc][r] = acc[r];
+      }
+      __syncthreads();
+    }
+  }
+
+  // store: element (r=k_local, c=d_local)
+  const int dl = lane & 31;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int r = acc_row(reg, hi);
+    const int k_r = kv0w + r;
+    if (k_r >= Skv) continue;
+    __hip_bfloat16* out = dkv_out + (((long)b * Skv + k_r) * Hkv + hkv) * (long)D + dl;
+#pragma unroll
+    for (int dc = 0; dc < DCOL; ++dc) out[dc * 32] = __float2bfloat16(acc_out[dc][reg]);
+  }
+}
+
+template <int D, int MOD>
+void launch_bwd_all(dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
+                    const __hip_bfloat16* q, const __hip_bfloat16* k, const __hip_bfloat16* v,
+                    const __hip_bfloat16* dout, const float* lse, const float* drow,
+                    __hip_bfloat16* dq, __hip_bfloat16* dk, __hip_bfloat16* dv,
+                    const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
+                    float scale, int modarg) {
+  attn_bwd_dq_kernel<D, MOD><<<gq, block, 0, stream>>>(
+      q, k, v, dout, lse, drow, dq, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+  attn_bwd_dkv_kernel<D, MOD, true><<<gkv, block, 0, stream>>>(
+      q, k, v, dout, lse, drow, dk, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+  attn_bwd_dkv_kernel<D, MOD, false><<<gkv, block, 0, stream>>>(
+      q, k, v, dout, lse, drow, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+}
+
+template <int D>
+void launch_bwd_mod(int mod, dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
+                    const __hip_bfloat16* q, const __hip_bfloat16* k, const __hip_bfloat16* v,
+                    const __hip_bfloat16* dout, const float* lse, const float* drow,
+                    __hip_bfloat16* dq, __hip_bfloat16* dk, __hip_bfloat16* dv,
+                    const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
+                    float scale, int modarg) {
+  switch (mod) {
+    case MOD_NONE:
+      launch_bwd_all<D, MOD_NONE>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      break;
+    case MOD_CAUSAL:
+      launch_bwd_all<D, MOD_CAUSAL>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      break;
+    case MOD_SLIDING_WINDOW:
+      launch_bwd_all<D, MOD_SLIDING_WINDOW>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      break;
+    case MOD_PREFIX_LM:
+      launch_bwd_all<D, MOD_PREFIX_LM>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      break;
+    case MOD_ALIBI:
+      launch_bwd_all<D, MOD_ALIBI>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      break;
+    default:
+      TORCH_CHECK(false, "attn_bwd: unknown mod ", mod);
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+                                 at::Tensor dout, at::Tensor lse, double scale, long mod,
+                                 long modarg, at::Tensor slopes) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && dout.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_bwd: bf16 only");
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1), Hkv = k.size(2);
+  TORCH_CHECK(D == 64 || D == 128, "attn_bwd: head_dim must be 64 or 128");
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto drow = at::empty({(long)B * Sq * Hq}, q.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+
+  {  // preprocess
+    const long rows = (long)B * Sq * Hq;
+    const long grid = cdiv(rows, TPB / WAVE);
+    bwd_preprocess_kernel<<<grid, TPB, 0, stream>>>(
+        reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(o.data_ptr()),
+        drow.data_ptr<float>(), rows, Hq, D);
+  }
+
+  dim3 gq(cdiv(Sq, QPB), Hq, B);
+  dim3 gkv(cdiv(Skv, KPB), Hkv, B);
+  dim3 block(TPB);
+  const float* sl = slopes.numel() > 0 ? slopes.data_ptr<float>() : nullptr;
+  auto* qp = reinterpret_cast<const __hip_bfloat16*>(q.data_ptr());
+  auto* kp = reinterpret_cast<const __hip_bfloat16*>(k.data_ptr());
+  auto* vp = reinterpret_cast<const __hip_bfloat16*>(v.data_ptr());
+  auto* dop = reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr());
+  auto* dqp = reinterpret_cast<__hip_bfloat16*>(dq.data_ptr());
+  auto* dkp = reinterpret_cast<__hip_bfloat16*>(dk.data_ptr());
+  auto* dvp = reinterpret_cast<__hip_bfloat16*>(dv.data_ptr());
+  if (D == 64)
+    launch_bwd_mod<64>((int)mod, gq, gkv, block, stream, qp, kp, vp, dop,
+                       lse.data_ptr<float>(), drow.data_ptr<float>(), dqp, dkp, dvp, sl,
+                       B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg);
+  else
+    launch_bwd_mod<128>((int)mod, gq, gkv, block, stream, qp, kp, vp, dop,
+                        lse.data_ptr<float>(), drow.data_ptr<float>(), dqp, dkp, dvp, sl,
+                        B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg);
+  return {dq, dk, dv};
+}

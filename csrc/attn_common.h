@@ -1,0 +1,77 @@
+// Shared pieces of the tiled attention kernels (K1/K2).
+//
+// MFMA tile convention (gfx950 mfma_f32_32x32x16_bf16, guide §3):
+//   A[32][16]: lane l holds A[l&31][8*(l>>5)+j], j=0..7  (bf16x8)
+//   B[16][32]: lane l holds B[8*(l>>5)+j][l&31]
+//   C/D[32][32] f32x16: element (r,c) lives in lane c+32*((r>>2)&1),
+//                       reg (r&3)+4*(r>>3); i.e. row r(reg,hi)=(reg&3)+8*(reg>>2)+4*hi.
+//
+// The workhorse trick (guide T12): QK^T is computed SWAPPED — mfma(K, Q)
+// yields S^T with the q index lane-local (lane&31 = q column), so the online
+// softmax runs entirely in registers; acc_to_afrag() then converts the
+// exp'd accumulator into the A-operand fragments of the P·V MFMA with
+// 8 v_cvt_pk_bf16_f32 + 4 v_permlane32_swap per 32x32 tile.
+#pragma once
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+// attention mask/score-mod variant codes (must match ops/attention.py)
+#define MOD_NONE 0
+#define MOD_CAUSAL 1
+#define MOD_SLIDING_WINDOW 2
+#define MOD_PREFIX_LM 3
+#define MOD_ALIBI 4
+
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned packed;
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(packed) : "v"(lo), "v"(hi));
+  return packed;
+}
+
+union Bf16x8U {
+  bf16x8 v;
+  unsigned u[4];
+  __hip_bfloat16 h[8];
+  ushort s[8];
+};
+
+// Convert a 32x32 f32 accumulator (element M[r][c] at lane c+32*((r>>2)&1))
+// into the two A-operand fragments of A = M^T (A[i][j] = M[j][i]):
+// frag[ks] lane l = A[l&31][16*ks + 8*(l>>5) + j]. Guide T12.
+__device__ __forceinline__ void acc_to_afrag(const float (&p)[16], bf16x8& f0, bf16x8& f1) {
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    unsigned a0 = cvt_pk_bf16(p[8 * ks + 0], p[8 * ks + 1]);
+    unsigned a1 = cvt_pk_bf16(p[8 * ks + 2], p[8 * ks + 3]);
+    unsigned b0 = cvt_pk_bf16(p[8 * ks + 4], p[8 * ks + 5]);
+    unsigned b1 = cvt_pk_bf16(p[8 * ks + 6], p[8 * ks + 7]);
+    auto r0 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+    auto r1 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+    Bf16x8U out;
+    out.u[0] = r0[0];
+    out.u[1] = r1[0];
+    out.u[2] = r0[1];
+    out.u[3] = r1[1];
+    if (ks == 0) f0 = out.v; else f1 = out.v;
+  }
+}
+
+// row index held by (reg, hi) in a 32x32 accumulator
+__device__ __forceinline__ int acc_row(int reg, int hi) {
+  return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
+}
+
+// keep/mask decision for one score element. q_pos/k_pos are ABSOLUTE
+// positions (q_pos = q_row + Skv - Sq handles KV-cache decode).
+template <int MOD>
+__device__ __forceinline__ bool attn_keep(int q_pos, int k_pos, int skv, int modarg) {
+  if (k_pos >= skv) return false;
+  if constexpr (MOD == MOD_NONE) return true;
+  if constexpr (MOD == MOD_CAUSAL || MOD == MOD_ALIBI) return k_pos <= q_pos;
+  if constexpr (MOD == MOD_SLIDING_WINDOW)
+    return k_pos <= q_pos && (q_pos - k_pos) < modarg;
+  if constexpr (MOD == MOD_PREFIX_LM) return k_pos <= q_pos || k_pos < modarg;
+  return true;
+}

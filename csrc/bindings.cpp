@@ -1,0 +1,58 @@
+// Python bindings for the MI355X HIP kernel layer (_mcdp_C).
+#include <torch/extension.h>
+
+// rmsnorm.hip
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor rstd, at::Tensor dy);
+// rope.hip
+at::Tensor rope_fwd(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
+                    long offset, bool conj);
+// swiglu.hip
+at::Tensor swiglu_fwd(at::Tensor gu);
+at::Tensor swiglu_bwd(at::Tensor gu, at::Tensor dy);
+// cross_entropy.hip
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets, long ignore_index);
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor scale,
+                  long ignore_index);
+// optim.hip
+at::Tensor sumsq(at::Tensor g);
+void adamw_step(at::Tensor param, at::Tensor master, at::Tensor grad, at::Tensor m,
+                at::Tensor v, at::Tensor sumsq_t, long step, double lr, double b1,
+                double b2, double eps, double wd, long decay_boundary, double max_norm);
+void lion_step(at::Tensor param, at::Tensor master, at::Tensor grad, at::Tensor m,
+               at::Tensor sumsq_t, double lr, double b1, double b2, double wd,
+               long decay_boundary, double max_norm);
+void sgd_step(at::Tensor param, at::Tensor master, at::Tensor grad, at::Tensor buf,
+              at::Tensor sumsq_t, double lr, double mom, double wd, long decay_boundary,
+              bool nesterov, double max_norm);
+// attn_fwd.hip / attn_bwd.hip
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale,
+                                 long mod, long modarg, at::Tensor slopes);
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+                                 at::Tensor dout, at::Tensor lse, double scale, long mod,
+                                 long modarg, at::Tensor slopes);
+// sampling.hip
+at::Tensor sample_token(at::Tensor logits, double temperature, double top_p, double min_p,
+                        long seed);
+// debug.hip
+at::Tensor mfma_tile_test(at::Tensor A, at::Tensor B);
+at::Tensor afrag_transform_test(at::Tensor M);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (y, rstd)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (dx, dw)");
+  m.def("rope_fwd", &rope_fwd, "RoPE apply (conj=true for backward)");
+  m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");
+  m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");
+  m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (loss_sum, ntok, lse)");
+  m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward (dlogits)");
+  m.def("sumsq", &sumsq, "sum of squares -> f32 scalar");
+  m.def("adamw_step", &adamw_step, "fused AdamW over flat buffers");
+  m.def("lion_step", &lion_step, "fused Lion over flat buffers");
+  m.def("sgd_step", &sgd_step, "fused SGD over flat buffers");
+  m.def("attn_fwd", &attn_fwd, "flash attention forward (o, lse)");
+  m.def("attn_bwd", &attn_bwd, "flash attention backward (dq, dk, dv)");
+  m.def("sample_token", &sample_token, "fused temperature/min-p sampling");
+  m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
+  m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");
+}

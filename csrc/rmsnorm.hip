@@ -1,0 +1,146 @@
+// RMSNorm forward/backward (kernel K3).
+// One workgroup per row, fp32 accumulation, bf16x8 vectorized loads.
+// Replaces /root/reference/models/llama.py:44-56 (Python-composed RMSNorm).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+// ---------------- forward ----------------
+// y = x * rsqrt(mean(x^2) + eps) * w ; saves rstd per row for backward.
+template <typename T, int BLOCK>
+__global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                   T* __restrict__ y, float* __restrict__ rstd,
+                                   int H, float eps) {
+  __shared__ float scratch[BLOCK / WAVE];
+  const long row = blockIdx.x;
+  const T* xr = x + row * (long)H;
+  T* yr = y + row * (long)H;
+
+  float ss = 0.f;
+  if constexpr (sizeof(T) == 2) {
+    const int HV = H / 8;
+    const uint4* xv = reinterpret_cast<const uint4*>(xr);
+    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+      U4 u; u.u = xv[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { float f = bf16_bits_to_f32(u.s[j]); ss += f * f; }
+    }
+    for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK) {
+      float f = to_f32(xr[i]); ss += f * f;
+    }
+  } else {
+    for (int i = threadIdx.x; i < H; i += BLOCK) { float f = to_f32(xr[i]); ss += f * f; }
+  }
+  ss = block_reduce_sum<BLOCK>(ss, scratch);
+  const float r = rsqrtf(ss / H + eps);
+  if (threadIdx.x == 0) rstd[row] = r;
+
+  if constexpr (sizeof(T) == 2) {
+    const int HV = H / 8;
+    const uint4* xv = reinterpret_cast<const uint4*>(xr);
+    const uint4* wv = reinterpret_cast<const uint4*>(w);
+    uint4* yv = reinterpret_cast<uint4*>(yr);
+    for (int i = threadIdx.x; i < HV; i += BLOCK) {
+      U4 u, ww, o; u.u = xv[i]; ww.u = wv[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.s[j] = f32_to_bf16_bits(bf16_bits_to_f32(u.s[j]) * r * bf16_bits_to_f32(ww.s[j]));
+      yv[i] = o.u;
+    }
+    for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK)
+      from_f32(&yr[i], to_f32(xr[i]) * r * to_f32(w[i]));
+  } else {
+    for (int i = threadIdx.x; i < H; i += BLOCK)
+      from_f32(&yr[i], to_f32(xr[i]) * r * to_f32(w[i]));
+  }
+}
+
+// ---------------- backward ----------------
+// dx = r*dy*w - x * r^3/H * sum(dy*w*x) ; dw_partial[blk] += dy * x * r
+// Grid-stride over rows; per-block dw accumulated in LDS then one global add.
+template <typename T, int BLOCK>
+__global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                   const float* __restrict__ rstd,
+                                   const T* __restrict__ dy, T* __restrict__ dx,
+                                   float* __restrict__ dw_partial,
+                                   long rows, int H) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* dw_acc = reinterpret_cast<float*>(smem);          // [H]
+  float* scratch = dw_acc + H;                             // [BLOCK/WAVE]
+
+  for (int i = threadIdx.x; i < H; i += BLOCK) dw_acc[i] = 0.f;
+  __syncthreads();
+
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + row * (long)H;
+    const T* dyr = dy + row * (long)H;
+    T* dxr = dx + row * (long)H;
+    const float r = rstd[row];
+
+    float c = 0.f;
+    for (int i = threadIdx.x; i < H; i += BLOCK)
+      c += to_f32(dyr[i]) * to_f32(w[i]) * to_f32(xr[i]);
+    c = block_reduce_sum<BLOCK>(c, scratch);
+    const float k = r * r * r * c / H;
+    for (int i = threadIdx.x; i < H; i += BLOCK) {
+      const float xi = to_f32(xr[i]);
+      const float dyi = to_f32(dyr[i]);
+      from_f32(&dxr[i], r * dyi * to_f32(w[i]) - xi * k);
+      dw_acc[i] += dyi * xi * r;
+    }
+    __syncthreads();
+  }
+  float* dwp = dw_partial + blockIdx.x * (long)H;
+  for (int i = threadIdx.x; i < H; i += BLOCK) dwp[i] = dw_acc[i];
+}
+
+}  // namespace
+
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto y = at::empty_like(x);
+  auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  constexpr int BLOCK = 256;
+  dim3 grid(rows);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "rmsnorm_fwd", [&] {
+    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
+    if constexpr (std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>) {
+      rmsnorm_fwd_kernel<T, BLOCK><<<grid, BLOCK, 0, stream>>>(
+          reinterpret_cast<const T*>(x.data_ptr()), reinterpret_cast<const T*>(w.data_ptr()),
+          reinterpret_cast<T*>(y.data_ptr()), rstd.data_ptr<float>(), H, (float)eps);
+    } else {
+      TORCH_CHECK(false, "rmsnorm: unsupported dtype");
+    }
+  });
+  return {y, rstd};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor rstd, at::Tensor dy) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && dy.is_contiguous());
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto dx = at::empty_like(x);
+  constexpr int BLOCK = 256;
+  const int nblocks = (int)std::min<long>(rows, 1024);
+  auto dw_partial = at::zeros({nblocks, H}, x.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  const size_t lds = (H + BLOCK / WAVE) * sizeof(float);
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, x.scalar_type(), "rmsnorm_bwd", [&] {
+    using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
+    if constexpr (std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>) {
+      rmsnorm_bwd_kernel<T, BLOCK><<<nblocks, BLOCK, lds, stream>>>(
+          reinterpret_cast<const T*>(x.data_ptr()), reinterpret_cast<const T*>(w.data_ptr()),
+          rstd.data_ptr<float>(), reinterpret_cast<const T*>(dy.data_ptr()),
+          reinterpret_cast<T*>(dx.data_ptr()), dw_partial.data_ptr<float>(), rows, H);
+    } else {
+      TORCH_CHECK(false, "rmsnorm: unsupported dtype");
+    }
+  });
+  auto dw = dw_partial.sum(0);
+  return {dx, dw};
+}

@@ -17,7 +17,10 @@ from torch.utils import cpp_extension  # noqa: E402
 ROOT = Path(__file__).resolve().parent
 CSRC = ROOT / "csrc"
 
-sources = sorted(str(p) for p in CSRC.glob("*.hip")) + [str(CSRC / "bindings.cpp")]
+# exclude the *_hip.hip copies torch's hipify pass generates in-place
+sources = sorted(
+    str(p) for p in CSRC.glob("*.hip") if not p.name.endswith("_hip.hip")
+) + [str(CSRC / "bindings.cpp")]
 
 setup(
     name="mlx_cuda_distributed_pretraining_amd",

@@ -1,0 +1,253 @@
+"""HIP kernel numerics on a real MI355X: each kernel vs the plain PyTorch
+fp32 reference of the same op (tolerances per dtype)."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def dev():
+    return torch.device("cuda:0")
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from mlx_cuda_distributed_pretraining_amd.ops import require_ext
+
+    return require_ext()
+
+
+# ---------------- MFMA layout self-checks (run these FIRST) ----------------
+def test_mfma_tile_layout(ext):
+    torch.manual_seed(0)
+    # ASYMMETRIC inputs (guide: symmetric B passes transposed writes)
+    A = (torch.arange(32 * 16, device=dev(), dtype=torch.float32).reshape(32, 16) % 7 - 3)
+    B = (torch.arange(16 * 32, device=dev(), dtype=torch.float32).reshape(16, 32) % 5 - 2) * 0.5
+    A += torch.randn(32, 16, device=dev())
+    B += torch.randn(16, 32, device=dev())
+    C = ext.mfma_tile_test(A.bfloat16(), B.bfloat16())
+    want = A.bfloat16().float() @ B.bfloat16().float()
+    assert torch.allclose(C, want, atol=1e-2, rtol=1e-2), (C - want).abs().max()
+
+
+def test_afrag_transform(ext):
+    torch.manual_seed(1)
+    M = torch.randn(32, 32, device=dev(), dtype=torch.float32)
+    A = ext.afrag_transform_test(M)
+    want = M.t().bfloat16().float()
+    assert torch.allclose(A, want, atol=1e-2), (A - want).abs().max()
+
+
+# ---------------- pointwise / reduction kernels ----------------
+@pytest.mark.parametrize("shape,H", [((4, 128), 2048), ((2, 64), 128), ((1, 7), 768)])
+def test_rmsnorm_fwd_bwd(ext, shape, H):
+    from mlx_cuda_distributed_pretraining_amd.ops import rms_norm
+
+    torch.manual_seed(0)
+    x = torch.randn(*shape, H, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(H, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    y = rms_norm(x, w, 1e-5)
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    ref = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + 1e-5) * wf
+    assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref.backward(dy.float())
+    assert torch.allclose(x.grad.float(), xf.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(w.grad.float(), wf.grad, atol=5e-1, rtol=5e-2)
+
+
+@pytest.mark.parametrize("traditional", [False, True])
+@pytest.mark.parametrize("offset", [0, 5])
+def test_rope_gpu_matches_ref(ext, traditional, offset):
+    from mlx_cuda_distributed_pretraining_amd.ops import RopeTable, apply_rope
+    from mlx_cuda_distributed_pretraining_amd.ops.rope import rope_ref
+
+    torch.manual_seed(0)
+    B, S, H, D = 2, 33, 4, 64
+    table = RopeTable(D)
+    cos, sin = table.get(S, dev(), offset)
+    x = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    y = apply_rope(x, cos, sin, traditional, offset)
+    ref = rope_ref(x.detach().float(), cos, sin, traditional, offset)
+    assert torch.allclose(y.float(), ref, atol=2e-2, rtol=2e-2)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    # backward is the inverse rotation
+    dx_ref = rope_ref(dy.float(), cos, sin, traditional, offset, conj=True)
+    assert torch.allclose(x.grad.float(), dx_ref, atol=2e-2, rtol=2e-2)
+
+
+def test_swiglu_gpu(ext):
+    from mlx_cuda_distributed_pretraining_amd.ops import swiglu
+
+    torch.manual_seed(0)
+    gu = torch.randn(8, 64, 2 * 256, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    y = swiglu(gu)
+    guf = gu.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.silu(guf[..., :256]) * guf[..., 256:]
+    assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref.backward(dy.float())
+    assert torch.allclose(gu.grad.float(), guf.grad, atol=5e-2, rtol=5e-2)
+
+
+def test_cross_entropy_gpu(ext):
+    from mlx_cuda_distributed_pretraining_amd.ops import fused_cross_entropy
+
+    torch.manual_seed(0)
+    N, V = 512, 32000
+    logits = torch.randn(N, V, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    targets = torch.randint(0, V, (N,), device=dev())
+    targets[::5] = -100
+    loss, ntok = fused_cross_entropy(logits, targets, ignore_index=-100)
+    lf = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lf, targets, ignore_index=-100)
+    assert ntok.item() == (targets != -100).sum().item()
+    assert torch.allclose(loss, ref, atol=2e-3, rtol=2e-3)
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(logits.grad.float(), lf.grad, atol=2e-4, rtol=5e-2)
+
+
+def test_sumsq_and_adamw_gpu(ext):
+    from mlx_cuda_distributed_pretraining_amd.ops import fused_optim
+
+    torch.manual_seed(0)
+    N = 100_000
+    g = torch.randn(N, device=dev(), dtype=torch.bfloat16)
+    ss = fused_optim.grad_sumsq(g)
+    assert torch.allclose(ss, g.float().pow(2).sum(), rtol=1e-3)
+
+    master = torch.randn(N, device=dev(), dtype=torch.float32)
+    param = master.bfloat16().clone()
+    m = torch.zeros(N, device=dev())
+    v = torch.zeros(N, device=dev())
+    grad = torch.randn(N, device=dev(), dtype=torch.bfloat16)
+
+    ref_p = torch.nn.Parameter(master.clone())
+    opt = torch.optim.AdamW([ref_p], lr=1e-2, betas=(0.9, 0.999), eps=1e-8, weight_decay=0.1)
+    ref_p.grad = grad.float()
+    opt.step()
+
+    fused_optim.adamw_step(param, master, grad, m, v, 1, 1e-2, 0.9, 0.999, 1e-8, 0.1,
+                           decay_boundary=N)
+    torch.cuda.synchronize()
+    assert torch.allclose(master, ref_p.detach(), atol=1e-5, rtol=1e-5)
+    assert torch.allclose(param.float(), master, atol=1e-2, rtol=1e-2)
+
+
+def test_lion_sgd_gpu(ext):
+    from mlx_cuda_distributed_pretraining_amd.ops import fused_optim
+
+    N = 4096
+    for step_fn in ("lion", "sgd"):
+        master = torch.randn(N, device=dev())
+        param = master.bfloat16().clone()
+        grad = torch.randn(N, device=dev(), dtype=torch.bfloat16)
+        buf = torch.zeros(N, device=dev())
+        if step_fn == "lion":
+            fused_optim.lion_step(param, master, grad, buf, 1, 1e-3, 0.9, 0.99, 0.1, N)
+        else:
+            fused_optim.sgd_step(param, master, grad, buf, 1, 1e-2, 0.9, 0.1, N, True)
+        torch.cuda.synchronize()
+        assert torch.isfinite(master).all()
+        assert torch.allclose(param.float(), master, atol=1e-2, rtol=1e-2)
+
+
+def test_sample_token_gpu(ext):
+    logits = torch.full((1000,), -10.0, device=dev())
+    logits[123] = 10.0
+    for seed in range(5):
+        t = ext.sample_token(logits, 0.8, 1.0, 0.0, seed)
+        assert t.item() == 123
+
+
+# ---------------- attention ----------------
+def attn_cases():
+    # (B, Sq, Skv, Hq, Hkv, D, mod_kwargs)
+    return [
+        (1, 128, 128, 2, 2, 64, {}),
+        (2, 256, 256, 4, 2, 128, {}),           # GQA
+        (1, 200, 200, 2, 2, 64, {}),            # ragged seq
+        (1, 256, 256, 2, 2, 64, {"window": 64}),
+        (1, 256, 256, 2, 2, 64, {"prefix_len": 100}),
+        (1, 128, 128, 4, 4, 64, {"alibi": True}),
+        (1, 1, 96, 2, 2, 64, {}),               # decode: Sq=1 with cache
+        (1, 128, 128, 2, 2, 64, {"causal": False}),
+    ]
+
+
+@pytest.mark.parametrize("case", attn_cases())
+def test_attn_fwd_gpu(ext, case):
+    from mlx_cuda_distributed_pretraining_amd.ops import attention_ref, flash_attention
+
+    B, Sq, Skv, Hq, Hkv, D, kw = case
+    torch.manual_seed(0)
+    q = torch.randn(B, Sq, Hq, D, device=dev(), dtype=torch.bfloat16)
+    k = torch.randn(B, Skv, Hkv, D, device=dev(), dtype=torch.bfloat16)
+    v = torch.randn(B, Skv, Hkv, D, device=dev(), dtype=torch.bfloat16)
+    causal = kw.get("causal", True)
+    alibi = None
+    if kw.get("alibi"):
+        alibi = torch.tensor([2 ** (-(i + 1)) for i in range(Hq)], device=dev())
+    o = flash_attention(q, k, v, causal=causal, window=kw.get("window"),
+                        prefix_len=kw.get("prefix_len"), alibi_slopes=alibi)
+    ref = attention_ref(q.float(), k.float(), v.float(), causal=causal,
+                        scale=1.0 / math.sqrt(D), window=kw.get("window"),
+                        prefix_len=kw.get("prefix_len"), alibi_slopes=alibi)
+    err = (o.float() - ref).abs().max().item()
+    assert err < 3e-2, f"attn fwd max err {err} for case {case}"
+
+
+def test_attn_fwd_lse_matches(ext):
+    from mlx_cuda_distributed_pretraining_amd.ops import attention_ref
+    from mlx_cuda_distributed_pretraining_amd.ops._ext import get_ext
+
+    torch.manual_seed(0)
+    B, S, H, D = 1, 128, 2, 64
+    q = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    k = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    v = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    slopes = torch.empty(0, device=dev())
+    o, lse = get_ext().attn_fwd(q, k, v, 1.0 / math.sqrt(D), 1, 0, slopes)
+    _, ref_lse = attention_ref(q.float(), k.float(), v.float(), causal=True,
+                               scale=1.0 / math.sqrt(D), return_lse=True)
+    assert torch.allclose(lse, ref_lse, atol=3e-2, rtol=1e-2)
+
+
+@pytest.mark.parametrize("case", attn_cases())
+def test_attn_bwd_gpu(ext, case):
+    from mlx_cuda_distributed_pretraining_amd.ops import attention_ref, flash_attention
+
+    B, Sq, Skv, Hq, Hkv, D, kw = case
+    if Sq != Skv:
+        pytest.skip("bwd only used in training (Sq == Skv)")
+    torch.manual_seed(0)
+    mk = lambda *s: torch.randn(*s, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    q, k, v = mk(B, Sq, Hq, D), mk(B, Skv, Hkv, D), mk(B, Skv, Hkv, D)
+    causal = kw.get("causal", True)
+    alibi = None
+    if kw.get("alibi"):
+        alibi = torch.tensor([2 ** (-(i + 1)) for i in range(Hq)], device=dev())
+    o = flash_attention(q, k, v, causal=causal, window=kw.get("window"),
+                        prefix_len=kw.get("prefix_len"), alibi_slopes=alibi)
+    dy = torch.randn_like(o)
+    o.backward(dy)
+
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    ref = attention_ref(qf, kf, vf, causal=causal, scale=1.0 / math.sqrt(D),
+                        window=kw.get("window"), prefix_len=kw.get("prefix_len"),
+                        alibi_slopes=alibi)
+    ref.backward(dy.float())
+    for name, got, want in [("dq", q.grad, qf.grad), ("dk", k.grad, kf.grad),
+                            ("dv", v.grad, vf.grad)]:
+        err = (got.float() - want).abs().max().item()
+        ref_mag = want.abs().max().item()
+        assert err < 0.05 * max(ref_mag, 1.0), f"{name} max err {err} (ref mag {ref_mag}) case {case}"

@@ -1,0 +1,84 @@
+"""End-to-end GPU training smoke: the full HIP-kernel path must train."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_flat_fused_train_steps_decrease_loss(tmp_path):
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+    from mlx_cuda_distributed_pretraining_amd.ops.cross_entropy import fused_cross_entropy
+    from mlx_cuda_distributed_pretraining_amd.optim.flat_fused import FusedFlatAdamW
+    from mlx_cuda_distributed_pretraining_amd.parallel.flat import FlatParamSpace
+
+    torch.manual_seed(0)
+    dev = torch.device("cuda:0")
+    args = ModelArgs(hidden_size=256, intermediate_size=512, num_layers=2,
+                     num_heads=4, num_kv_heads=2, head_dim=64, vocab_size=1024)
+    model = Model(args).to(dev, torch.bfloat16)
+    space = FlatParamSpace(model)
+    opt = FusedFlatAdamW(space, lr=1e-3, weight_decay=0.01, max_grad_norm=1.0)
+    x = torch.randint(0, 1024, (4, 128), device=dev)
+    losses = []
+    for _ in range(20):
+        opt.zero_grad()
+        logits = model(x[:, :-1])
+        loss, _ = fused_cross_entropy(
+            logits.reshape(-1, 1024).contiguous(), x[:, 1:].reshape(-1), -100
+        )
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0] * 0.8, f"loss did not fall: {losses[0]} -> {losses[-1]}"
+
+
+def test_trainer_on_gpu(tmp_path):
+    from pathlib import Path
+
+    from mlx_cuda_distributed_pretraining_amd.core.config import Config
+    from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer
+
+    repo = Path(__file__).resolve().parents[1]
+    cfg = Config.from_yaml(str(repo / "configs" / "model-config-124m.yaml"))
+    cfg.model.dimensions = {"hidden_size": 256, "intermediate_size": 512, "num_layers": 2}
+    cfg.model.attention = {"num_heads": 4, "num_kv_heads": 2, "head_dim": 64,
+                           "max_position_embeddings": 256}
+    cfg.data.preprocessing["max_context_size"] = 256
+    cfg.training.hyperparameters["batch_size"] = 4
+    cfg.training.hyperparameters["iters"] = 5
+    cfg.logging.steps = {"logging_interval": 1, "checkpoint_interval": 0,
+                         "validation_interval": 0}
+    trainer = Trainer(cfg, runs_root=str(tmp_path / "runs"))
+    trainer.train()
+    log = (tmp_path / "runs" / cfg.name / "log.txt").read_text()
+    assert "Step 5" in log
+
+
+def test_generation_on_gpu():
+    from mlx_cuda_distributed_pretraining_amd.core.config import DataConfig
+    from mlx_cuda_distributed_pretraining_amd.data import TokenizerManager
+    from mlx_cuda_distributed_pretraining_amd.inference.generate import generate
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+
+    tok = TokenizerManager(DataConfig())
+    args = ModelArgs(hidden_size=128, intermediate_size=256, num_layers=2,
+                     num_heads=2, head_dim=64, vocab_size=tok.vocab_size)
+    model = Model(args).to("cuda:0", torch.bfloat16)
+    text, stats = generate(model, tok, "hello", max_tokens=16, temperature=0.8)
+    assert isinstance(text, str)
+    assert stats["generated_tokens"] >= 1
+
+
+def test_native_extension_is_the_compute_path():
+    """On a GPU box the HIP extension must be loaded and used — no silent
+    eager fallback (driver records which .so files are loaded)."""
+    from mlx_cuda_distributed_pretraining_amd.ops import require_ext
+    from mlx_cuda_distributed_pretraining_amd.ops.rmsnorm import rms_norm
+
+    ext = require_ext()
+    assert ext is not None
+    x = torch.randn(4, 128, device="cuda:0", dtype=torch.bfloat16)
+    w = torch.ones(128, device="cuda:0", dtype=torch.bfloat16)
+    y = rms_norm(x, w, 1e-5)  # raises if the extension is missing
+    assert y.shape == x.shape
