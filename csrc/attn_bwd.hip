@@ -64,7 +64,8 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
     __hip_bfloat16* __restrict__ dq, const float* __restrict__ slopes,
-    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg) {
+    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
+    long q_rs, long k_rs, long v_rs, long do_rs) {
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSTR = D + 8;
@@ -90,14 +91,15 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
   // Q and dO fragments (B-operand layout: lane holds row q=lq, 8 d values)
   bf16x8 qf[DBLK], dof[DBLK];
   {
-    const long base = (((long)b * Sq + (q_valid ? qrow : 0)) * Hq + hq) * (long)D + hi * 8;
+    const long row0 = (long)b * Sq + (q_valid ? qrow : 0);
+    const long hd = (long)hq * D + hi * 8;
 #pragma unroll
     for (int dblk = 0; dblk < DBLK; ++dblk) {
       Bf16x8U uq, ud;
       *reinterpret_cast<uint4*>(uq.s) =
-          q_valid ? *reinterpret_cast<const uint4*>(q + base + dblk * 16) : uint4{0, 0, 0, 0};
+          q_valid ? *reinterpret_cast<const uint4*>(q + row0 * q_rs + hd + dblk * 16) : uint4{0, 0, 0, 0};
       *reinterpret_cast<uint4*>(ud.s) =
-          q_valid ? *reinterpret_cast<const uint4*>(dout + base + dblk * 16) : uint4{0, 0, 0, 0};
+          q_valid ? *reinterpret_cast<const uint4*>(dout + row0 * do_rs + hd + dblk * 16) : uint4{0, 0, 0, 0};
       qf[dblk] = uq.v;
       dof[dblk] = ud.v;
     }
@@ -132,12 +134,13 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
         const int row = u / U4ROW;
         const int d0 = (u % U4ROW) * 8;
         const bool valid = kv0 + row < Skv;
-        const long src = (((long)b * Skv + (valid ? kv0 + row : 0)) * Hkv + hkv) * (long)D + d0;
+        const long kr0 = (long)b * Skv + (valid ? kv0 + row : 0);
+        const long khd = (long)hkv * D + d0;
         Bf16x8U ku, vu;
         *reinterpret_cast<uint4*>(ku.s) =
-            valid ? *reinterpret_cast<const uint4*>(k + src) : uint4{0, 0, 0, 0};
+            valid ? *reinterpret_cast<const uint4*>(k + kr0 * k_rs + khd) : uint4{0, 0, 0, 0};
         *reinterpret_cast<uint4*>(vu.s) =
-            valid ? *reinterpret_cast<const uint4*>(v + src) : uint4{0, 0, 0, 0};
+            valid ? *reinterpret_cast<const uint4*>(v + kr0 * v_rs + khd) : uint4{0, 0, 0, 0};
         *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(ku.s);
         *reinterpret_cast<uint4*>(v_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(vu.s);
 #pragma unroll
@@ -211,7 +214,8 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
     __hip_bfloat16* __restrict__ dkv_out, const float* __restrict__ slopes,
-    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg) {
+    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
+    long q_rs, long k_rs, long v_rs, long do_rs) {
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSTR = D + 8;
@@ -237,17 +241,18 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
   // K (and V for dK) fragments: lane holds row k=lk, 8 d values
   bf16x8 kf[DBLK], vf[WANT_DK ? DBLK : 1];
   {
-    const long base = (((long)b * Skv + (k_valid ? krow : 0)) * Hkv + hkv) * (long)D + hi * 8;
+    const long kr0 = (long)b * Skv + (k_valid ? krow : 0);
+    const long khd = (long)hkv * D + hi * 8;
 #pragma unroll
     for (int dblk = 0; dblk < DBLK; ++dblk) {
       Bf16x8U ku;
       *reinterpret_cast<uint4*>(ku.s) =
-          k_valid ? *reinterpret_cast<const uint4*>(k + base + dblk * 16) : uint4{0, 0, 0, 0};
+          k_valid ? *reinterpret_cast<const uint4*>(k + kr0 * k_rs + khd + dblk * 16) : uint4{0, 0, 0, 0};
       kf[dblk] = ku.v;
       if constexpr (WANT_DK) {
         Bf16x8U vu;
         *reinterpret_cast<uint4*>(vu.s) =
-            k_valid ? *reinterpret_cast<const uint4*>(v + base + dblk * 16) : uint4{0, 0, 0, 0};
+            k_valid ? *reinterpret_cast<const uint4*>(v + kr0 * v_rs + khd + dblk * 16) : uint4{0, 0, 0, 0};
         vf[dblk] = vu.v;
       }
     }
@@ -273,12 +278,13 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
           const int row = u / U4ROW;
           const int d0 = (u % U4ROW) * 8;
           const bool valid = q0 + row < Sq;
-          const long src = (((long)b * Sq + (valid ? q0 + row : 0)) * Hq + hq) * (long)D + d0;
+          const long qr0 = (long)b * Sq + (valid ? q0 + row : 0);
+          const long qhd = (long)hq * D + d0;
           Bf16x8U qu, du;
           *reinterpret_cast<uint4*>(qu.s) =
-              valid ? *reinterpret_cast<const uint4*>(q + src) : uint4{0, 0, 0, 0};
+              valid ? *reinterpret_cast<const uint4*>(q + qr0 * q_rs + qhd) : uint4{0, 0, 0, 0};
           *reinterpret_cast<uint4*>(du.s) =
-              valid ? *reinterpret_cast<const uint4*>(dout + src) : uint4{0, 0, 0, 0};
+              valid ? *reinterpret_cast<const uint4*>(dout + qr0 * do_rs + qhd) : uint4{0, 0, 0, 0};
           *reinterpret_cast<uint4*>(q_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(qu.s);
           if constexpr (WANT_DK) {
             *reinterpret_cast<uint4*>(do_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(du.s);
@@ -374,13 +380,13 @@ void launch_bwd_all(dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
                     const __hip_bfloat16* dout, const float* lse, const float* drow,
                     __hip_bfloat16* dq, __hip_bfloat16* dk, __hip_bfloat16* dv,
                     const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
-                    float scale, int modarg) {
+                    float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs) {
   attn_bwd_dq_kernel<D, MOD><<<gq, block, 0, stream>>>(
-      q, k, v, dout, lse, drow, dq, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      q, k, v, dout, lse, drow, dq, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
   attn_bwd_dkv_kernel<D, MOD, true><<<gkv, block, 0, stream>>>(
-      q, k, v, dout, lse, drow, dk, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      q, k, v, dout, lse, drow, dk, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
   attn_bwd_dkv_kernel<D, MOD, false><<<gkv, block, 0, stream>>>(
-      q, k, v, dout, lse, drow, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      q, k, v, dout, lse, drow, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
 }
 
 template <int D>
@@ -389,22 +395,22 @@ void launch_bwd_mod(int mod, dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
                     const __hip_bfloat16* dout, const float* lse, const float* drow,
                     __hip_bfloat16* dq, __hip_bfloat16* dk, __hip_bfloat16* dv,
                     const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
-                    float scale, int modarg) {
+                    float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs) {
   switch (mod) {
     case MOD_NONE:
-      launch_bwd_all<D, MOD_NONE>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      launch_bwd_all<D, MOD_NONE>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
       break;
     case MOD_CAUSAL:
-      launch_bwd_all<D, MOD_CAUSAL>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      launch_bwd_all<D, MOD_CAUSAL>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
       break;
     case MOD_SLIDING_WINDOW:
-      launch_bwd_all<D, MOD_SLIDING_WINDOW>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      launch_bwd_all<D, MOD_SLIDING_WINDOW>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
       break;
     case MOD_PREFIX_LM:
-      launch_bwd_all<D, MOD_PREFIX_LM>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      launch_bwd_all<D, MOD_PREFIX_LM>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
       break;
     case MOD_ALIBI:
-      launch_bwd_all<D, MOD_ALIBI>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      launch_bwd_all<D, MOD_ALIBI>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
       break;
     default:
       TORCH_CHECK(false, "attn_bwd: unknown mod ", mod);
@@ -416,14 +422,23 @@ void launch_bwd_mod(int mod, dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
 std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
                                  at::Tensor dout, at::Tensor lse, double scale, long mod,
                                  long modarg, at::Tensor slopes) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && dout.is_contiguous());
+  TORCH_CHECK(q.is_cuda());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_bwd: bf16 only");
+  const int Dq_ = q.size(3);
+  auto rs = [Dq_](at::Tensor& t) {
+    const int S = t.size(1), H = t.size(2), D = t.size(3);
+    if (!(t.stride(3) == 1 && t.stride(2) == D && t.stride(0) == (long)S * t.stride(1)))
+      t = t.contiguous();
+    return t.stride(1);
+  };
+  const long q_rs = rs(q), k_rs = rs(k), v_rs = rs(v), do_rs = rs(dout);
+  o = o.contiguous();
   const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Skv = k.size(1), Hkv = k.size(2);
   TORCH_CHECK(D == 64 || D == 128, "attn_bwd: head_dim must be 64 or 128");
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
+  auto dq = at::empty({B, Sq, Hq, D}, q.options());
+  auto dk = at::empty({B, Skv, Hkv, D}, k.options());
+  auto dv = at::empty({B, Skv, Hkv, D}, v.options());
   auto drow = at::empty({(long)B * Sq * Hq}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
 
@@ -450,10 +465,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::T
   if (D == 64)
     launch_bwd_mod<64>((int)mod, gq, gkv, block, stream, qp, kp, vp, dop,
                        lse.data_ptr<float>(), drow.data_ptr<float>(), dqp, dkp, dvp, sl,
-                       B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg);
+                       B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs, do_rs);
   else
     launch_bwd_mod<128>((int)mod, gq, gkv, block, stream, qp, kp, vp, dop,
                         lse.data_ptr<float>(), drow.data_ptr<float>(), dqp, dkp, dvp, sl,
-                        B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg);
+                        B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs, do_rs);
   return {dq, dk, dv};
 }

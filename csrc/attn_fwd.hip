@@ -25,7 +25,8 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, const float* __restrict__ slopes,
-    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg) {
+    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
+    long q_rs, long k_rs, long v_rs) {
   constexpr int DBLK = D / 16;   // QK^T k-slots (over d)
   constexpr int DCOL = D / 32;   // PV output column tiles
   constexpr int KPAD = 8;        // elements of row padding (16B) against bank conflicts
@@ -56,7 +57,7 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
   bf16x8 qf[DBLK];
   {
     const __hip_bfloat16* qp =
-        q + (((long)b * Sq + (q_valid ? qrow : 0)) * Hq + hq) * D + hi * 8;
+        q + ((long)b * Sq + (q_valid ? qrow : 0)) * q_rs + (long)hq * D + hi * 8;
 #pragma unroll
     for (int dblk = 0; dblk < DBLK; ++dblk) {
       Bf16x8U u;
@@ -96,14 +97,15 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
         const int row = u / U4ROW;
         const int d0 = (u % U4ROW) * 8;
         const bool valid = kv0 + row < Skv;
-        const long src = (((long)b * Skv + (valid ? kv0 + row : 0)) * Hkv + hkv) * D + d0;
+        const long r0 = (long)b * Skv + (valid ? kv0 + row : 0);
+        const long hd = (long)hkv * D + d0;
         Bf16x8U kv_u;
         *reinterpret_cast<uint4*>(kv_u.s) =
-            valid ? *reinterpret_cast<const uint4*>(k + src) : uint4{0, 0, 0, 0};
+            valid ? *reinterpret_cast<const uint4*>(k + r0 * k_rs + hd) : uint4{0, 0, 0, 0};
         *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(kv_u.s);
         Bf16x8U vv;
         *reinterpret_cast<uint4*>(vv.s) =
-            valid ? *reinterpret_cast<const uint4*>(v + src) : uint4{0, 0, 0, 0};
+            valid ? *reinterpret_cast<const uint4*>(v + r0 * v_rs + hd) : uint4{0, 0, 0, 0};
 #pragma unroll
         for (int j = 0; j < 8; ++j) vt_lds[(d0 + j) * VSTR + row] = vv.h[j];
       }
@@ -201,22 +203,23 @@ template <int D>
 void launch_fwd(int mod, dim3 grid, dim3 block, hipStream_t stream,
                 const __hip_bfloat16* q, const __hip_bfloat16* k, const __hip_bfloat16* v,
                 __hip_bfloat16* o, float* lse, const float* slopes,
-                int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg) {
+                int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
+                long q_rs, long k_rs, long v_rs) {
   switch (mod) {
     case MOD_NONE:
-      attn_fwd_kernel<D, MOD_NONE><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      attn_fwd_kernel<D, MOD_NONE><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_CAUSAL:
-      attn_fwd_kernel<D, MOD_CAUSAL><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      attn_fwd_kernel<D, MOD_CAUSAL><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_SLIDING_WINDOW:
-      attn_fwd_kernel<D, MOD_SLIDING_WINDOW><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      attn_fwd_kernel<D, MOD_SLIDING_WINDOW><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_PREFIX_LM:
-      attn_fwd_kernel<D, MOD_PREFIX_LM><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      attn_fwd_kernel<D, MOD_PREFIX_LM><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_ALIBI:
-      attn_fwd_kernel<D, MOD_ALIBI><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg);
+      attn_fwd_kernel<D, MOD_ALIBI><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     default:
       TORCH_CHECK(false, "attn_fwd: unknown mod ", mod);
@@ -225,15 +228,25 @@ void launch_fwd(int mod, dim3 grid, dim3 block, hipStream_t stream,
 
 }  // namespace
 
+static long bshd_row_stride(at::Tensor& t) {
+  // accept [B,S,H,D] views whose (h,d) inner block is contiguous (e.g. slices
+  // of the fused QKV projection); otherwise materialize.
+  const int S = t.size(1), H = t.size(2), D = t.size(3);
+  if (!(t.stride(3) == 1 && t.stride(2) == D && t.stride(0) == (long)S * t.stride(1)))
+    t = t.contiguous();
+  return t.stride(1);
+}
+
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale,
                                  long mod, long modarg, at::Tensor slopes) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.is_cuda());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_fwd: bf16 only");
+  const long q_rs = bshd_row_stride(q), k_rs = bshd_row_stride(k), v_rs = bshd_row_stride(v);
   const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Skv = k.size(1), Hkv = k.size(2);
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
   TORCH_CHECK(D == 64 || D == 128, "attn_fwd: head_dim must be 64 or 128, got ", D);
-  auto o = at::empty_like(q);
+  auto o = at::empty({B, Sq, Hq, D}, q.options());
   auto lse = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(cdiv(Sq, QPB), Hq, B);
@@ -244,8 +257,8 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, doubl
   auto* vp = reinterpret_cast<const __hip_bfloat16*>(v.data_ptr());
   auto* op = reinterpret_cast<__hip_bfloat16*>(o.data_ptr());
   if (D == 64)
-    launch_fwd<64>((int)mod, grid, block, stream, qp, kp, vp, op, lse.data_ptr<float>(), sl, B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg);
+    launch_fwd<64>((int)mod, grid, block, stream, qp, kp, vp, op, lse.data_ptr<float>(), sl, B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs);
   else
-    launch_fwd<128>((int)mod, grid, block, stream, qp, kp, vp, op, lse.data_ptr<float>(), sl, B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg);
+    launch_fwd<128>((int)mod, grid, block, stream, qp, kp, vp, op, lse.data_ptr<float>(), sl, B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs);
   return {o, lse};
 }

@@ -80,15 +80,52 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict_
     const float r = rstd[row];
 
     float c = 0.f;
-    for (int i = threadIdx.x; i < H; i += BLOCK)
-      c += to_f32(dyr[i]) * to_f32(w[i]) * to_f32(xr[i]);
-    c = block_reduce_sum<BLOCK>(c, scratch);
-    const float k = r * r * r * c / H;
-    for (int i = threadIdx.x; i < H; i += BLOCK) {
-      const float xi = to_f32(xr[i]);
-      const float dyi = to_f32(dyr[i]);
-      from_f32(&dxr[i], r * dyi * to_f32(w[i]) - xi * k);
-      dw_acc[i] += dyi * xi * r;
+    if constexpr (sizeof(T) == 2) {
+      const int HV = H / 8;
+      const uint4* xv = reinterpret_cast<const uint4*>(xr);
+      const uint4* dyv = reinterpret_cast<const uint4*>(dyr);
+      const uint4* wv = reinterpret_cast<const uint4*>(w);
+      for (int i = threadIdx.x; i < HV; i += BLOCK) {
+        U4 xu, du, wu;
+        xu.u = xv[i]; du.u = dyv[i]; wu.u = wv[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          c += bf16_bits_to_f32(du.s[j]) * bf16_bits_to_f32(wu.s[j]) * bf16_bits_to_f32(xu.s[j]);
+      }
+      for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK)
+        c += to_f32(dyr[i]) * to_f32(w[i]) * to_f32(xr[i]);
+      c = block_reduce_sum<BLOCK>(c, scratch);
+      const float kk = r * r * r * c / H;
+      uint4* dxv = reinterpret_cast<uint4*>(dxr);
+      for (int i = threadIdx.x; i < HV; i += BLOCK) {
+        U4 xu, du, wu, ou;
+        xu.u = xv[i]; du.u = dyv[i]; wu.u = wv[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float xi = bf16_bits_to_f32(xu.s[j]);
+          const float dyi = bf16_bits_to_f32(du.s[j]);
+          ou.s[j] = f32_to_bf16_bits(r * dyi * bf16_bits_to_f32(wu.s[j]) - xi * kk);
+          dw_acc[i * 8 + j] += dyi * xi * r;
+        }
+        dxv[i] = ou.u;
+      }
+      for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK) {
+        const float xi = to_f32(xr[i]);
+        const float dyi = to_f32(dyr[i]);
+        from_f32(&dxr[i], r * dyi * to_f32(w[i]) - xi * kk);
+        dw_acc[i] += dyi * xi * r;
+      }
+    } else {
+      for (int i = threadIdx.x; i < H; i += BLOCK)
+        c += to_f32(dyr[i]) * to_f32(w[i]) * to_f32(xr[i]);
+      c = block_reduce_sum<BLOCK>(c, scratch);
+      const float kk = r * r * r * c / H;
+      for (int i = threadIdx.x; i < H; i += BLOCK) {
+        const float xi = to_f32(xr[i]);
+        const float dyi = to_f32(dyr[i]);
+        from_f32(&dxr[i], r * dyi * to_f32(w[i]) - xi * kk);
+        dw_acc[i] += dyi * xi * r;
+      }
     }
     __syncthreads();
   }

@@ -51,15 +51,33 @@ __global__ void swiglu_bwd_kernel(const T* __restrict__ gu, const T* __restrict_
     const T* d = dy + row * (long)I + col;
     T* dg = dgu + row * (long)(2 * I) + col;
     T* du = dg + I;
+    if constexpr (sizeof(T) == 2) {
+      U4 gv, uv, dv, dgv, duv;
+      gv.u = *reinterpret_cast<const uint4*>(g);
+      uv.u = *reinterpret_cast<const uint4*>(u);
+      dv.u = *reinterpret_cast<const uint4*>(d);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const float gf = to_f32(g[j]);
-      const float uf = to_f32(u[j]);
-      const float df = to_f32(d[j]);
-      const float sg = 1.f / (1.f + __expf(-gf));
-      const float silu = gf * sg;
-      from_f32(&dg[j], df * uf * (sg * (1.f + gf * (1.f - sg))));
-      from_f32(&du[j], df * silu);
+      for (int j = 0; j < 8; ++j) {
+        const float gf = bf16_bits_to_f32(gv.s[j]);
+        const float uf = bf16_bits_to_f32(uv.s[j]);
+        const float df = bf16_bits_to_f32(dv.s[j]);
+        const float sg = 1.f / (1.f + __expf(-gf));
+        dgv.s[j] = f32_to_bf16_bits(df * uf * (sg * (1.f + gf * (1.f - sg))));
+        duv.s[j] = f32_to_bf16_bits(df * gf * sg);
+      }
+      *reinterpret_cast<uint4*>(dg) = dgv.u;
+      *reinterpret_cast<uint4*>(du) = duv.u;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float gf = to_f32(g[j]);
+        const float uf = to_f32(u[j]);
+        const float df = to_f32(d[j]);
+        const float sg = 1.f / (1.f + __expf(-gf));
+        const float silu = gf * sg;
+        from_f32(&dg[j], df * uf * (sg * (1.f + gf * (1.f - sg))));
+        from_f32(&du[j], df * silu);
+      }
     }
   }
 }

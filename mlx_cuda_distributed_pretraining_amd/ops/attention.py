@@ -126,9 +126,7 @@ class _FlashAttnFn(torch.autograd.Function):
                 if alibi_slopes is not None
                 else torch.empty(0, dtype=torch.float32, device=q.device)
             )
-            o, lse = ext.attn_fwd(
-                q.contiguous(), k.contiguous(), v.contiguous(), scale, mod, modarg, slopes
-            )
+            o, lse = ext.attn_fwd(q, k, v, scale, mod, modarg, slopes)
             ctx.save_for_backward(q, k, v, o, lse, slopes)
             ctx.meta = (causal, scale, window, prefix_len, True)
             return o
@@ -155,10 +153,7 @@ class _FlashAttnFn(torch.autograd.Function):
             modarg = int(
                 window if window is not None else (prefix_len if prefix_len is not None else 0)
             )
-            dq, dk, dv = ext.attn_bwd(
-                q.contiguous(), k.contiguous(), v.contiguous(),
-                o.contiguous(), do.contiguous(), lse, scale, mod, modarg, slopes,
-            )
+            dq, dk, dv = ext.attn_bwd(q, k, v, o, do, lse, scale, mod, modarg, slopes)
             return dq, dk, dv, None, None, None, None, None
         # CPU reference backward: recompute with autograd in fp32
         with torch.enable_grad():

@@ -83,7 +83,7 @@ class _RopeFn(torch.autograd.Function):
         ctx.save_for_backward(cos, sin)
         if use_hip(x):
             ext = get_ext()
-            return ext.rope_fwd(x.contiguous(), cos, sin, traditional, offset, False)
+            return ext.rope_fwd(x, cos, sin, traditional, offset, False)
         return rope_ref(x, cos, sin, traditional, offset, conj=False)
 
     @staticmethod
