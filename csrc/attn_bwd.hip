@@ -9,35 +9,40 @@
 //   dV kernel : block per kv-tile; accumulates dV = P^T @ dO in registers
 // Splitting dK/dV keeps each kernel under the 256-VGPR budget at D=128
 // (merged, the two 32x128 fp32 accumulators alone are 128 VGPRs).
-// All use the same MFMA layout + acc_to_afrag transform as the forward.
+//
+// Same gfx950 structure as the forward (attn_fwd.hip): 8 waves/block sharing
+// every staged tile, T14 register-prefetch of the next tile under the MFMA
+// clusters, pair-row b32 writes for the transposed LDS images, s_setprio
+// around MFMAs, MFMA layout + acc_to_afrag transform from attn_common.h.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "attn_common.h"
 
 namespace {
 
-constexpr int KVB = 32;
-constexpr int NW = 4;
+constexpr int NW = 8;
 constexpr int TPB = NW * WAVE;
 constexpr int QPB = 32 * NW;   // q rows per block (dQ kernel)
 constexpr int KPB = 32 * NW;   // kv rows per block (dK/dV kernels)
+constexpr int KVB = 32;        // kv tile per iteration (dQ kernel)
 
 // ---------------- preprocess: Drow = rowsum(dO * O) ----------------
 __global__ void bwd_preprocess_kernel(const __hip_bfloat16* __restrict__ dout,
                                       const __hip_bfloat16* __restrict__ o,
                                       float* __restrict__ drow,
-                                      long rows, int Hq, int D) {
-  // one wave per (b,s,h) row; rows-major layout is BSHD so row index = (b*S+s)*Hq+h
-  const long row = blockIdx.x * (long)(TPB / WAVE) + threadIdx.x / WAVE;
+                                      long rows, int D, long do_rs, int Hq) {
+  // one wave per (b,s,h) row; row index = (b*S+s)*Hq+h; o is contiguous BSHD,
+  // dout may be row-strided.
+  const long row = blockIdx.x * (long)(256 / WAVE) + threadIdx.x / WAVE;
   if (row >= rows) return;
   const int lane = threadIdx.x % WAVE;
-  const __hip_bfloat16* dp = dout + row * D;
+  const long bs = row / Hq;
+  const int h = (int)(row % Hq);
+  const __hip_bfloat16* dp = dout + bs * do_rs + (long)h * D;
   const __hip_bfloat16* op = o + row * D;
   float acc = 0.f;
   for (int i = lane; i < D; i += WAVE) acc += to_f32(dp[i]) * to_f32(op[i]);
   acc = wave_reduce_sum(acc);
-  // drow stored in BSHD-row order: index (b*S+s)*Hq+h, matching how the
-  // dq/dkv kernels read it.
   if (lane == 0) drow[row] = acc;
 }
 
@@ -57,7 +62,7 @@ __device__ __forceinline__ void q_range_for_kv(int kv0, int kpb, int q_off, int 
   }
 }
 
-// ---------------- dQ kernel ----------------
+// ---------------- dQ kernel (block per q-tile of QPB rows) ----------------
 template <int D, int MOD>
 __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
@@ -120,37 +125,68 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     kv_hi = min(Skv, max(blk_qpos_hi + 1, modarg));
   }
 
+  // T14 staging: chunks of (2 kv rows x 8 d). K chunks write row-major twice
+  // (plain + pair-transposed); V chunks write row-major.
+  constexpr int CH_TOT = (KVB / 2) * (D / 8);  // chunks per tile
+  constexpr int NCH = (CH_TOT + TPB - 1) / TPB;
+  uint4 kreg[NCH][2], vreg[NCH][2];
+
+  auto stage_load = [&](int kv0) {
+#pragma unroll
+    for (int c = 0; c < NCH; ++c) {
+      const int u = tid + c * TPB;
+      const int row = (u / (D / 8)) * 2;
+      const int d0 = (u % (D / 8)) * 8;
+      if (row >= KVB) continue;
+      const bool ok0 = kv0 + row < Skv;
+      const bool ok1 = kv0 + row + 1 < Skv;
+      const long kb = ((long)b * Skv + kv0 + row) * k_rs + (long)hkv * D + d0;
+      const long vb = ((long)b * Skv + kv0 + row) * v_rs + (long)hkv * D + d0;
+      kreg[c][0] = ok0 ? *reinterpret_cast<const uint4*>(k + kb) : uint4{0, 0, 0, 0};
+      kreg[c][1] = ok1 ? *reinterpret_cast<const uint4*>(k + kb + k_rs) : uint4{0, 0, 0, 0};
+      vreg[c][0] = ok0 ? *reinterpret_cast<const uint4*>(v + vb) : uint4{0, 0, 0, 0};
+      vreg[c][1] = ok1 ? *reinterpret_cast<const uint4*>(v + vb + v_rs) : uint4{0, 0, 0, 0};
+    }
+  };
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int c = 0; c < NCH; ++c) {
+      const int u = tid + c * TPB;
+      const int row = (u / (D / 8)) * 2;
+      const int d0 = (u % (D / 8)) * 8;
+      if (row >= KVB) continue;
+      *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = kreg[c][0];
+      *reinterpret_cast<uint4*>(k_lds + (row + 1) * KSTR + d0) = kreg[c][1];
+      *reinterpret_cast<uint4*>(v_lds + row * KSTR + d0) = vreg[c][0];
+      *reinterpret_cast<uint4*>(v_lds + (row + 1) * KSTR + d0) = vreg[c][1];
+      Bf16x8U k0, k1;
+      *reinterpret_cast<uint4*>(k0.s) = kreg[c][0];
+      *reinterpret_cast<uint4*>(k1.s) = kreg[c][1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const uint pair = (uint)k0.s[j] | ((uint)k1.s[j] << 16);
+        *reinterpret_cast<uint*>(kt_lds + (d0 + j) * VSTR + row) = pair;
+      }
+    }
+  };
+
   float dq_acc[DCOL][16];
 #pragma unroll
   for (int dc = 0; dc < DCOL; ++dc)
 #pragma unroll
     for (int r = 0; r < 16; ++r) dq_acc[dc][r] = 0.f;
 
+  stage_load(kv_lo);
+  stage_write();
+  __syncthreads();
+
   for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
-    {  // stage K (row-major + transposed) and V (row-major)
-      constexpr int U4ROW = D / 8;
-      constexpr int TOT = KVB * U4ROW;
-      for (int u = tid; u < TOT; u += TPB) {
-        const int row = u / U4ROW;
-        const int d0 = (u % U4ROW) * 8;
-        const bool valid = kv0 + row < Skv;
-        const long kr0 = (long)b * Skv + (valid ? kv0 + row : 0);
-        const long khd = (long)hkv * D + d0;
-        Bf16x8U ku, vu;
-        *reinterpret_cast<uint4*>(ku.s) =
-            valid ? *reinterpret_cast<const uint4*>(k + kr0 * k_rs + khd) : uint4{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(vu.s) =
-            valid ? *reinterpret_cast<const uint4*>(v + kr0 * v_rs + khd) : uint4{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(ku.s);
-        *reinterpret_cast<uint4*>(v_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(vu.s);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) kt_lds[(d0 + j) * VSTR + row] = ku.h[j];
-      }
-    }
-    __syncthreads();
+    const bool has_next = kv0 + KVB < kv_hi;
+    if (has_next) stage_load(kv0 + KVB);
 
     // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
     f32x16 st = {}, dpt = {};
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dblk = 0; dblk < DBLK; ++dblk) {
       Bf16x8U kf, vf;
@@ -161,6 +197,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
       dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf.v, dof[dblk], dpt, 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     float ds[16];
 #pragma unroll
@@ -175,6 +212,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
 
     bf16x8 da0, da1;
     acc_to_afrag(ds, da0, da1);  // -> dS[32q x 16k] A-fragments
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dc = 0; dc < DCOL; ++dc) {
       f32x16 acc;
@@ -190,6 +228,10 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
     }
+    __builtin_amdgcn_s_setprio(0);
+
+    __syncthreads();
+    if (has_next) stage_write();
     __syncthreads();
   }
 
@@ -207,7 +249,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
 }
 
 // ---------------- dK / dV kernels (block per kv-tile, loop GQA group) -------
-// TEMPLATE WANT_DK: true -> compute dK (needs dP, V frags); false -> dV.
+// WANT_DK: true -> dK (needs dP: V frags + dO rm); false -> dV (needs dO^T).
 template <int D, int MOD, bool WANT_DK>
 __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
@@ -221,19 +263,18 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
   constexpr int KSTR = D + 8;
   constexpr int VSTR = 32 + 8;
 
-  // q-tile images: row-major Q [32][KSTR]; transposed X^T [D][VSTR] where
-  // X = Q (dK) or dO (dV); plus dO row-major for dP (dK only); lse/drow [32].
+  // q-tile images: Q row-major; X^T (X = Q for dK, dO for dV); dO rm (dK only)
   __shared__ __hip_bfloat16 smem[32 * KSTR + D * VSTR + 32 * KSTR];
   __shared__ float stats_lds[2][32];
-  __hip_bfloat16* q_lds = smem;                 // Q row-major (A-frags)
-  __hip_bfloat16* xt_lds = smem + 32 * KSTR;    // Q^T (dK) / dO^T (dV)
-  __hip_bfloat16* do_lds = xt_lds + D * VSTR;   // dO row-major (dK only)
+  __hip_bfloat16* q_lds = smem;
+  __hip_bfloat16* xt_lds = smem + 32 * KSTR;
+  __hip_bfloat16* do_lds = xt_lds + D * VSTR;
 
   const int b = blockIdx.z, hkv = blockIdx.y, kvtile = blockIdx.x;
   const int group = Hq / Hkv;
   const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
   const int lk = lane & 31, hi = lane >> 5;
-  const int kv0w = kvtile * KPB + wave * 32;  // this wave's kv rows
+  const int kv0w = kvtile * KPB + wave * 32;
   const int krow = kv0w + lk;
   const bool k_valid = krow < Skv;
   const int q_off = Skv - Sq;
@@ -267,47 +308,86 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
   int q_lo, q_hi;
   q_range_for_kv<MOD>(kvtile * KPB, KPB, q_off, Sq, modarg, q_lo, q_hi);
 
-  for (int g = 0; g < group; ++g) {
-    const int hq = hkv * group + g;
-    const float slope = (MOD == MOD_ALIBI) ? slopes[hq] : 0.f;
-    for (int q0 = q_lo; q0 < q_hi; q0 += 32) {
-      {  // stage q-tile: Q rm (+ dO rm for dK), X^T, lse/drow
-        constexpr int U4ROW = D / 8;
-        constexpr int TOT = 32 * U4ROW;
-        for (int u = tid; u < TOT; u += TPB) {
-          const int row = u / U4ROW;
-          const int d0 = (u % U4ROW) * 8;
-          const bool valid = q0 + row < Sq;
-          const long qr0 = (long)b * Sq + (valid ? q0 + row : 0);
-          const long qhd = (long)hq * D + d0;
-          Bf16x8U qu, du;
-          *reinterpret_cast<uint4*>(qu.s) =
-              valid ? *reinterpret_cast<const uint4*>(q + qr0 * q_rs + qhd) : uint4{0, 0, 0, 0};
-          *reinterpret_cast<uint4*>(du.s) =
-              valid ? *reinterpret_cast<const uint4*>(dout + qr0 * do_rs + qhd) : uint4{0, 0, 0, 0};
-          *reinterpret_cast<uint4*>(q_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(qu.s);
-          if constexpr (WANT_DK) {
-            *reinterpret_cast<uint4*>(do_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(du.s);
-#pragma unroll
-            for (int j = 0; j < 8; ++j) xt_lds[(d0 + j) * VSTR + row] = qu.h[j];
-          } else {
-#pragma unroll
-            for (int j = 0; j < 8; ++j) xt_lds[(d0 + j) * VSTR + row] = du.h[j];
-          }
-        }
-        if (tid < 32) {
-          const int qr = q0 + tid;
-          stats_lds[0][tid] = (qr < Sq) ? lse[((long)b * Hq + hq) * Sq + qr] : INFINITY;
-          stats_lds[1][tid] = (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + hq] : 0.f;
-        }
-      }
-      __syncthreads();
+  // T14 staging of one 32-row q tile: threads 0..255 own Q chunks
+  // (2 rows x 8 d), threads 256..511 own dO chunks.
+  constexpr int CH_TOT = (32 / 2) * (D / 8);  // 256 chunks per tensor
+  const bool is_q_half = tid < CH_TOT;
+  uint4 sreg[2];
 
-      // S = mfma(Q, K^T): A=Q rm frags from LDS, B=K^T = register kf
+  auto stage_load = [&](int hq_, int q0) {
+    const int u = is_q_half ? tid : tid - CH_TOT;
+    if (u >= CH_TOT) return;
+    const int row = (u / (D / 8)) * 2;
+    const int d0 = (u % (D / 8)) * 8;
+    const bool ok0 = q0 + row < Sq;
+    const bool ok1 = q0 + row + 1 < Sq;
+    const __hip_bfloat16* src = is_q_half ? q : dout;
+    const long rs_ = is_q_half ? q_rs : do_rs;
+    const long base = ((long)b * Sq + q0 + row) * rs_ + (long)hq_ * D + d0;
+    sreg[0] = ok0 ? *reinterpret_cast<const uint4*>(src + base) : uint4{0, 0, 0, 0};
+    sreg[1] = ok1 ? *reinterpret_cast<const uint4*>(src + base + rs_) : uint4{0, 0, 0, 0};
+  };
+  auto stage_write = [&]() {
+    const int u = is_q_half ? tid : tid - CH_TOT;
+    if (u >= CH_TOT) return;
+    const int row = (u / (D / 8)) * 2;
+    const int d0 = (u % (D / 8)) * 8;
+    // row-major images: Q always; dO only for dK
+    if (is_q_half) {
+      *reinterpret_cast<uint4*>(q_lds + row * KSTR + d0) = sreg[0];
+      *reinterpret_cast<uint4*>(q_lds + (row + 1) * KSTR + d0) = sreg[1];
+    } else if constexpr (WANT_DK) {
+      *reinterpret_cast<uint4*>(do_lds + row * KSTR + d0) = sreg[0];
+      *reinterpret_cast<uint4*>(do_lds + (row + 1) * KSTR + d0) = sreg[1];
+    }
+    // transposed image: Q^T for dK, dO^T for dV
+    if (is_q_half == WANT_DK) {
+      Bf16x8U x0, x1;
+      *reinterpret_cast<uint4*>(x0.s) = sreg[0];
+      *reinterpret_cast<uint4*>(x1.s) = sreg[1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const uint pair = (uint)x0.s[j] | ((uint)x1.s[j] << 16);
+        *reinterpret_cast<uint*>(xt_lds + (d0 + j) * VSTR + row) = pair;
+      }
+    }
+  };
+
+  // iterate (GQA head, q tile) pairs with a flat prefetch pipeline
+  const int ntiles = (q_hi - q_lo + 31) / 32;
+  const int total_iters = group * ntiles;
+  if (total_iters == 0) goto store;
+
+  {
+    auto iter_to = [&](int it, int& hq_, int& q0) {
+      hq_ = hkv * group + it / ntiles;
+      q0 = q_lo + (it % ntiles) * 32;
+    };
+    int hq_, q0;
+    iter_to(0, hq_, q0);
+    stage_load(hq_, q0);
+    stage_write();
+    if (tid < 32) {
+      const int qr = q0 + tid;
+      stats_lds[0][tid] = (qr < Sq) ? lse[((long)b * Hq + hq_) * Sq + qr] : INFINITY;
+      stats_lds[1][tid] = (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + hq_] : 0.f;
+    }
+    __syncthreads();
+
+    for (int it = 0; it < total_iters; ++it) {
+      iter_to(it, hq_, q0);
+      const float slope = (MOD == MOD_ALIBI) ? slopes[hq_] : 0.f;
+      const bool has_next = it + 1 < total_iters;
+      int nhq, nq0;
+      if (has_next) {
+        iter_to(it + 1, nhq, nq0);
+        stage_load(nhq, nq0);
+      }
+
+      // S = mfma(Q, K^T): A=Q rm frags from LDS, B = register kf.
       // element (r=q_local, c=k_local)
       f32x16 s_acc = {}, dp_acc = {};
-      // A-operand row index is lane&31 (= q_local for Q/dO); B-operand column
-      // index is also lane&31 (= k_local for the register K^T/V^T fragments).
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int dblk = 0; dblk < DBLK; ++dblk) {
         Bf16x8U qa;
@@ -321,6 +401,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
           dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da.v, vf[dblk], dp_acc, 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 
       float pv[16];
 #pragma unroll
@@ -339,9 +420,10 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
         }
       }
 
-      // transform: acc holds M[r=q][c=k]; A-frags of M^T = P^T (dV) or dS^T (dK)
+      // transform: acc holds M[r=q][c=k]; A-frags of M^T = dS^T (dK) / P^T (dV)
       bf16x8 a0, a1;
       acc_to_afrag(pv, a0, a1);
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int dc = 0; dc < DCOL; ++dc) {
         f32x16 acc;
@@ -349,7 +431,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
         for (int r = 0; r < 16; ++r) acc[r] = acc_out[dc][r];
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-          Bf16x8U xb;  // B = X[16q x 32d] from X^T image (X = Q for dK, dO for dV)
+          Bf16x8U xb;  // B = X[16q x 32d] from X^T image
           *reinterpret_cast<uint4*>(xb.s) = *reinterpret_cast<const uint4*>(
               xt_lds + (dc * 32 + lk) * VSTR + ks * 16 + hi * 8);
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb.v, acc, 0, 0, 0);
@@ -357,20 +439,34 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) acc_out[dc][r] = acc[r];
       }
+      __builtin_amdgcn_s_setprio(0);
+
+      __syncthreads();
+      if (has_next) {
+        stage_write();
+        if (tid < 32) {
+          const int qr = nq0 + tid;
+          stats_lds[0][tid] = (qr < Sq) ? lse[((long)b * Hq + nhq) * Sq + qr] : INFINITY;
+          stats_lds[1][tid] = (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + nhq] : 0.f;
+        }
+      }
       __syncthreads();
     }
   }
 
+store:
   // store: element (r=k_local, c=d_local)
-  const int dl = lane & 31;
+  {
+    const int dl = lane & 31;
 #pragma unroll
-  for (int reg = 0; reg < 16; ++reg) {
-    const int r = acc_row(reg, hi);
-    const int k_r = kv0w + r;
-    if (k_r >= Skv) continue;
-    __hip_bfloat16* out = dkv_out + (((long)b * Skv + k_r) * Hkv + hkv) * (long)D + dl;
+    for (int reg = 0; reg < 16; ++reg) {
+      const int r = acc_row(reg, hi);
+      const int k_r = kv0w + r;
+      if (k_r >= Skv) continue;
+      __hip_bfloat16* out = dkv_out + (((long)b * Skv + k_r) * Hkv + hkv) * (long)D + dl;
 #pragma unroll
-    for (int dc = 0; dc < DCOL; ++dc) out[dc * 32] = __float2bfloat16(acc_out[dc][reg]);
+      for (int dc = 0; dc < DCOL; ++dc) out[dc * 32] = __float2bfloat16(acc_out[dc][reg]);
+    }
   }
 }
 
@@ -424,8 +520,7 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::T
                                  long modarg, at::Tensor slopes) {
   TORCH_CHECK(q.is_cuda());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_bwd: bf16 only");
-  const int Dq_ = q.size(3);
-  auto rs = [Dq_](at::Tensor& t) {
+  auto rs = [](at::Tensor& t) {
     const int S = t.size(1), H = t.size(2), D = t.size(3);
     if (!(t.stride(3) == 1 && t.stride(2) == D && t.stride(0) == (long)S * t.stride(1)))
       t = t.contiguous();
@@ -442,13 +537,14 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::T
   auto drow = at::empty({(long)B * Sq * Hq}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
 
-  {  // preprocess
+  {  // preprocess (one wave per row; dout may be strided, o is contiguous)
     const long rows = (long)B * Sq * Hq;
-    const long grid = cdiv(rows, TPB / WAVE);
-    bwd_preprocess_kernel<<<grid, TPB, 0, stream>>>(
+    const long grid = cdiv(rows, 256 / WAVE);
+    // note: dout row here means the (b,s) row; kernel re-derives h
+    bwd_preprocess_kernel<<<grid, 256, 0, stream>>>(
         reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
         reinterpret_cast<const __hip_bfloat16*>(o.data_ptr()),
-        drow.data_ptr<float>(), rows, Hq, D);
+        drow.data_ptr<float>(), rows, D, do_rs, Hq);
   }
 
   dim3 gq(cdiv(Sq, QPB), Hq, B);

@@ -3,39 +3,45 @@
 // (/root/reference/models/attention/flash_attention.py:78-156 — "Simple
 // approach without tiling for now" — and simple_attention.py, flex_attention.py).
 //
-// Design (gfx950): 4 waves/block, each wave owns 32 q rows (128/block);
-// KV tiles of 32 staged cooperatively in LDS (K row-major for A-fragments,
-// V transposed for B-fragments); swapped QK^T (see attn_common.h) keeps the
-// online softmax lane-local; GQA reads the shared KV head directly (no
-// repeat); causal/sliding-window tiles are skipped at block level.
-// BSHD layout: q [B,Sq,Hq,D], k/v [B,Skv,Hkv,D], o [B,Sq,Hq,D], lse [B,Hq,Sq].
+// Structure (gfx950), following the guide's 8-wave attention ladder:
+//   - NW waves/block, each wave owns 32 q rows (8 waves = 256 q rows share
+//     every K/V tile -> staging traffic amortized 8x),
+//   - KVB-row K/V tiles staged in LDS: K row-major (A-fragments via
+//     ds_read_b128, +16B row pad = conflict-free for the 16-lane groups),
+//     V transposed (B-fragments contiguous; pair-rows staged as b32 writes),
+//   - T14 async-stage split: tile t+1's global loads ISSUE before tile t's
+//     MFMA work, the LDS write happens after the barrier — HBM latency hides
+//     under QK^T/PV (guide §6 Guideline 15),
+//   - swapped QK^T (attn_common.h) keeps the online softmax lane-local,
+//   - s_setprio(1) around the MFMA clusters (guide T5: +4-7% on attention),
+//   - causal/sliding-window tiles skipped at block level; GQA reads the
+//     shared KV head directly; BSHD layout, strided views accepted.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "attn_common.h"
 
 namespace {
 
-constexpr int KVB = 32;   // kv tile
-constexpr int QPW = 32;   // q rows per wave
-constexpr int NW = 4;     // waves per block
-constexpr int QPB = QPW * NW;  // q rows per block
+constexpr int QPW = 32;  // q rows per wave
 
-template <int D, int MOD>
-__global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
+template <int D, int MOD, int NW, int KVB>
+__global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, const float* __restrict__ slopes,
     int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
     long q_rs, long k_rs, long v_rs) {
-  constexpr int DBLK = D / 16;   // QK^T k-slots (over d)
-  constexpr int DCOL = D / 32;   // PV output column tiles
-  constexpr int KPAD = 8;        // elements of row padding (16B) against bank conflicts
-  constexpr int KSTR = D + KPAD;
-  constexpr int VSTR = KVB + 8;
+  constexpr int TPB = NW * WAVE;
+  constexpr int QPB = NW * QPW;
+  constexpr int DBLK = D / 16;  // QK^T d-slots
+  constexpr int DCOL = D / 32;  // PV output column tiles
+  constexpr int KT = KVB / 32;  // 32-row k sub-tiles
+  constexpr int KSTR = D + 8;   // K image row stride (elements)
+  constexpr int VSTR = KVB + 8; // V^T image row stride
 
   __shared__ __hip_bfloat16 smem[KVB * KSTR + D * VSTR];
-  __hip_bfloat16* k_lds = smem;            // [KVB][KSTR] row-major
-  __hip_bfloat16* vt_lds = smem + KVB * KSTR;  // [D][VSTR] transposed V
+  __hip_bfloat16* k_lds = smem;
+  __hip_bfloat16* vt_lds = smem + KVB * KSTR;
 
   const int b = blockIdx.z;
   const int hq = blockIdx.y;
@@ -53,7 +59,7 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
   const int q_off = Skv - Sq;
   const int q_pos = qrow + q_off;
 
-  // ---- Q fragments (B-operand: lane holds Q[q=lq][16*dblk + 8*hi + j]) ----
+  // ---- Q fragments ----
   bf16x8 qf[DBLK];
   {
     const __hip_bfloat16* qp =
@@ -67,7 +73,7 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
     }
   }
 
-  // ---- kv range for this block ----
+  // ---- kv range ----
   const int blk_qpos_lo = qtile * QPB + q_off;
   const int blk_qpos_hi = blk_qpos_lo + QPB - 1;
   int kv_lo = 0, kv_hi = Skv;
@@ -81,6 +87,68 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
   }
   const float slope = (MOD == MOD_ALIBI) ? slopes[hq] : 0.f;
 
+  // ---- staging helpers (T14 split: load -> regs early, write -> LDS late) --
+  // K: KU4 uint4 chunks per thread (row-major image, write as b128).
+  // V: VC chunks of (2 rows x 8 d) per thread, written as 8 b32 into the
+  //    transposed image (2 k-elements per write).
+  constexpr int K_TOT = KVB * (D / 8);            // uint4 per K tile
+  constexpr int KU4 = (K_TOT + TPB - 1) / TPB;
+  constexpr int V_TOT = (KVB / 2) * (D / 8);      // chunks per V tile
+  constexpr int VC = (V_TOT + TPB - 1) / TPB;
+
+  uint4 kreg[KU4], vreg[VC][2];
+
+  auto stage_load = [&](int kv0) {
+#pragma unroll
+    for (int c = 0; c < KU4; ++c) {
+      const int u = tid + c * TPB;
+      const int row = u / (D / 8);
+      const int d0 = (u % (D / 8)) * 8;
+      const bool ok = row < KVB && kv0 + row < Skv;
+      kreg[c] = ok ? *reinterpret_cast<const uint4*>(
+                         k + ((long)b * Skv + kv0 + row) * k_rs + (long)hkv * D + d0)
+                   : uint4{0, 0, 0, 0};
+    }
+#pragma unroll
+    for (int c = 0; c < VC; ++c) {
+      const int u = tid + c * TPB;
+      const int row = (u / (D / 8)) * 2;
+      const int d0 = (u % (D / 8)) * 8;
+      const bool ok0 = row < KVB && kv0 + row < Skv;
+      const bool ok1 = row + 1 < KVB && kv0 + row + 1 < Skv;
+      const long base = ((long)b * Skv + kv0 + row) * v_rs + (long)hkv * D + d0;
+      vreg[c][0] = ok0 ? *reinterpret_cast<const uint4*>(v + base) : uint4{0, 0, 0, 0};
+      vreg[c][1] = ok1 ? *reinterpret_cast<const uint4*>(v + base + v_rs) : uint4{0, 0, 0, 0};
+    }
+  };
+
+  auto stage_write = [&]() {
+#pragma unroll
+    for (int c = 0; c < KU4; ++c) {
+      const int u = tid + c * TPB;
+      const int row = u / (D / 8);
+      const int d0 = (u % (D / 8)) * 8;
+      if (row < KVB)
+        *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = kreg[c];
+    }
+#pragma unroll
+    for (int c = 0; c < VC; ++c) {
+      const int u = tid + c * TPB;
+      const int row = (u / (D / 8)) * 2;
+      const int d0 = (u % (D / 8)) * 8;
+      if (row < KVB) {
+        Bf16x8U v0, v1;
+        *reinterpret_cast<uint4*>(v0.s) = vreg[c][0];
+        *reinterpret_cast<uint4*>(v1.s) = vreg[c][1];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const uint pair = (uint)v0.s[j] | ((uint)v1.s[j] << 16);
+          *reinterpret_cast<uint*>(vt_lds + (d0 + j) * VSTR + row) = pair;
+        }
+      }
+    }
+  };
+
   float m = -INFINITY, l = 0.f;
   float o_acc[DCOL][16];
 #pragma unroll
@@ -88,70 +156,61 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[dc][r] = 0.f;
 
+  // prologue: stage tile 0
+  stage_load(kv_lo);
+  stage_write();
+  __syncthreads();
+
   for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
-    // ---- cooperative staging: K row-major, V transposed ----
-    {
-      constexpr int U4ROW = D / 8;                      // uint4 per row
-      constexpr int TOT = KVB * U4ROW;                  // uint4 per tile
-      for (int u = tid; u < TOT; u += NW * WAVE) {
-        const int row = u / U4ROW;
-        const int d0 = (u % U4ROW) * 8;
-        const bool valid = kv0 + row < Skv;
-        const long r0 = (long)b * Skv + (valid ? kv0 + row : 0);
-        const long hd = (long)hkv * D + d0;
-        Bf16x8U kv_u;
-        *reinterpret_cast<uint4*>(kv_u.s) =
-            valid ? *reinterpret_cast<const uint4*>(k + r0 * k_rs + hd) : uint4{0, 0, 0, 0};
-        *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = *reinterpret_cast<uint4*>(kv_u.s);
-        Bf16x8U vv;
-        *reinterpret_cast<uint4*>(vv.s) =
-            valid ? *reinterpret_cast<const uint4*>(v + r0 * v_rs + hd) : uint4{0, 0, 0, 0};
+    const bool has_next = kv0 + KVB < kv_hi;
+    if (has_next) stage_load(kv0 + KVB);  // loads in flight under the MFMAs
+
+    // ---- S^T = mfma(K, Q) per 32-row k sub-tile ----
+    f32x16 st[KT];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vt_lds[(d0 + j) * VSTR + row] = vv.h[j];
+    for (int kt = 0; kt < KT; ++kt) st[kt] = f32x16{};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kt = 0; kt < KT; ++kt)
+#pragma unroll
+      for (int dblk = 0; dblk < DBLK; ++dblk) {
+        Bf16x8U kf;
+        *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(
+            k_lds + (kt * 32 + lq) * KSTR + dblk * 16 + hi * 8);
+        st[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st[kt], 0, 0, 0);
       }
-    }
-    __syncthreads();
+    __builtin_amdgcn_s_setprio(0);
 
-    // ---- S^T = mfma(K, Q): element (r=k_local, c=q_local) ----
-    f32x16 st = {};
+    // ---- scale + mask + online softmax ----
+    float p[KT][16];
+    float tmax = -INFINITY;
 #pragma unroll
-    for (int dblk = 0; dblk < DBLK; ++dblk) {
-      Bf16x8U kf;
-      *reinterpret_cast<uint4*>(kf.s) =
-          *reinterpret_cast<const uint4*>(k_lds + lq * KSTR + dblk * 16 + hi * 8);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
-    }
-
-    // ---- scale + mask (+alibi) ----
-    float p[16];
+    for (int kt = 0; kt < KT; ++kt)
 #pragma unroll
-    for (int reg = 0; reg < 16; ++reg) {
-      const int k_pos = kv0 + acc_row(reg, hi);
-      const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
-      float s = st[reg] * scale;
-      if constexpr (MOD == MOD_ALIBI) s += slope * (k_pos - q_pos);
-      p[reg] = keep ? s : -INFINITY;
-    }
-
-    // ---- online softmax (lane-local in q) ----
-    float tmax = p[0];
-#pragma unroll
-    for (int reg = 1; reg < 16; ++reg) tmax = fmaxf(tmax, p[reg]);
+      for (int reg = 0; reg < 16; ++reg) {
+        const int k_pos = kv0 + kt * 32 + acc_row(reg, hi);
+        const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+        float s = st[kt][reg] * scale;
+        if constexpr (MOD == MOD_ALIBI) s += slope * (k_pos - q_pos);
+        p[kt][reg] = keep ? s : -INFINITY;
+        tmax = fmaxf(tmax, p[kt][reg]);
+      }
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32));
     const float m_new = fmaxf(m, tmax);
-    const float mc = fmaxf(m_new, -1e30f);  // clamp only inside exponentials
-    const float alpha = __expf(m - mc);      // m=-inf -> 0 on the first live tile
+    const float mc = fmaxf(m_new, -1e30f);
+    const float alpha = __expf(m - mc);
     float psum = 0.f;
 #pragma unroll
-    for (int reg = 0; reg < 16; ++reg) {
-      p[reg] = __expf(p[reg] - mc);  // -inf -> 0
-      psum += p[reg];
-    }
+    for (int kt = 0; kt < KT; ++kt)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        p[kt][reg] = __expf(p[kt][reg] - mc);
+        psum += p[kt][reg];
+      }
     psum += __shfl_xor(psum, 32);
     l = l * alpha + psum;
     m = m_new;
 
-    // rescale o_acc rows (alpha gathered per accumulator row)
 #pragma unroll
     for (int reg = 0; reg < 16; ++reg) {
       const float ar = __shfl(alpha, acc_row(reg, hi));
@@ -159,24 +218,32 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
       for (int dc = 0; dc < DCOL; ++dc) o_acc[dc][reg] *= ar;
     }
 
-    // ---- P -> A fragments, PV ----
-    bf16x8 pa0, pa1;
-    acc_to_afrag(p, pa0, pa1);
+    // ---- P fragments + PV ----
+    bf16x8 pa[KT][2];
+#pragma unroll
+    for (int kt = 0; kt < KT; ++kt) acc_to_afrag(p[kt], pa[kt][0], pa[kt][1]);
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dc = 0; dc < DCOL; ++dc) {
       f32x16 acc;
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc[r] = o_acc[dc][r];
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        Bf16x8U vf;
-        *reinterpret_cast<uint4*>(vf.s) = *reinterpret_cast<const uint4*>(
-            vt_lds + (dc * 32 + lq) * VSTR + ks * 16 + hi * 8);
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? pa0 : pa1, vf.v, acc, 0, 0, 0);
-      }
+      for (int kt = 0; kt < KT; ++kt)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          Bf16x8U vfr;
+          *reinterpret_cast<uint4*>(vfr.s) = *reinterpret_cast<const uint4*>(
+              vt_lds + (dc * 32 + lq) * VSTR + kt * 32 + ks * 16 + hi * 8);
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kt][ks], vfr.v, acc, 0, 0, 0);
+        }
 #pragma unroll
       for (int r = 0; r < 16; ++r) o_acc[dc][r] = acc[r];
     }
+    __builtin_amdgcn_s_setprio(0);
+
+    __syncthreads();  // all waves done reading this tile's LDS
+    if (has_next) stage_write();
     __syncthreads();
   }
 
@@ -184,7 +251,6 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
   if (q_valid && hi == 0)
     lse[((long)b * Hq + hq) * Sq + qrow] = m + __logf(l);
 
-  // o_acc element (r=q_local, c=d_local): lane&31 = d_local here.
   const int dl = lane & 31;
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
@@ -199,6 +265,34 @@ __global__ __launch_bounds__(NW * WAVE) void attn_fwd_kernel(
   }
 }
 
+int fwd_qpb() {  // q rows per block (needed by the host for grid sizing)
+  static int nw = []() {
+    const char* e = getenv("MCDP_ATTN_NW");
+    return e ? atoi(e) : 8;
+  }();
+  return nw * QPW;
+}
+
+template <int D, int MOD>
+void launch_fwd_cfg(dim3 grid, dim3 block, hipStream_t stream,
+                    const __hip_bfloat16* q, const __hip_bfloat16* k, const __hip_bfloat16* v,
+                    __hip_bfloat16* o, float* lse, const float* slopes,
+                    int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
+                    long q_rs, long k_rs, long v_rs) {
+  const char* e = getenv("MCDP_ATTN_KVB");
+  const int kvb = e ? atoi(e) : 64;
+  const int nw = fwd_qpb() / QPW;
+#define LAUNCH(NW_, KVB_)                                                              \
+  attn_fwd_kernel<D, MOD, NW_, KVB_><<<grid, block, 0, stream>>>(                      \
+      q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs)
+  if (nw == 8 && kvb == 64) LAUNCH(8, 64);
+  else if (nw == 8 && kvb == 32) LAUNCH(8, 32);
+  else if (nw == 4 && kvb == 64) LAUNCH(4, 64);
+  else if (nw == 4 && kvb == 32) LAUNCH(4, 32);
+  else TORCH_CHECK(false, "attn_fwd: unsupported NW/KVB ", nw, "/", kvb);
+#undef LAUNCH
+}
+
 template <int D>
 void launch_fwd(int mod, dim3 grid, dim3 block, hipStream_t stream,
                 const __hip_bfloat16* q, const __hip_bfloat16* k, const __hip_bfloat16* v,
@@ -207,19 +301,19 @@ void launch_fwd(int mod, dim3 grid, dim3 block, hipStream_t stream,
                 long q_rs, long k_rs, long v_rs) {
   switch (mod) {
     case MOD_NONE:
-      attn_fwd_kernel<D, MOD_NONE><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
+      launch_fwd_cfg<D, MOD_NONE>(grid, block, stream, q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_CAUSAL:
-      attn_fwd_kernel<D, MOD_CAUSAL><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
+      launch_fwd_cfg<D, MOD_CAUSAL>(grid, block, stream, q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_SLIDING_WINDOW:
-      attn_fwd_kernel<D, MOD_SLIDING_WINDOW><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
+      launch_fwd_cfg<D, MOD_SLIDING_WINDOW>(grid, block, stream, q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_PREFIX_LM:
-      attn_fwd_kernel<D, MOD_PREFIX_LM><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
+      launch_fwd_cfg<D, MOD_PREFIX_LM>(grid, block, stream, q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     case MOD_ALIBI:
-      attn_fwd_kernel<D, MOD_ALIBI><<<grid, block, 0, stream>>>(q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
+      launch_fwd_cfg<D, MOD_ALIBI>(grid, block, stream, q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs);
       break;
     default:
       TORCH_CHECK(false, "attn_fwd: unknown mod ", mod);
@@ -249,8 +343,9 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, doubl
   auto o = at::empty({B, Sq, Hq, D}, q.options());
   auto lse = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
-  dim3 grid(cdiv(Sq, QPB), Hq, B);
-  dim3 block(NW * WAVE);
+  const int qpb = fwd_qpb();
+  dim3 grid(cdiv(Sq, qpb), Hq, B);
+  dim3 block(qpb / QPW * WAVE);
   const float* sl = slopes.numel() > 0 ? slopes.data_ptr<float>() : nullptr;
   auto* qp = reinterpret_cast<const __hip_bfloat16*>(q.data_ptr());
   auto* kp = reinterpret_cast<const __hip_bfloat16*>(k.data_ptr());
