@@ -58,7 +58,9 @@ def _worker(rank, world_size, port, result_q):
         loss.backward()
         ddp.finalize()
         if rank == 0:
-            grads = {n: p.grad.clone() for n, p in model.named_parameters()}
+            # send numpy copies: CPU-tensor fd-passing over the spawn queue can
+            # race with worker exit (ConnectionResetError in recvfds)
+            grads = {n: p.grad.numpy().copy() for n, p in model.named_parameters()}
             result_q.put(grads)
     finally:
         dist.destroy_process_group()
@@ -81,7 +83,8 @@ def test_ddp_grads_match_single_process():
     # == grad of mean over both halves == single-process grad over full batch
     # (equal token counts per rank).
     for n, g in expected.items():
-        assert torch.allclose(got[n], g, atol=1e-5), f"grad mismatch on {n}"
+        assert torch.allclose(torch.from_numpy(got[n]), g, atol=1e-5), \
+            f"grad mismatch on {n}"
 
 
 def _worker_zero1(rank, world_size, port, result_q):
@@ -108,7 +111,7 @@ def _worker_zero1(rank, world_size, port, result_q):
         loss.backward()
         opt.step()
         if rank == 0:
-            result_q.put(space.flat_param.clone())
+            result_q.put(space.flat_param.numpy().copy())
     finally:
         dist.destroy_process_group()
 
@@ -122,7 +125,7 @@ def test_zero1_step_runs_and_syncs():
     procs = [ctx.Process(target=_worker_zero1, args=(r, 2, 29513, q)) for r in range(2)]
     for p in procs:
         p.start()
-    flat = q.get()
+    flat = torch.from_numpy(q.get())
     for p in procs:
         p.join(120)
         assert p.exitcode == 0
