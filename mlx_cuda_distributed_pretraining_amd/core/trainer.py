@@ -118,6 +118,9 @@ class Trainer:
         self.start_step = 0
         self.total_tokens = 0
         self.validation_losses: List[tuple] = []
+        self.flat_space = None
+        self.optimizer = None
+        self.ddp = None
         if for_training:
             self.setup_training()
 
