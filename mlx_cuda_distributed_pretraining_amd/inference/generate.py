@@ -26,23 +26,30 @@ def generate_step(
     logits_processors: Optional[List[Callable]] = None,
     prefill_step_size: int = 512,
     stop_tokens: Optional[List[int]] = None,
+    kv_bits: Optional[int] = None,
+    kv_group_size: int = 64,
+    quantized_kv_start: int = 0,
 ) -> Generator[int, None, None]:
+    from .kv_cache import make_cache, maybe_quantize_kv_cache
+
     device = next(model.parameters()).device
     sampler = sampler or (lambda logits: logits.reshape(-1).argmax())
     logits_processors = logits_processors or []
     stop_tokens = set(stop_tokens or [])
 
-    cache = make_prompt_cache(model)
+    cache = make_cache(model, kind="chunked")
     tokens = list(prompt_tokens)
     y = torch.tensor([tokens], dtype=torch.long, device=device)
 
     # chunked prefill (reference :144-154)
     while y.shape[1] > prefill_step_size:
         model(y[:, :prefill_step_size], cache=cache)
+        cache = maybe_quantize_kv_cache(cache, quantized_kv_start, kv_bits, kv_group_size)
         y = y[:, prefill_step_size:]
     logits = model(y, cache=cache)[:, -1, :]
 
     for _ in range(max_tokens):
+        cache = maybe_quantize_kv_cache(cache, quantized_kv_start, kv_bits, kv_group_size)
         for proc in logits_processors:
             logits = proc(tokens, logits)
         tok = int(sampler(logits.reshape(-1)).item())
