@@ -1,7 +1,7 @@
 // Tiled FlashAttention-2 backward (kernel K1 bwd).
 //
 // Deterministic three-kernel design (no atomics):
-//   preprocess: Drow[b,h,s] = sum_d dO*O
+//   preprocess: Drow[b,h,s] = sum_d dO*O  (vectorized 16B loads, 4 rows/wave)
 //   dQ kernel : block per q-tile; recomputes P from (Q,K,lse); accumulates
 //               dQ = (P∘(dP−Drow))·scale @ K in registers
 //   dK kernel : block per kv-tile; loops the GQA group's q heads; accumulates
@@ -11,9 +11,12 @@
 // (merged, the two 32x128 fp32 accumulators alone are 128 VGPRs).
 //
 // Same gfx950 structure as the forward (attn_fwd.hip): 8 waves/block sharing
-// every staged tile, T14 register-prefetch of the next tile under the MFMA
-// clusters, pair-row b32 writes for the transposed LDS images, s_setprio
-// around MFMAs, MFMA layout + acc_to_afrag transform from attn_common.h.
+// every staged tile, XOR-swizzled LDS images (guide §6 Guideline 4 — padding
+// does not fix the D=128 column-read conflict), DOUBLE-BUFFERED tiles with
+// one barrier per iteration, T14 register-prefetch of the next tile under
+// the MFMA clusters, base-2 exponentials with log2(e) folded into the scale,
+// mask-free fast path on interior causal tiles, s_setprio around MFMAs,
+// MFMA layout + acc_to_afrag transform from attn_common.h.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "attn_common.h"
@@ -25,25 +28,34 @@ constexpr int TPB = NW * WAVE;
 constexpr int QPB = 32 * NW;   // q rows per block (dQ kernel)
 constexpr int KPB = 32 * NW;   // kv rows per block (dK/dV kernels)
 constexpr int KVB = 32;        // kv tile per iteration (dQ kernel)
+constexpr float LOG2E = 1.4426950408889634f;
 
 // ---------------- preprocess: Drow = rowsum(dO * O) ----------------
+// 4 rows per wave: 16 lanes x 8 elements cover a D=128 row in one uint4 load.
 __global__ void bwd_preprocess_kernel(const __hip_bfloat16* __restrict__ dout,
                                       const __hip_bfloat16* __restrict__ o,
                                       float* __restrict__ drow,
                                       long rows, int D, long do_rs, int Hq) {
-  // one wave per (b,s,h) row; row index = (b*S+s)*Hq+h; o is contiguous BSHD,
-  // dout may be row-strided.
-  const long row = blockIdx.x * (long)(256 / WAVE) + threadIdx.x / WAVE;
-  if (row >= rows) return;
+  const int lpr = D / 8;                  // lanes needed per row (16 at D=128)
+  const int rpw = WAVE / lpr;             // rows per wave (4 at D=128)
   const int lane = threadIdx.x % WAVE;
+  const long row = (blockIdx.x * (long)(256 / WAVE) + threadIdx.x / WAVE) * rpw
+                   + lane / lpr;
+  if (row >= rows) return;
   const long bs = row / Hq;
   const int h = (int)(row % Hq);
-  const __hip_bfloat16* dp = dout + bs * do_rs + (long)h * D;
-  const __hip_bfloat16* op = o + row * D;
+  const int d0 = (lane % lpr) * 8;
+  Bf16x8U du, ou;
+  *reinterpret_cast<uint4*>(du.s) =
+      *reinterpret_cast<const uint4*>(dout + bs * do_rs + (long)h * D + d0);
+  *reinterpret_cast<uint4*>(ou.s) =
+      *reinterpret_cast<const uint4*>(o + row * D + d0);
   float acc = 0.f;
-  for (int i = lane; i < D; i += WAVE) acc += to_f32(dp[i]) * to_f32(op[i]);
-  acc = wave_reduce_sum(acc);
-  if (lane == 0) drow[row] = acc;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc += to_f32(du.h[j]) * to_f32(ou.h[j]);
+  // segmented reduce within the lpr-lane group
+  for (int off = lpr / 2; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
+  if (lane % lpr == 0) drow[row] = acc;
 }
 
 // q-range of kv-tile blocks per mod (in q-row space)
@@ -73,14 +85,11 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     long q_rs, long k_rs, long v_rs, long do_rs) {
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
-  constexpr int KSTR = D + 8;
-  constexpr int VSTR = KVB + 8;
+  constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K/V row-major images
+  constexpr int TSWZ = 3;                      // K^T image (KVB=32 -> 64 B rows)
+  constexpr int TILE = KVB * D * 2 + D * KVB;  // K rm + V rm + K^T
 
-  // K row-major + K^T + V row-major
-  __shared__ __hip_bfloat16 smem[KVB * KSTR + D * VSTR + KVB * KSTR];
-  __hip_bfloat16* k_lds = smem;
-  __hip_bfloat16* kt_lds = smem + KVB * KSTR;
-  __hip_bfloat16* v_lds = kt_lds + D * VSTR;
+  __shared__ __hip_bfloat16 smem[2 * TILE];
 
   const int b = blockIdx.z, hq = blockIdx.y, qtile = blockIdx.x;
   const int hkv = hq / (Hq / Hkv);
@@ -91,7 +100,8 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
   const bool q_valid = qrow < Sq;
   const int q_off = Skv - Sq;
   const int q_pos = qrow + q_off;
-  const float slope = (MOD == MOD_ALIBI) ? slopes[hq] : 0.f;
+  const float scale2 = scale * LOG2E;
+  const float slope2 = (MOD == MOD_ALIBI) ? slopes[hq] * LOG2E : 0.f;
 
   // Q and dO fragments (B-operand layout: lane holds row q=lq, 8 d values)
   bf16x8 qf[DBLK], dof[DBLK];
@@ -109,7 +119,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
       dof[dblk] = ud.v;
     }
   }
-  const float Lq = q_valid ? lse[((long)b * Hq + hq) * Sq + qrow] : INFINITY;
+  const float Lq2 = (q_valid ? lse[((long)b * Hq + hq) * Sq + qrow] : INFINITY) * LOG2E;
   const float Dq = q_valid ? drow[((long)b * Sq + qrow) * Hq + hq] : 0.f;
 
   // kv range (same as forward)
@@ -148,24 +158,28 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
       vreg[c][1] = ok1 ? *reinterpret_cast<const uint4*>(v + vb + v_rs) : uint4{0, 0, 0, 0};
     }
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](int bufsel) {
+    __hip_bfloat16* k_lds = smem + bufsel * TILE;
+    __hip_bfloat16* v_lds = k_lds + KVB * D;
+    __hip_bfloat16* kt_lds = v_lds + KVB * D;
 #pragma unroll
     for (int c = 0; c < NCH; ++c) {
       const int u = tid + c * TPB;
       const int row = (u / (D / 8)) * 2;
       const int d0 = (u % (D / 8)) * 8;
       if (row >= KVB) continue;
-      *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = kreg[c][0];
-      *reinterpret_cast<uint4*>(k_lds + (row + 1) * KSTR + d0) = kreg[c][1];
-      *reinterpret_cast<uint4*>(v_lds + row * KSTR + d0) = vreg[c][0];
-      *reinterpret_cast<uint4*>(v_lds + (row + 1) * KSTR + d0) = vreg[c][1];
+      *reinterpret_cast<uint4*>(k_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = kreg[c][0];
+      *reinterpret_cast<uint4*>(k_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = kreg[c][1];
+      *reinterpret_cast<uint4*>(v_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = vreg[c][0];
+      *reinterpret_cast<uint4*>(v_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = vreg[c][1];
       Bf16x8U k0, k1;
       *reinterpret_cast<uint4*>(k0.s) = kreg[c][0];
       *reinterpret_cast<uint4*>(k1.s) = kreg[c][1];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
+        const int drw = d0 + j;
         const uint pair = (uint)k0.s[j] | ((uint)k1.s[j] << 16);
-        *reinterpret_cast<uint*>(kt_lds + (d0 + j) * VSTR + row) = pair;
+        *reinterpret_cast<uint*>(kt_lds + drw * KVB + (row ^ ((drw & TSWZ) << 3))) = pair;
       }
     }
   };
@@ -177,37 +191,65 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     for (int r = 0; r < 16; ++r) dq_acc[dc][r] = 0.f;
 
   stage_load(kv_lo);
-  stage_write();
+  stage_write(0);
   __syncthreads();
 
+  int buf = 0;
   for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
     const bool has_next = kv0 + KVB < kv_hi;
     if (has_next) stage_load(kv0 + KVB);
+
+    const __hip_bfloat16* k_lds = smem + buf * TILE;
+    const __hip_bfloat16* v_lds = k_lds + KVB * D;
+    const __hip_bfloat16* kt_lds = v_lds + KVB * D;
 
     // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
     f32x16 st = {}, dpt = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dblk = 0; dblk < DBLK; ++dblk) {
+      const int col = (dblk * 16 + hi * 8) ^ ((lq & KSWZ) << 3);
       Bf16x8U kf, vf;
-      *reinterpret_cast<uint4*>(kf.s) =
-          *reinterpret_cast<const uint4*>(k_lds + lq * KSTR + dblk * 16 + hi * 8);
-      *reinterpret_cast<uint4*>(vf.s) =
-          *reinterpret_cast<const uint4*>(v_lds + lq * KSTR + dblk * 16 + hi * 8);
+      *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(k_lds + lq * D + col);
+      *reinterpret_cast<uint4*>(vf.s) = *reinterpret_cast<const uint4*>(v_lds + lq * D + col);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
       dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf.v, dof[dblk], dpt, 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
+    if (has_next) stage_write(buf ^ 1);  // vmcnt hides under the MFMAs (T14)
+
+    // interior tiles: every k row of this tile is kept for every q row of
+    // this wave (wave-uniform branch)
+    bool full = kv0 + KVB <= Skv;
+    if constexpr (MOD == MOD_CAUSAL) {
+      full = full && (kv0 + KVB - 1 <= q0w + q_off);
+    } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+      full = full && (kv0 + KVB - 1 <= q0w + q_off) &&
+             ((q0w + 31 + q_off) - kv0 < modarg);
+    } else if constexpr (MOD == MOD_PREFIX_LM) {
+      full = full && ((kv0 + KVB - 1 <= q0w + q_off) || (kv0 + KVB <= modarg));
+    } else if constexpr (MOD == MOD_ALIBI) {
+      full = false;
+    }
+
     float ds[16];
+    if (full) {
 #pragma unroll
-    for (int reg = 0; reg < 16; ++reg) {
-      const int k_pos = kv0 + acc_row(reg, hi);
-      const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
-      float s = st[reg] * scale;
-      if constexpr (MOD == MOD_ALIBI) s += slope * (k_pos - q_pos);
-      const float p = keep ? __expf(s - Lq) : 0.f;
-      ds[reg] = p * (dpt[reg] - Dq) * scale;
+      for (int reg = 0; reg < 16; ++reg) {
+        const float p = __builtin_amdgcn_exp2f(st[reg] * scale2 - Lq2);
+        ds[reg] = p * (dpt[reg] - Dq) * scale;
+      }
+    } else {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int k_pos = kv0 + acc_row(reg, hi);
+        const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+        float s2 = st[reg] * scale2;
+        if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
+        const float p = keep ? __builtin_amdgcn_exp2f(s2 - Lq2) : 0.f;
+        ds[reg] = p * (dpt[reg] - Dq) * scale;
+      }
     }
 
     bf16x8 da0, da1;
@@ -220,9 +262,10 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
       for (int r = 0; r < 16; ++r) acc[r] = dq_acc[dc][r];
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
+        const int drw = dc * 32 + lq;
         Bf16x8U kb;  // B = K[16k x 32d] from K^T image
         *reinterpret_cast<uint4*>(kb.s) = *reinterpret_cast<const uint4*>(
-            kt_lds + (dc * 32 + lq) * VSTR + ks * 16 + hi * 8);
+            kt_lds + drw * KVB + ((ks * 16 + hi * 8) ^ ((drw & TSWZ) << 3)));
         acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb.v, acc, 0, 0, 0);
       }
 #pragma unroll
@@ -231,8 +274,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     __syncthreads();
-    if (has_next) stage_write();
-    __syncthreads();
+    buf ^= 1;
   }
 
   // store dq: element (r=q_local, c=d_local)
@@ -260,15 +302,13 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
     long q_rs, long k_rs, long v_rs, long do_rs) {
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
-  constexpr int KSTR = D + 8;
-  constexpr int VSTR = 32 + 8;
+  constexpr int KSWZ = (D >= 128) ? 15 : 7;  // row-major q/do images
+  constexpr int TSWZ = 3;                    // X^T image (32 rows -> 64 B)
+  // q-tile images: Q rm; X^T (X = Q for dK, dO for dV); dO rm (dK only)
+  constexpr int TILE = 32 * D + D * 32 + (WANT_DK ? 32 * D : 0);
 
-  // q-tile images: Q row-major; X^T (X = Q for dK, dO for dV); dO rm (dK only)
-  __shared__ __hip_bfloat16 smem[32 * KSTR + D * VSTR + 32 * KSTR];
-  __shared__ float stats_lds[2][32];
-  __hip_bfloat16* q_lds = smem;
-  __hip_bfloat16* xt_lds = smem + 32 * KSTR;
-  __hip_bfloat16* do_lds = xt_lds + D * VSTR;
+  __shared__ __hip_bfloat16 smem[2 * TILE];
+  __shared__ float stats_lds[2][2][32];  // [buf][L|D][qrow]
 
   const int b = blockIdx.z, hkv = blockIdx.y, kvtile = blockIdx.x;
   const int group = Hq / Hkv;
@@ -278,6 +318,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
   const int krow = kv0w + lk;
   const bool k_valid = krow < Skv;
   const int q_off = Skv - Sq;
+  const float scale2 = scale * LOG2E;
 
   // K (and V for dK) fragments: lane holds row k=lk, 8 d values
   bf16x8 kf[DBLK], vf[WANT_DK ? DBLK : 1];
@@ -327,18 +368,21 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
     sreg[0] = ok0 ? *reinterpret_cast<const uint4*>(src + base) : uint4{0, 0, 0, 0};
     sreg[1] = ok1 ? *reinterpret_cast<const uint4*>(src + base + rs_) : uint4{0, 0, 0, 0};
   };
-  auto stage_write = [&]() {
+  auto stage_write = [&](int bufsel) {
+    __hip_bfloat16* q_lds = smem + bufsel * TILE;
+    __hip_bfloat16* xt_lds = q_lds + 32 * D;
+    __hip_bfloat16* do_lds = xt_lds + D * 32;  // only sized/used when WANT_DK
     const int u = is_q_half ? tid : tid - CH_TOT;
     if (u >= CH_TOT) return;
     const int row = (u / (D / 8)) * 2;
     const int d0 = (u % (D / 8)) * 8;
     // row-major images: Q always; dO only for dK
     if (is_q_half) {
-      *reinterpret_cast<uint4*>(q_lds + row * KSTR + d0) = sreg[0];
-      *reinterpret_cast<uint4*>(q_lds + (row + 1) * KSTR + d0) = sreg[1];
+      *reinterpret_cast<uint4*>(q_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = sreg[0];
+      *reinterpret_cast<uint4*>(q_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = sreg[1];
     } else if constexpr (WANT_DK) {
-      *reinterpret_cast<uint4*>(do_lds + row * KSTR + d0) = sreg[0];
-      *reinterpret_cast<uint4*>(do_lds + (row + 1) * KSTR + d0) = sreg[1];
+      *reinterpret_cast<uint4*>(do_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = sreg[0];
+      *reinterpret_cast<uint4*>(do_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = sreg[1];
     }
     // transposed image: Q^T for dK, dO^T for dV
     if (is_q_half == WANT_DK) {
@@ -347,8 +391,9 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
       *reinterpret_cast<uint4*>(x1.s) = sreg[1];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
+        const int drw = d0 + j;
         const uint pair = (uint)x0.s[j] | ((uint)x1.s[j] << 16);
-        *reinterpret_cast<uint*>(xt_lds + (d0 + j) * VSTR + row) = pair;
+        *reinterpret_cast<uint*>(xt_lds + drw * 32 + (row ^ ((drw & TSWZ) << 3))) = pair;
       }
     }
   };
@@ -363,20 +408,26 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
       hq_ = hkv * group + it / ntiles;
       q0 = q_lo + (it % ntiles) * 32;
     };
+    auto load_stats = [&](int bufsel, int hq_, int q0) {
+      if (tid < 32) {
+        const int qr = q0 + tid;
+        stats_lds[bufsel][0][tid] =
+            ((qr < Sq) ? lse[((long)b * Hq + hq_) * Sq + qr] : INFINITY) * LOG2E;
+        stats_lds[bufsel][1][tid] =
+            (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + hq_] : 0.f;
+      }
+    };
     int hq_, q0;
     iter_to(0, hq_, q0);
     stage_load(hq_, q0);
-    stage_write();
-    if (tid < 32) {
-      const int qr = q0 + tid;
-      stats_lds[0][tid] = (qr < Sq) ? lse[((long)b * Hq + hq_) * Sq + qr] : INFINITY;
-      stats_lds[1][tid] = (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + hq_] : 0.f;
-    }
+    stage_write(0);
+    load_stats(0, hq_, q0);
     __syncthreads();
 
+    int buf = 0;
     for (int it = 0; it < total_iters; ++it) {
       iter_to(it, hq_, q0);
-      const float slope = (MOD == MOD_ALIBI) ? slopes[hq_] : 0.f;
+      const float slope2 = (MOD == MOD_ALIBI) ? slopes[hq_] * LOG2E : 0.f;
       const bool has_next = it + 1 < total_iters;
       int nhq, nq0;
       if (has_next) {
@@ -384,39 +435,75 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
         stage_load(nhq, nq0);
       }
 
+      const __hip_bfloat16* q_lds = smem + buf * TILE;
+      const __hip_bfloat16* xt_lds = q_lds + 32 * D;
+      const __hip_bfloat16* do_lds = xt_lds + D * 32;
+
       // S = mfma(Q, K^T): A=Q rm frags from LDS, B = register kf.
       // element (r=q_local, c=k_local)
       f32x16 s_acc = {}, dp_acc = {};
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int dblk = 0; dblk < DBLK; ++dblk) {
+        const int col = (dblk * 16 + hi * 8) ^ ((lk & KSWZ) << 3);
         Bf16x8U qa;
-        *reinterpret_cast<uint4*>(qa.s) =
-            *reinterpret_cast<const uint4*>(q_lds + lk * KSTR + dblk * 16 + hi * 8);
+        *reinterpret_cast<uint4*>(qa.s) = *reinterpret_cast<const uint4*>(q_lds + lk * D + col);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa.v, kf[dblk], s_acc, 0, 0, 0);
         if constexpr (WANT_DK) {
           Bf16x8U da;
-          *reinterpret_cast<uint4*>(da.s) =
-              *reinterpret_cast<const uint4*>(do_lds + lk * KSTR + dblk * 16 + hi * 8);
+          *reinterpret_cast<uint4*>(da.s) = *reinterpret_cast<const uint4*>(do_lds + lk * D + col);
           dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da.v, vf[dblk], dp_acc, 0, 0, 0);
         }
       }
       __builtin_amdgcn_s_setprio(0);
 
+      if (has_next) {
+        stage_write(buf ^ 1);  // vmcnt hides under the MFMAs (T14)
+        load_stats(buf ^ 1, nhq, nq0);
+      }
+
+      // interior (q tile entirely below this wave's k rows): mask-free
+      bool full = q0 + 31 < Sq;
+      if constexpr (MOD == MOD_CAUSAL) {
+        full = full && (q0 + q_off >= kv0w + 31);
+      } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+        full = full && (q0 + q_off >= kv0w + 31) &&
+               ((q0 + 31 + q_off) - kv0w < modarg);
+      } else if constexpr (MOD == MOD_PREFIX_LM) {
+        // second clause wave-uniform: whole wave's k rows inside the prefix
+        full = full && ((q0 + q_off >= kv0w + 31) || (kv0w + 31 < modarg));
+      } else {
+        full = false;  // MOD_NONE boundary Sq checks + ALIBI slope
+      }
+
       float pv[16];
+      if (full) {
 #pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        const int q_r = q0 + acc_row(reg, hi);
-        const int q_pos = q_r + q_off;
-        const int k_pos = krow;
-        const bool keep = k_valid && q_r < Sq && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
-        float s = s_acc[reg] * scale;
-        if constexpr (MOD == MOD_ALIBI) s += slope * (k_pos - q_pos);
-        const float p = keep ? __expf(s - stats_lds[0][acc_row(reg, hi)]) : 0.f;
-        if constexpr (WANT_DK) {
-          pv[reg] = p * (dp_acc[reg] - stats_lds[1][acc_row(reg, hi)]) * scale;  // dS
-        } else {
-          pv[reg] = p;
+        for (int reg = 0; reg < 16; ++reg) {
+          const int qr = acc_row(reg, hi);
+          const float p = __builtin_amdgcn_exp2f(s_acc[reg] * scale2 - stats_lds[buf][0][qr]);
+          if constexpr (WANT_DK) {
+            pv[reg] = p * (dp_acc[reg] - stats_lds[buf][1][qr]) * scale;  // dS
+          } else {
+            pv[reg] = p;
+          }
+        }
+      } else {
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          const int qr = acc_row(reg, hi);
+          const int q_r = q0 + qr;
+          const int q_pos = q_r + q_off;
+          const int k_pos = krow;
+          const bool keep = k_valid && q_r < Sq && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+          float s2 = s_acc[reg] * scale2;
+          if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
+          const float p = keep ? __builtin_amdgcn_exp2f(s2 - stats_lds[buf][0][qr]) : 0.f;
+          if constexpr (WANT_DK) {
+            pv[reg] = p * (dp_acc[reg] - stats_lds[buf][1][qr]) * scale;  // dS
+          } else {
+            pv[reg] = p;
+          }
         }
       }
 
@@ -431,9 +518,10 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
         for (int r = 0; r < 16; ++r) acc[r] = acc_out[dc][r];
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
+          const int drw = dc * 32 + lk;
           Bf16x8U xb;  // B = X[16q x 32d] from X^T image
           *reinterpret_cast<uint4*>(xb.s) = *reinterpret_cast<const uint4*>(
-              xt_lds + (dc * 32 + lk) * VSTR + ks * 16 + hi * 8);
+              xt_lds + drw * 32 + ((ks * 16 + hi * 8) ^ ((drw & TSWZ) << 3)));
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb.v, acc, 0, 0, 0);
         }
 #pragma unroll
@@ -442,15 +530,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
       __builtin_amdgcn_s_setprio(0);
 
       __syncthreads();
-      if (has_next) {
-        stage_write();
-        if (tid < 32) {
-          const int qr = nq0 + tid;
-          stats_lds[0][tid] = (qr < Sq) ? lse[((long)b * Hq + nhq) * Sq + qr] : INFINITY;
-          stats_lds[1][tid] = (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + nhq] : 0.f;
-        }
-      }
-      __syncthreads();
+      buf ^= 1;
     }
   }
 
@@ -537,10 +617,10 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::T
   auto drow = at::empty({(long)B * Sq * Hq}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
 
-  {  // preprocess (one wave per row; dout may be strided, o is contiguous)
+  {  // preprocess: 4 rows per wave at D=128 (vectorized 16 B loads)
     const long rows = (long)B * Sq * Hq;
-    const long grid = cdiv(rows, 256 / WAVE);
-    // note: dout row here means the (b,s) row; kernel re-derives h
+    const int rpw = WAVE / (D / 8);
+    const long grid = cdiv(rows, (long)(256 / WAVE) * rpw);
     bwd_preprocess_kernel<<<grid, 256, 0, stream>>>(
         reinterpret_cast<const __hip_bfloat16*>(dout.data_ptr()),
         reinterpret_cast<const __hip_bfloat16*>(o.data_ptr()),

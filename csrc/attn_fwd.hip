@@ -6,14 +6,19 @@
 // Structure (gfx950), following the guide's 8-wave attention ladder:
 //   - NW waves/block, each wave owns 32 q rows (8 waves = 256 q rows share
 //     every K/V tile -> staging traffic amortized 8x),
-//   - KVB-row K/V tiles staged in LDS: K row-major (A-fragments via
-//     ds_read_b128, +16B row pad = conflict-free for the 16-lane groups),
-//     V transposed (B-fragments contiguous; pair-rows staged as b32 writes),
+//   - KVB-row K/V tiles in DOUBLE-BUFFERED LDS, ONE barrier per tile:
+//     tile t+1 is written into buf^1 while buf is being consumed,
+//   - XOR-swizzled images (guide §6 Guideline 4: row-major D=128 bf16 read
+//     column-wise by a lane group is an up-to-16-way bank conflict, and +1
+//     padding does NOT fix it): K row-major [KVB][D] with element column
+//     col ^ ((row&15)<<3); V transposed [D][KVB] with col ^ ((drow&7)<<3),
 //   - T14 async-stage split: tile t+1's global loads ISSUE before tile t's
-//     MFMA work, the LDS write happens after the barrier — HBM latency hides
-//     under QK^T/PV (guide §6 Guideline 15),
+//     MFMA work, the LDS write lands right after QK^T (vmcnt hidden under
+//     the MFMAs; guide §6 Guideline 15),
 //   - swapped QK^T (attn_common.h) keeps the online softmax lane-local,
-//   - s_setprio(1) around the MFMA clusters (guide T5: +4-7% on attention),
+//     exponentials in base-2 with log2(e) folded into the scale,
+//   - interior causal tiles take a mask-free fast path (wave-uniform branch),
+//   - s_setprio(1) around the MFMA clusters (guide T5),
 //   - causal/sliding-window tiles skipped at block level; GQA reads the
 //     shared KV head directly; BSHD layout, strided views accepted.
 #include <torch/extension.h>
@@ -23,6 +28,8 @@
 namespace {
 
 constexpr int QPW = 32;  // q rows per wave
+constexpr float LOG2E = 1.4426950408889634f;
+constexpr float LN2 = 0.6931471805599453f;
 
 template <int D, int MOD, int NW, int KVB>
 __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
@@ -36,12 +43,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   constexpr int DBLK = D / 16;  // QK^T d-slots
   constexpr int DCOL = D / 32;  // PV output column tiles
   constexpr int KT = KVB / 32;  // 32-row k sub-tiles
-  constexpr int KSTR = D + 8;   // K image row stride (elements)
-  constexpr int VSTR = KVB + 8; // V^T image row stride
+  // XOR swizzle masks (element units; <<3 = 8-element/16-byte granules)
+  constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K image rows (row bytes = 2D)
+  constexpr int VSWZ = (KVB >= 128) ? 15 : (KVB >= 64 ? 7 : 3);  // V^T rows
+  constexpr int TILE = KVB * D + D * KVB;     // elements per buffer
 
-  __shared__ __hip_bfloat16 smem[KVB * KSTR + D * VSTR];
-  __hip_bfloat16* k_lds = smem;
-  __hip_bfloat16* vt_lds = smem + KVB * KSTR;
+  __shared__ __hip_bfloat16 smem[2 * TILE];
 
   const int b = blockIdx.z;
   const int hq = blockIdx.y;
@@ -58,6 +65,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   const bool q_valid = qrow < Sq;
   const int q_off = Skv - Sq;
   const int q_pos = qrow + q_off;
+  const float scale2 = scale * LOG2E;
 
   // ---- Q fragments ----
   bf16x8 qf[DBLK];
@@ -85,12 +93,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   } else if constexpr (MOD == MOD_PREFIX_LM) {
     kv_hi = min(Skv, max(blk_qpos_hi + 1, modarg));
   }
-  const float slope = (MOD == MOD_ALIBI) ? slopes[hq] : 0.f;
+  const float slope2 = (MOD == MOD_ALIBI) ? slopes[hq] * LOG2E : 0.f;
 
   // ---- staging helpers (T14 split: load -> regs early, write -> LDS late) --
-  // K: KU4 uint4 chunks per thread (row-major image, write as b128).
-  // V: VC chunks of (2 rows x 8 d) per thread, written as 8 b32 into the
-  //    transposed image (2 k-elements per write).
+  // K: KU4 uint4 chunks per thread (row-major swizzled image, b128 writes).
+  // V: VC chunks of (2 rows x 8 d), written as 8 b32 into the transposed
+  //    swizzled image (2 k-elements per write).
   constexpr int K_TOT = KVB * (D / 8);            // uint4 per K tile
   constexpr int KU4 = (K_TOT + TPB - 1) / TPB;
   constexpr int V_TOT = (KVB / 2) * (D / 8);      // chunks per V tile
@@ -122,14 +130,16 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     }
   };
 
-  auto stage_write = [&]() {
+  auto stage_write = [&](int buf) {
+    __hip_bfloat16* k_lds = smem + buf * TILE;
+    __hip_bfloat16* vt_lds = k_lds + KVB * D;
 #pragma unroll
     for (int c = 0; c < KU4; ++c) {
       const int u = tid + c * TPB;
       const int row = u / (D / 8);
       const int d0 = (u % (D / 8)) * 8;
       if (row < KVB)
-        *reinterpret_cast<uint4*>(k_lds + row * KSTR + d0) = kreg[c];
+        *reinterpret_cast<uint4*>(k_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = kreg[c];
     }
 #pragma unroll
     for (int c = 0; c < VC; ++c) {
@@ -142,28 +152,33 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
         *reinterpret_cast<uint4*>(v1.s) = vreg[c][1];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
+          const int drow = d0 + j;
           const uint pair = (uint)v0.s[j] | ((uint)v1.s[j] << 16);
-          *reinterpret_cast<uint*>(vt_lds + (d0 + j) * VSTR + row) = pair;
+          *reinterpret_cast<uint*>(vt_lds + drow * KVB + (row ^ ((drow & VSWZ) << 3))) = pair;
         }
       }
     }
   };
 
-  float m = -INFINITY, l = 0.f;
+  float m = -INFINITY, l = 0.f;  // m in base-2 domain
   float o_acc[DCOL][16];
 #pragma unroll
   for (int dc = 0; dc < DCOL; ++dc)
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[dc][r] = 0.f;
 
-  // prologue: stage tile 0
+  // prologue: stage tile 0 into buf 0
   stage_load(kv_lo);
-  stage_write();
+  stage_write(0);
   __syncthreads();
 
+  int buf = 0;
   for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
     const bool has_next = kv0 + KVB < kv_hi;
     if (has_next) stage_load(kv0 + KVB);  // loads in flight under the MFMAs
+
+    const __hip_bfloat16* k_lds = smem + buf * TILE;
+    const __hip_bfloat16* vt_lds = k_lds + KVB * D;
 
     // ---- S^T = mfma(K, Q) per 32-row k sub-tile ----
     f32x16 st[KT];
@@ -171,40 +186,69 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     for (int kt = 0; kt < KT; ++kt) st[kt] = f32x16{};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int kt = 0; kt < KT; ++kt)
+    for (int kt = 0; kt < KT; ++kt) {
+      const int krow = kt * 32 + lq;
 #pragma unroll
       for (int dblk = 0; dblk < DBLK; ++dblk) {
         Bf16x8U kf;
         *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(
-            k_lds + (kt * 32 + lq) * KSTR + dblk * 16 + hi * 8);
+            k_lds + krow * D + ((dblk * 16 + hi * 8) ^ ((krow & KSWZ) << 3)));
         st[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st[kt], 0, 0, 0);
       }
+    }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- scale + mask + online softmax ----
+    // write tile t+1 into the other buffer: the vmcnt wait on stage_load's
+    // global loads sits under the QK^T MFMAs we just issued (T14)
+    if (has_next) stage_write(buf ^ 1);
+
+    // ---- scale + mask + online softmax (base-2 domain) ----
+    // interior tiles (wave-uniform): every (q,k) pair of this wave is kept
+    bool full = kv0 + KVB <= Skv;
+    if constexpr (MOD == MOD_CAUSAL) {
+      full = full && (kv0 + KVB - 1 <= q0w + q_off);
+    } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+      full = full && (kv0 + KVB - 1 <= q0w + q_off) &&
+             ((q0w + QPW - 1 + q_off) - kv0 < modarg);
+    } else if constexpr (MOD == MOD_PREFIX_LM) {
+      full = full && ((kv0 + KVB - 1 <= q0w + q_off) || (kv0 + KVB <= modarg));
+    } else if constexpr (MOD == MOD_ALIBI) {
+      full = false;  // slope term needs per-element positions anyway
+    }
+
     float p[KT][16];
     float tmax = -INFINITY;
+    if (full) {
 #pragma unroll
-    for (int kt = 0; kt < KT; ++kt)
+      for (int kt = 0; kt < KT; ++kt)
 #pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        const int k_pos = kv0 + kt * 32 + acc_row(reg, hi);
-        const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
-        float s = st[kt][reg] * scale;
-        if constexpr (MOD == MOD_ALIBI) s += slope * (k_pos - q_pos);
-        p[kt][reg] = keep ? s : -INFINITY;
-        tmax = fmaxf(tmax, p[kt][reg]);
-      }
+        for (int reg = 0; reg < 16; ++reg) {
+          p[kt][reg] = st[kt][reg] * scale2;
+          tmax = fmaxf(tmax, p[kt][reg]);
+        }
+    } else {
+#pragma unroll
+      for (int kt = 0; kt < KT; ++kt)
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          const int k_pos = kv0 + kt * 32 + acc_row(reg, hi);
+          const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+          float s = st[kt][reg] * scale2;
+          if constexpr (MOD == MOD_ALIBI) s += slope2 * (k_pos - q_pos);
+          p[kt][reg] = keep ? s : -INFINITY;
+          tmax = fmaxf(tmax, p[kt][reg]);
+        }
+    }
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32));
     const float m_new = fmaxf(m, tmax);
     const float mc = fmaxf(m_new, -1e30f);
-    const float alpha = __expf(m - mc);
+    const float alpha = __builtin_amdgcn_exp2f(m - mc);
     float psum = 0.f;
 #pragma unroll
     for (int kt = 0; kt < KT; ++kt)
 #pragma unroll
       for (int reg = 0; reg < 16; ++reg) {
-        p[kt][reg] = __expf(p[kt][reg] - mc);
+        p[kt][reg] = __builtin_amdgcn_exp2f(p[kt][reg] - mc);
         psum += p[kt][reg];
       }
     psum += __shfl_xor(psum, 32);
@@ -232,9 +276,10 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
       for (int kt = 0; kt < KT; ++kt)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
+          const int drow = dc * 32 + lq;
           Bf16x8U vfr;
           *reinterpret_cast<uint4*>(vfr.s) = *reinterpret_cast<const uint4*>(
-              vt_lds + (dc * 32 + lq) * VSTR + kt * 32 + ks * 16 + hi * 8);
+              vt_lds + drow * KVB + ((kt * 32 + ks * 16 + hi * 8) ^ ((drow & VSWZ) << 3)));
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kt][ks], vfr.v, acc, 0, 0, 0);
         }
 #pragma unroll
@@ -242,14 +287,13 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    __syncthreads();  // all waves done reading this tile's LDS
-    if (has_next) stage_write();
-    __syncthreads();
+    __syncthreads();  // buf consumed by all waves; buf^1 fully written
+    buf ^= 1;
   }
 
   // ---- epilogue ----
   if (q_valid && hi == 0)
-    lse[((long)b * Hq + hq) * Sq + qrow] = m + __logf(l);
+    lse[((long)b * Hq + hq) * Sq + qrow] = (m + __builtin_amdgcn_logf(l)) * LN2;
 
   const int dl = lane & 31;
 #pragma unroll
