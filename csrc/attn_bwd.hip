@@ -86,8 +86,8 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K/V row-major images
-  constexpr int TSWZ = 3;                      // K^T image (KVB=32 -> 64 B rows)
-  constexpr int TILE = KVB * D * 2 + D * KVB;  // K rm + V rm + K^T
+  constexpr int TROW = 64;                     // K^T image row (elements; swzt())
+  constexpr int TILE = KVB * D * 2 + D * TROW; // K rm + V rm + K^T
 
   __shared__ __hip_bfloat16 smem[2 * TILE];
 
@@ -179,7 +179,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
       for (int j = 0; j < 8; ++j) {
         const int drw = d0 + j;
         const uint pair = (uint)k0.s[j] | ((uint)k1.s[j] << 16);
-        *reinterpret_cast<uint*>(kt_lds + drw * KVB + (row ^ ((drw & TSWZ) << 3))) = pair;
+        *reinterpret_cast<uint*>(kt_lds + drw * TROW + (row ^ swzt(drw))) = pair;
       }
     }
   };
@@ -265,7 +265,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
         const int drw = dc * 32 + lq;
         Bf16x8U kb;  // B = K[16k x 32d] from K^T image
         *reinterpret_cast<uint4*>(kb.s) = *reinterpret_cast<const uint4*>(
-            kt_lds + drw * KVB + ((ks * 16 + hi * 8) ^ ((drw & TSWZ) << 3)));
+            kt_lds + drw * TROW + ((ks * 16 + hi * 8) ^ swzt(drw)));
         acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb.v, acc, 0, 0, 0);
       }
 #pragma unroll
@@ -303,9 +303,9 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSWZ = (D >= 128) ? 15 : 7;  // row-major q/do images
-  constexpr int TSWZ = 3;                    // X^T image (32 rows -> 64 B)
+  constexpr int TROW = 64;                   // X^T image row (elements; swzt())
   // q-tile images: Q rm; X^T (X = Q for dK, dO for dV); dO rm (dK only)
-  constexpr int TILE = 32 * D + D * 32 + (WANT_DK ? 32 * D : 0);
+  constexpr int TILE = 32 * D + D * TROW + (WANT_DK ? 32 * D : 0);
 
   __shared__ __hip_bfloat16 smem[2 * TILE];
   __shared__ float stats_lds[2][2][32];  // [buf][L|D][qrow]
@@ -371,7 +371,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
   auto stage_write = [&](int bufsel) {
     __hip_bfloat16* q_lds = smem + bufsel * TILE;
     __hip_bfloat16* xt_lds = q_lds + 32 * D;
-    __hip_bfloat16* do_lds = xt_lds + D * 32;  // only sized/used when WANT_DK
+    __hip_bfloat16* do_lds = xt_lds + D * TROW;  // only sized/used when WANT_DK
     const int u = is_q_half ? tid : tid - CH_TOT;
     if (u >= CH_TOT) return;
     const int row = (u / (D / 8)) * 2;
@@ -393,7 +393,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
       for (int j = 0; j < 8; ++j) {
         const int drw = d0 + j;
         const uint pair = (uint)x0.s[j] | ((uint)x1.s[j] << 16);
-        *reinterpret_cast<uint*>(xt_lds + drw * 32 + (row ^ ((drw & TSWZ) << 3))) = pair;
+        *reinterpret_cast<uint*>(xt_lds + drw * TROW + (row ^ swzt(drw))) = pair;
       }
     }
   };
@@ -437,7 +437,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
 
       const __hip_bfloat16* q_lds = smem + buf * TILE;
       const __hip_bfloat16* xt_lds = q_lds + 32 * D;
-      const __hip_bfloat16* do_lds = xt_lds + D * 32;
+      const __hip_bfloat16* do_lds = xt_lds + D * TROW;
 
       // S = mfma(Q, K^T): A=Q rm frags from LDS, B = register kf.
       // element (r=q_local, c=k_local)
@@ -521,7 +521,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
           const int drw = dc * 32 + lk;
           Bf16x8U xb;  // B = X[16q x 32d] from X^T image
           *reinterpret_cast<uint4*>(xb.s) = *reinterpret_cast<const uint4*>(
-              xt_lds + drw * 32 + ((ks * 16 + hi * 8) ^ ((drw & TSWZ) << 3)));
+              xt_lds + drw * TROW + ((ks * 16 + hi * 8) ^ swzt(drw)));
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb.v, acc, 0, 0, 0);
         }
 #pragma unroll

@@ -63,6 +63,17 @@ __device__ __forceinline__ int acc_row(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
 }
 
+// Swizzle (element units) for TRANSPOSED LDS images stored as [D][64] rows
+// (128 B rows). ds_write bank = (byte/4) % 32, so the row base (drow*32
+// dwords) contributes nothing — the spread must come from the column offset.
+// The staging writes touch d-rows stride 8 within a lane group ((drow&7)
+// constant there), so the swizzle mixes BOTH drow&7 and (drow>>3)&7:
+// conflict-free b128 reads (16-lane group covers all banks via the
+// drow-parity split), 2-way b32 writes. Columns used must be < 64 elements.
+__device__ __forceinline__ int swzt(int drow) {
+  return (((drow & 7) ^ ((drow >> 3) & 7)) << 3);
+}
+
 // keep/mask decision for one score element. q_pos/k_pos are ABSOLUTE
 // positions (q_pos = q_row + Skv - Sq handles KV-cache decode).
 template <int MOD>

@@ -45,8 +45,8 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   constexpr int KT = KVB / 32;  // 32-row k sub-tiles
   // XOR swizzle masks (element units; <<3 = 8-element/16-byte granules)
   constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K image rows (row bytes = 2D)
-  constexpr int VSWZ = (KVB >= 128) ? 15 : (KVB >= 64 ? 7 : 3);  // V^T rows
-  constexpr int TILE = KVB * D + D * KVB;     // elements per buffer
+  constexpr int VROW = 64;                    // V^T image row (elements; swzt())
+  constexpr int TILE = KVB * D + D * VROW;    // elements per buffer
 
   __shared__ __hip_bfloat16 smem[2 * TILE];
 
@@ -154,7 +154,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
         for (int j = 0; j < 8; ++j) {
           const int drow = d0 + j;
           const uint pair = (uint)v0.s[j] | ((uint)v1.s[j] << 16);
-          *reinterpret_cast<uint*>(vt_lds + drow * KVB + (row ^ ((drow & VSWZ) << 3))) = pair;
+          *reinterpret_cast<uint*>(vt_lds + drow * VROW + (row ^ swzt(drow))) = pair;
         }
       }
     }
@@ -279,7 +279,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
           const int drow = dc * 32 + lq;
           Bf16x8U vfr;
           *reinterpret_cast<uint4*>(vfr.s) = *reinterpret_cast<const uint4*>(
-              vt_lds + drow * KVB + ((kt * 32 + ks * 16 + hi * 8) ^ ((drow & VSWZ) << 3)));
+              vt_lds + drow * VROW + ((kt * 32 + ks * 16 + hi * 8) ^ swzt(drow)));
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kt][ks], vfr.v, acc, 0, 0, 0);
         }
 #pragma unroll

@@ -38,6 +38,9 @@ class DataConfig:
     # MI355X extension: synthetic data of a given shape (no-network benches).
     synthetic: bool = False
     synthetic_vocab_size: int = 32000
+    # Streaming shard pipeline (reference fineweb_stream*.py): e.g.
+    # {"source": "/data/shards", "max_cache_gb": 10, "cache_dir": "/tmp/cache"}
+    streaming: Optional[Dict[str, Any]] = None
 
 
 @dataclass

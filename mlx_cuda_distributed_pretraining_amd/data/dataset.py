@@ -50,6 +50,24 @@ class DataManager:
             self.synthetic_vocab = int(getattr(data_cfg, "synthetic_vocab_size", 32000))
             return
 
+        self.stream = None
+        stream_cfg = getattr(data_cfg, "streaming", None)
+        if stream_cfg:
+            from .streaming import DiskSpaceManager, StreamingTokenDataset
+
+            mgr = None
+            if stream_cfg.get("cache_dir"):
+                mgr = DiskSpaceManager(
+                    stream_cfg["cache_dir"],
+                    max_bytes=int(float(stream_cfg.get("max_cache_gb", 10)) * 2**30),
+                )
+            ds = StreamingTokenDataset(
+                stream_cfg["source"], tokenizer, seq_len=self.max_context_size,
+                rank=rank, world_size=world_size,
+                max_tokens=stream_cfg.get("max_tokens"), disk_manager=mgr,
+            )
+            self.stream = ds.iter_batches(batch_size)
+
         input_file = getattr(data_cfg, "input_file", None)
         if input_file:
             self.docs = self._load_jsonl(input_file)
@@ -129,6 +147,11 @@ class DataManager:
                 generator=g,
                 dtype=torch.long,
             )
+        if getattr(self, "stream", None) is not None:
+            try:
+                return next(self.stream)
+            except StopIteration:
+                raise RuntimeError("streaming source exhausted (max_tokens reached)")
         if not self.batch_order:
             raise RuntimeError("No training data loaded")
         idx = (step * self.world_size + self.rank) % len(self.batch_order)
