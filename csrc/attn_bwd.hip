@@ -217,8 +217,6 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    if (has_next) stage_write(buf ^ 1);  // vmcnt hides under the MFMAs (T14)
-
     // interior tiles: every k row of this tile is kept for every q row of
     // this wave (wave-uniform branch)
     bool full = kv0 + KVB <= Skv;
@@ -272,6 +270,10 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
       for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
     }
     __builtin_amdgcn_s_setprio(0);
+
+    // write tile t+1 LAST: the whole iteration hides the global-load flight
+    // (buf^1 was last read before the previous barrier)
+    if (has_next) stage_write(buf ^ 1);
 
     __syncthreads();
     buf ^= 1;
@@ -457,11 +459,6 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
 
-      if (has_next) {
-        stage_write(buf ^ 1);  // vmcnt hides under the MFMAs (T14)
-        load_stats(buf ^ 1, nhq, nq0);
-      }
-
       // interior (q tile entirely below this wave's k rows): mask-free
       bool full = q0 + 31 < Sq;
       if constexpr (MOD == MOD_CAUSAL) {
@@ -528,6 +525,12 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
         for (int r = 0; r < 16; ++r) acc_out[dc][r] = acc[r];
       }
       __builtin_amdgcn_s_setprio(0);
+
+      // write tile t+1 LAST: the whole iteration hides the global-load flight
+      if (has_next) {
+        stage_write(buf ^ 1);
+        load_stats(buf ^ 1, nhq, nq0);
+      }
 
       __syncthreads();
       buf ^= 1;

@@ -198,10 +198,6 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // write tile t+1 into the other buffer: the vmcnt wait on stage_load's
-    // global loads sits under the QK^T MFMAs we just issued (T14)
-    if (has_next) stage_write(buf ^ 1);
-
     // ---- scale + mask + online softmax (base-2 domain) ----
     // interior tiles (wave-uniform): every (q,k) pair of this wave is kept
     bool full = kv0 + KVB <= Skv;
@@ -240,9 +236,15 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
         }
     }
     tmax = fmaxf(tmax, __shfl_xor(tmax, 32));
-    const float m_new = fmaxf(m, tmax);
-    const float mc = fmaxf(m_new, -1e30f);
-    const float alpha = __builtin_amdgcn_exp2f(m - mc);
+    // defer-max (guide T13): while no lane's tile max exceeds the running
+    // max by more than THR (base-2), keep the old max and SKIP the O/l
+    // rescale entirely — P is then bounded by 2^THR, which the fp32
+    // accumulators tolerate (~3x max-abs error vs THR=0). The decision is
+    // wave-uniform and taken before any of this tile's P*V.
+    const float m_old = m;
+    const bool defer = __all(tmax - m <= 8.0f);
+    if (!defer) m = fmaxf(m, tmax);
+    const float mc = fmaxf(m, -1e30f);
     float psum = 0.f;
 #pragma unroll
     for (int kt = 0; kt < KT; ++kt)
@@ -252,14 +254,17 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
         psum += p[kt][reg];
       }
     psum += __shfl_xor(psum, 32);
-    l = l * alpha + psum;
-    m = m_new;
-
+    if (defer) {
+      l += psum;
+    } else {
+      const float alpha = __builtin_amdgcn_exp2f(m_old - mc);
+      l = l * alpha + psum;
 #pragma unroll
-    for (int reg = 0; reg < 16; ++reg) {
-      const float ar = __shfl(alpha, acc_row(reg, hi));
+      for (int reg = 0; reg < 16; ++reg) {
+        const float ar = __shfl(alpha, acc_row(reg, hi));
 #pragma unroll
-      for (int dc = 0; dc < DCOL; ++dc) o_acc[dc][reg] *= ar;
+        for (int dc = 0; dc < DCOL; ++dc) o_acc[dc][reg] *= ar;
+      }
     }
 
     // ---- P fragments + PV ----
@@ -286,6 +291,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
       for (int r = 0; r < 16; ++r) o_acc[dc][r] = acc[r];
     }
     __builtin_amdgcn_s_setprio(0);
+
+    // write tile t+1 into the other buffer LAST: the vmcnt wait on
+    // stage_load's global loads has the whole iteration's MFMA/softmax work
+    // to hide under (T14/async-STAGE). buf^1 was last read in iteration
+    // t-1, before the barrier every wave has already passed.
+    if (has_next) stage_write(buf ^ 1);
 
     __syncthreads();  // buf consumed by all waves; buf^1 fully written
     buf ^= 1;
