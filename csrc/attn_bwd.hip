@@ -82,7 +82,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ drow,
     __hip_bfloat16* __restrict__ dq, const float* __restrict__ slopes,
     int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
-    long q_rs, long k_rs, long v_rs, long do_rs) {
+    long q_rs, long k_rs, long v_rs, long do_rs, long dq_rs) {
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K/V row-major images
@@ -286,7 +286,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
     const int r = acc_row(reg, hi);
     const int q_r = q0w + r;
     if (q_r >= Sq) continue;
-    __hip_bfloat16* dqr = dq + (((long)b * Sq + q_r) * Hq + hq) * (long)D + dl;
+    __hip_bfloat16* dqr = dq + ((long)b * Sq + q_r) * dq_rs + (long)hq * D + dl;
 #pragma unroll
     for (int dc = 0; dc < DCOL; ++dc) dqr[dc * 32] = __float2bfloat16(dq_acc[dc][reg]);
   }
@@ -301,7 +301,7 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
     const float* __restrict__ lse, const float* __restrict__ drow,
     __hip_bfloat16* __restrict__ dkv_out, const float* __restrict__ slopes,
     int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
-    long q_rs, long k_rs, long v_rs, long do_rs) {
+    long q_rs, long k_rs, long v_rs, long do_rs, long dkv_rs) {
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSWZ = (D >= 128) ? 15 : 7;  // row-major q/do images
@@ -546,7 +546,7 @@ store:
       const int r = acc_row(reg, hi);
       const int k_r = kv0w + r;
       if (k_r >= Skv) continue;
-      __hip_bfloat16* out = dkv_out + (((long)b * Skv + k_r) * Hkv + hkv) * (long)D + dl;
+      __hip_bfloat16* out = dkv_out + ((long)b * Skv + k_r) * dkv_rs + (long)hkv * D + dl;
 #pragma unroll
       for (int dc = 0; dc < DCOL; ++dc) out[dc * 32] = __float2bfloat16(acc_out[dc][reg]);
     }
@@ -559,13 +559,14 @@ void launch_bwd_all(dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
                     const __hip_bfloat16* dout, const float* lse, const float* drow,
                     __hip_bfloat16* dq, __hip_bfloat16* dk, __hip_bfloat16* dv,
                     const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
-                    float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs) {
+                    float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs,
+                    long dq_rs, long dk_rs, long dv_rs) {
   attn_bwd_dq_kernel<D, MOD><<<gq, block, 0, stream>>>(
-      q, k, v, dout, lse, drow, dq, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      q, k, v, dout, lse, drow, dq, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs);
   attn_bwd_dkv_kernel<D, MOD, true><<<gkv, block, 0, stream>>>(
-      q, k, v, dout, lse, drow, dk, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      q, k, v, dout, lse, drow, dk, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dk_rs);
   attn_bwd_dkv_kernel<D, MOD, false><<<gkv, block, 0, stream>>>(
-      q, k, v, dout, lse, drow, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      q, k, v, dout, lse, drow, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dv_rs);
 }
 
 template <int D>
@@ -574,22 +575,23 @@ void launch_bwd_mod(int mod, dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
                     const __hip_bfloat16* dout, const float* lse, const float* drow,
                     __hip_bfloat16* dq, __hip_bfloat16* dk, __hip_bfloat16* dv,
                     const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
-                    float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs) {
+                    float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs,
+                    long dq_rs, long dk_rs, long dv_rs) {
   switch (mod) {
     case MOD_NONE:
-      launch_bwd_all<D, MOD_NONE>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      launch_bwd_all<D, MOD_NONE>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_CAUSAL:
-      launch_bwd_all<D, MOD_CAUSAL>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      launch_bwd_all<D, MOD_CAUSAL>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_SLIDING_WINDOW:
-      launch_bwd_all<D, MOD_SLIDING_WINDOW>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      launch_bwd_all<D, MOD_SLIDING_WINDOW>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_PREFIX_LM:
-      launch_bwd_all<D, MOD_PREFIX_LM>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      launch_bwd_all<D, MOD_PREFIX_LM>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_ALIBI:
-      launch_bwd_all<D, MOD_ALIBI>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs);
+      launch_bwd_all<D, MOD_ALIBI>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     default:
       TORCH_CHECK(false, "attn_bwd: unknown mod ", mod);
@@ -598,9 +600,10 @@ void launch_bwd_mod(int mod, dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
 
 }  // namespace
 
-std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
-                                 at::Tensor dout, at::Tensor lse, double scale, long mod,
-                                 long modarg, at::Tensor slopes) {
+std::vector<at::Tensor> attn_bwd_out(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+                                     at::Tensor dout, at::Tensor lse, double scale, long mod,
+                                     long modarg, at::Tensor slopes,
+                                     at::Tensor dq, at::Tensor dk, at::Tensor dv) {
   TORCH_CHECK(q.is_cuda());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn_bwd: bf16 only");
   auto rs = [](at::Tensor& t) {
@@ -614,9 +617,22 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::T
   const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Skv = k.size(1), Hkv = k.size(2);
   TORCH_CHECK(D == 64 || D == 128, "attn_bwd: head_dim must be 64 or 128");
-  auto dq = at::empty({B, Sq, Hq, D}, q.options());
-  auto dk = at::empty({B, Skv, Hkv, D}, k.options());
-  auto dv = at::empty({B, Skv, Hkv, D}, v.options());
+  // outputs may be pre-allocated row-strided views (slices of a fused dQKV
+  // grad buffer) — element (b,s,h,d) at (b*S+s)*stride(1) + h*D + d
+  auto check_out = [&](at::Tensor& t, int S_, int H_) {
+    if (t.numel() == 0) {
+      t = at::empty({B, S_, H_, D}, q.options());
+    } else {
+      TORCH_CHECK(t.size(0) == B && t.size(1) == S_ && t.size(2) == H_ && t.size(3) == D &&
+                      t.stride(3) == 1 && t.stride(2) == D &&
+                      t.stride(0) == (long)S_ * t.stride(1),
+                  "attn_bwd: bad out view");
+    }
+    return t.stride(1);
+  };
+  const long dq_rs = check_out(dq, Sq, Hq);
+  const long dk_rs = check_out(dk, Skv, Hkv);
+  const long dv_rs = check_out(dv, Skv, Hkv);
   auto drow = at::empty({(long)B * Sq * Hq}, q.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentHIPStream();
 
@@ -644,10 +660,20 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::T
   if (D == 64)
     launch_bwd_mod<64>((int)mod, gq, gkv, block, stream, qp, kp, vp, dop,
                        lse.data_ptr<float>(), drow.data_ptr<float>(), dqp, dkp, dvp, sl,
-                       B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs, do_rs);
+                       B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs, do_rs,
+                       dq_rs, dk_rs, dv_rs);
   else
     launch_bwd_mod<128>((int)mod, gq, gkv, block, stream, qp, kp, vp, dop,
                         lse.data_ptr<float>(), drow.data_ptr<float>(), dqp, dkp, dvp, sl,
-                        B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs, do_rs);
+                        B, Sq, Skv, Hq, Hkv, (float)scale, (int)modarg, q_rs, k_rs, v_rs, do_rs,
+                        dq_rs, dk_rs, dv_rs);
   return {dq, dk, dv};
+}
+
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+                                 at::Tensor dout, at::Tensor lse, double scale, long mod,
+                                 long modarg, at::Tensor slopes) {
+  auto none = at::empty({0}, q.options());
+  return attn_bwd_out(q, k, v, o, dout, lse, scale, mod, modarg, slopes,
+                      none, none.clone(), none.clone());
 }

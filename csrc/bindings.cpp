@@ -7,6 +7,8 @@ std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor rstd,
 // rope.hip
 at::Tensor rope_fwd(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
                     long offset, bool conj);
+at::Tensor rope_fwd_out(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
+                        long offset, bool conj, at::Tensor y);
 // swiglu.hip
 at::Tensor swiglu_fwd(at::Tensor gu);
 at::Tensor swiglu_bwd(at::Tensor gu, at::Tensor dy);
@@ -31,6 +33,10 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, doubl
 std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
                                  at::Tensor dout, at::Tensor lse, double scale, long mod,
                                  long modarg, at::Tensor slopes);
+std::vector<at::Tensor> attn_bwd_out(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
+                                     at::Tensor dout, at::Tensor lse, double scale, long mod,
+                                     long modarg, at::Tensor slopes,
+                                     at::Tensor dq, at::Tensor dk, at::Tensor dv);
 // sampling.hip
 at::Tensor sample_token(at::Tensor logits, double temperature, double top_p, double min_p,
                         long seed);
@@ -42,6 +48,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (y, rstd)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (dx, dw)");
   m.def("rope_fwd", &rope_fwd, "RoPE apply (conj=true for backward)");
+  m.def("rope_fwd_out", &rope_fwd_out, "RoPE apply into a strided out view");
   m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");
   m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (loss_sum, ntok, lse)");
@@ -52,6 +59,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step, "fused SGD over flat buffers");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (o, lse)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward (dq, dk, dv)");
+  m.def("attn_bwd_out", &attn_bwd_out,
+        "flash attention backward into strided out views (fused dQKV)");
   m.def("sample_token", &sample_token, "fused temperature/min-p sampling");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");

@@ -14,9 +14,10 @@ template <typename T, bool TRAD>
 __global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
                             const float* __restrict__ cost, const float* __restrict__ sint,
                             long total_groups, int S, int H, int D, int offset, float sgn,
-                            long x_row_stride) {
-  // x may be a strided view (e.g. the q slice of the fused QKV output):
-  // element (b,s,h,d) sits at (b*S+s)*x_row_stride + h*D + d. y is contiguous.
+                            long x_row_stride, long y_row_stride) {
+  // x and y may be strided views (e.g. the q slice of the fused QKV output /
+  // of the fused dQKV grad buffer): element (b,s,h,d) sits at
+  // (b*S+s)*row_stride + h*D + d.
   const int half = D / 2;
   const int groups_per_row = half / 4;  // 4 pairs per thread
   for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total_groups;
@@ -26,6 +27,7 @@ __global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
     const int s = (int)((row / H) % S);
     const int h = (int)(row % H);
     const long xrow_off = (row / H) * x_row_stride + (long)h * D;
+    const long yrow_off = (row / H) * y_row_stride + (long)h * D;
     const int d0 = gi * 4;
     const float* crow = cost + (long)(s + offset) * half + d0;
     const float* srow = sint + (long)(s + offset) * half + d0;
@@ -33,7 +35,7 @@ __global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
     if constexpr (TRAD) {
       // interleaved pairs: (2d, 2d+1); 4 pairs = 8 contiguous elements
       const T* xr = x + xrow_off + 2 * d0;
-      T* yr = y + row * (long)D + 2 * d0;
+      T* yr = y + yrow_off + 2 * d0;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const float c = crow[j], s_ = sgn * srow[j];
@@ -44,7 +46,7 @@ __global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
     } else {
       const T* xa = x + xrow_off + d0;
       const T* xb = xa + half;
-      T* ya = y + row * (long)D + d0;
+      T* ya = y + yrow_off + d0;
       T* yb = ya + half;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
@@ -59,17 +61,24 @@ __global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ y,
 
 }  // namespace
 
-at::Tensor rope_fwd(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
-                    long offset, bool conj) {
+at::Tensor rope_fwd_out(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
+                        long offset, bool conj, at::Tensor y) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 4, "rope: x must be [B,S,H,D]");
   const int B = x.size(0), S = x.size(1), H = x.size(2), D = x.size(3);
-  // allow a row-strided view (q/k slices of the fused QKV projection)
+  // allow row-strided views (q/k slices of the fused QKV projection / grad)
   TORCH_CHECK(x.stride(3) == 1 && x.stride(2) == D && x.stride(0) == S * x.stride(1),
               "rope: x must be [B,S,H,D] with contiguous (h,d) inner block");
   const long x_row_stride = x.stride(1);
   TORCH_CHECK(D % 8 == 0, "rope: head_dim must be a multiple of 8");
   TORCH_CHECK(cost.size(0) >= S + offset, "rope table too small");
-  auto y = at::empty({B, S, H, D}, x.options());  // output always contiguous
+  if (y.numel() == 0) {
+    y = at::empty({B, S, H, D}, x.options());
+  } else {
+    TORCH_CHECK(y.sizes() == x.sizes() && y.stride(3) == 1 && y.stride(2) == D &&
+                    y.stride(0) == S * y.stride(1),
+                "rope: out must be [B,S,H,D] with contiguous (h,d) inner block");
+  }
+  const long y_row_stride = y.stride(1);
   const long total_groups = (long)B * S * H * (D / 8);
   auto stream = at::cuda::getCurrentHIPStream();
   const int block = 256;
@@ -81,14 +90,20 @@ at::Tensor rope_fwd(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditi
       if (traditional)
         rope_kernel<T, true><<<grid, block, 0, stream>>>(
             reinterpret_cast<const T*>(x.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
-            cost.data_ptr<float>(), sint.data_ptr<float>(), total_groups, S, H, D, (int)offset, sgn, x_row_stride);
+            cost.data_ptr<float>(), sint.data_ptr<float>(), total_groups, S, H, D, (int)offset, sgn, x_row_stride, y_row_stride);
       else
         rope_kernel<T, false><<<grid, block, 0, stream>>>(
             reinterpret_cast<const T*>(x.data_ptr()), reinterpret_cast<T*>(y.data_ptr()),
-            cost.data_ptr<float>(), sint.data_ptr<float>(), total_groups, S, H, D, (int)offset, sgn, x_row_stride);
+            cost.data_ptr<float>(), sint.data_ptr<float>(), total_groups, S, H, D, (int)offset, sgn, x_row_stride, y_row_stride);
     } else {
       TORCH_CHECK(false, "rope: unsupported dtype");
     }
   });
   return y;
+}
+
+at::Tensor rope_fwd(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
+                    long offset, bool conj) {
+  return rope_fwd_out(x, cost, sint, traditional, offset, conj,
+                      at::empty({0}, x.options()));
 }

@@ -25,7 +25,7 @@ from typing import Any, Dict, List, Optional
 import torch
 import torch.nn as nn
 
-from ..ops.attention import attention_ref, flash_attention
+from ..ops.attention import attention_ref, flash_attention, rope_flash_attention_qkv
 from ..ops.rmsnorm import RMSNorm
 from ..ops.rope import RopeTable, apply_rope
 from ..ops.swiglu import swiglu
@@ -144,6 +144,23 @@ class Attention(nn.Module):
     def forward(self, x: torch.Tensor, cache: Optional[KVCache] = None) -> torch.Tensor:
         B, S, _ = x.shape
         qkv = self.wqkv(x)
+
+        atype = self.args.attention_type
+        if cache is None and atype != "simple":
+            # training fast path: one fused autograd node over the fused QKV
+            # tensor (RoPE + attention; backward writes dQKV in place — no
+            # split-backward grad cat)
+            cos, sin = self.rope_table.get(S, x.device, 0)
+            o = rope_flash_attention_qkv(
+                qkv, cos, sin, self.n_heads, self.n_kv_heads, self.head_dim,
+                traditional=self.args.rope_traditional, causal=True,
+                scale=self.scale,
+                window=self.args.attention_window if atype == "flex" else None,
+                prefix_len=self.args.attention_prefix_len if atype == "flex" else None,
+                alibi_slopes=self.alibi_slopes if self.args.use_alibi else None,
+            )
+            return self.wo(o.reshape(B, S, -1))
+
         q, k, v = qkv.split(
             [
                 self.n_heads * self.head_dim,

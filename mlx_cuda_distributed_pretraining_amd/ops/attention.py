@@ -186,6 +186,89 @@ def flash_attention(
     return _FlashAttnFn.apply(q, k, v, causal, scale, window, prefix_len, alibi_slopes)
 
 
+class _RopeAttnQKVFn(torch.autograd.Function):
+    """Fused qkv-slice -> RoPE(q,k) -> flash attention, HIP path only.
+
+    Takes the FUSED [B, S, (Hq+2*Hkv)*D] projection output directly so the
+    backward can write dQ/dK/dV straight into ONE dqkv buffer (attn_bwd_out /
+    rope_fwd_out strided-out variants) — eliminating autograd's
+    split-backward torch.cat over the three grads (a full extra read+write
+    of the qkv-grad per layer, the CatArrayBatchedCopy kernel in profiles)."""
+
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, Hq, Hkv, D, traditional, causal, scale,
+                window, prefix_len, alibi_slopes):
+        B, S, _ = qkv.shape
+        scale = scale if scale is not None else 1.0 / math.sqrt(D)
+        q = qkv[..., : Hq * D].view(B, S, Hq, D)
+        k = qkv[..., Hq * D : (Hq + Hkv) * D].view(B, S, Hkv, D)
+        v = qkv[..., (Hq + Hkv) * D :].view(B, S, Hkv, D)
+        ext = get_ext()
+        qr = ext.rope_fwd(q, cos, sin, traditional, 0, False)
+        kr = ext.rope_fwd(k, cos, sin, traditional, 0, False)
+        mod = _mods_to_code(causal, window, prefix_len, alibi_slopes is not None)
+        modarg = int(window if window is not None
+                     else (prefix_len if prefix_len is not None else 0))
+        slopes = (alibi_slopes.float().contiguous() if alibi_slopes is not None
+                  else torch.empty(0, dtype=torch.float32, device=qkv.device))
+        o, lse = ext.attn_fwd(qr, kr, v, scale, mod, modarg, slopes)
+        ctx.save_for_backward(qkv, qr, kr, o, lse, cos, sin, slopes)
+        ctx.meta = (Hq, Hkv, D, traditional, scale, mod, modarg)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, qr, kr, o, lse, cos, sin, slopes = ctx.saved_tensors
+        Hq, Hkv, D, traditional, scale, mod, modarg = ctx.meta
+        B, S, _ = qkv.shape
+        ext = get_ext()
+        v = qkv[..., (Hq + Hkv) * D :].view(B, S, Hkv, D)
+        dqkv = torch.empty_like(qkv)
+        dq_t = dqkv[..., : Hq * D].view(B, S, Hq, D)
+        dk_t = dqkv[..., Hq * D : (Hq + Hkv) * D].view(B, S, Hkv, D)
+        dv_t = dqkv[..., (Hq + Hkv) * D :].view(B, S, Hkv, D)
+        none = torch.empty(0, dtype=qkv.dtype, device=qkv.device)
+        dq_r, dk_r, _ = ext.attn_bwd_out(qr, kr, v, o, do, lse, scale, mod,
+                                         modarg, slopes, none, none.clone(), dv_t)
+        ext.rope_fwd_out(dq_r, cos, sin, traditional, 0, True, dq_t)
+        ext.rope_fwd_out(dk_r, cos, sin, traditional, 0, True, dk_t)
+        return (dqkv,) + (None,) * 11
+
+
+def rope_flash_attention_qkv(
+    qkv: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    n_heads: int,
+    n_kv_heads: int,
+    head_dim: int,
+    traditional: bool = False,
+    causal: bool = True,
+    scale: Optional[float] = None,
+    window: Optional[int] = None,
+    prefix_len: Optional[int] = None,
+    alibi_slopes: Optional[torch.Tensor] = None,
+) -> torch.Tensor:
+    """qkv: [B, S, (Hq+2*Hkv)*D] fused projection output -> o [B, S, Hq, D].
+
+    HIP fast path (one fused autograd node, no grad cat); CPU falls back to
+    the differentiable composition of the same ops."""
+    if use_hip(qkv):
+        return _RopeAttnQKVFn.apply(qkv, cos, sin, n_heads, n_kv_heads, head_dim,
+                                    traditional, causal, scale, window,
+                                    prefix_len, alibi_slopes)
+    from .rope import apply_rope
+
+    B, S, _ = qkv.shape
+    Hq, Hkv, D = n_heads, n_kv_heads, head_dim
+    q, k, v = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    q = apply_rope(q.view(B, S, Hq, D), cos, sin, traditional, 0)
+    k = apply_rope(k.view(B, S, Hkv, D), cos, sin, traditional, 0)
+    return flash_attention(q, k, v.view(B, S, Hkv, D), causal=causal, scale=scale,
+                           window=window, prefix_len=prefix_len,
+                           alibi_slopes=alibi_slopes)
+
+
 def flex_attention(
     q: torch.Tensor,
     k: torch.Tensor,

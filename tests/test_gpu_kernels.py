@@ -251,3 +251,35 @@ def test_attn_bwd_gpu(ext, case):
         err = (got.float() - want).abs().max().item()
         ref_mag = want.abs().max().item()
         assert err < 0.05 * max(ref_mag, 1.0), f"{name} max err {err} (ref mag {ref_mag}) case {case}"
+
+
+def test_fused_qkv_rope_attention_matches_composition(ext):
+    """Fused qkv->RoPE->attention autograd node vs the op composition
+    (incl. dQKV written in place by attn_bwd_out/rope_fwd_out)."""
+    from mlx_cuda_distributed_pretraining_amd.ops.attention import (
+        flash_attention, rope_flash_attention_qkv,
+    )
+    from mlx_cuda_distributed_pretraining_amd.ops.rope import RopeTable, apply_rope
+
+    torch.manual_seed(0)
+    B, S, Hq, Hkv, D = 2, 256, 8, 4, 128
+    table = RopeTable(D, 10000.0)
+    cos, sin = table.get(S, dev(), 0)
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D, device=dev(),
+                      dtype=torch.bfloat16, requires_grad=True)
+
+    o = rope_flash_attention_qkv(qkv, cos, sin, Hq, Hkv, D)
+    do = torch.randn_like(o)
+    o.backward(do)
+    got_o, got_g = o.detach(), qkv.grad.clone()
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    q, k, v = qkv2.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    qr = apply_rope(q.view(B, S, Hq, D), cos, sin, False, 0)
+    kr = apply_rope(k.view(B, S, Hkv, D), cos, sin, False, 0)
+    o2 = flash_attention(qr, kr, v.view(B, S, Hkv, D), causal=True)
+    o2.backward(do)
+
+    assert torch.allclose(got_o.float(), o2.detach().float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(got_g.float(), qkv2.grad.float(), atol=5e-2, rtol=5e-2), \
+        (got_g.float() - qkv2.grad.float()).abs().max()
