@@ -23,12 +23,19 @@
 
 namespace {
 
-constexpr int NW = 8;
-constexpr int TPB = NW * WAVE;
-constexpr int QPB = 32 * NW;   // q rows per block (dQ kernel)
-constexpr int KPB = 32 * NW;   // kv rows per block (dK/dV kernels)
 constexpr int KVB = 32;        // kv tile per iteration (dQ kernel)
 constexpr float LOG2E = 1.4426950408889634f;
+
+// waves per block (8 or 16): 16-wave blocks halve staging traffic and
+// barrier frequency per unit of MFMA work at the same 4 waves/SIMD
+// (1 block/CU instead of 2). Env MCDP_ATTN_BWD_NW selects at launch.
+inline int bwd_nw() {
+  static int nw = []() {
+    const char* e = getenv("MCDP_ATTN_BWD_NW");
+    return e ? atoi(e) : 8;
+  }();
+  return nw;
+}
 
 // ---------------- preprocess: Drow = rowsum(dO * O) ----------------
 // 4 rows per wave: 16 lanes x 8 elements cover a D=128 row in one uint4 load.
@@ -75,14 +82,16 @@ __device__ __forceinline__ void q_range_for_kv(int kv0, int kpb, int q_off, int 
 }
 
 // ---------------- dQ kernel (block per q-tile of QPB rows) ----------------
-template <int D, int MOD>
-__global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
+template <int D, int MOD, int NW>
+__global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
     __hip_bfloat16* __restrict__ dq, const float* __restrict__ slopes,
     int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
     long q_rs, long k_rs, long v_rs, long do_rs, long dq_rs) {
+  constexpr int TPB = NW * WAVE;
+  constexpr int QPB = 32 * NW;
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K/V row-major images
@@ -294,14 +303,16 @@ __global__ __launch_bounds__(TPB) void attn_bwd_dq_kernel(
 
 // ---------------- dK / dV kernels (block per kv-tile, loop GQA group) -------
 // WANT_DK: true -> dK (needs dP: V frags + dO rm); false -> dV (needs dO^T).
-template <int D, int MOD, bool WANT_DK>
-__global__ __launch_bounds__(TPB) void attn_bwd_dkv_kernel(
+template <int D, int MOD, bool WANT_DK, int NW>
+__global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, const __hip_bfloat16* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
     __hip_bfloat16* __restrict__ dkv_out, const float* __restrict__ slopes,
     int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
     long q_rs, long k_rs, long v_rs, long do_rs, long dkv_rs) {
+  constexpr int TPB = NW * WAVE;
+  constexpr int KPB = 32 * NW;
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
   constexpr int KSWZ = (D >= 128) ? 15 : 7;  // row-major q/do images
@@ -553,7 +564,7 @@ store:
   }
 }
 
-template <int D, int MOD>
+template <int D, int MOD, int NW>
 void launch_bwd_all(dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
                     const __hip_bfloat16* q, const __hip_bfloat16* k, const __hip_bfloat16* v,
                     const __hip_bfloat16* dout, const float* lse, const float* drow,
@@ -561,12 +572,30 @@ void launch_bwd_all(dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
                     const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
                     float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs,
                     long dq_rs, long dk_rs, long dv_rs) {
-  attn_bwd_dq_kernel<D, MOD><<<gq, block, 0, stream>>>(
+  attn_bwd_dq_kernel<D, MOD, NW><<<gq, block, 0, stream>>>(
       q, k, v, dout, lse, drow, dq, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs);
-  attn_bwd_dkv_kernel<D, MOD, true><<<gkv, block, 0, stream>>>(
+  attn_bwd_dkv_kernel<D, MOD, true, NW><<<gkv, block, 0, stream>>>(
       q, k, v, dout, lse, drow, dk, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dk_rs);
-  attn_bwd_dkv_kernel<D, MOD, false><<<gkv, block, 0, stream>>>(
+  attn_bwd_dkv_kernel<D, MOD, false, NW><<<gkv, block, 0, stream>>>(
       q, k, v, dout, lse, drow, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dv_rs);
+}
+
+template <int D, int MOD>
+void launch_all_nw(dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
+                   const __hip_bfloat16* q, const __hip_bfloat16* k, const __hip_bfloat16* v,
+                   const __hip_bfloat16* dout, const float* lse, const float* drow,
+                   __hip_bfloat16* dq, __hip_bfloat16* dk, __hip_bfloat16* dv,
+                   const float* slopes, int B, int Sq, int Skv, int Hq, int Hkv,
+                   float scale, int modarg, long q_rs, long k_rs, long v_rs, long do_rs,
+                   long dq_rs, long dk_rs, long dv_rs) {
+  if (bwd_nw() == 16)
+    launch_bwd_all<D, MOD, 16>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv,
+                               slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs,
+                               do_rs, dq_rs, dk_rs, dv_rs);
+  else
+    launch_bwd_all<D, MOD, 8>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv,
+                              slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs,
+                              do_rs, dq_rs, dk_rs, dv_rs);
 }
 
 template <int D>
@@ -579,19 +608,19 @@ void launch_bwd_mod(int mod, dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
                     long dq_rs, long dk_rs, long dv_rs) {
   switch (mod) {
     case MOD_NONE:
-      launch_bwd_all<D, MOD_NONE>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
+      launch_all_nw<D, MOD_NONE>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_CAUSAL:
-      launch_bwd_all<D, MOD_CAUSAL>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
+      launch_all_nw<D, MOD_CAUSAL>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_SLIDING_WINDOW:
-      launch_bwd_all<D, MOD_SLIDING_WINDOW>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
+      launch_all_nw<D, MOD_SLIDING_WINDOW>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_PREFIX_LM:
-      launch_bwd_all<D, MOD_PREFIX_LM>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
+      launch_all_nw<D, MOD_PREFIX_LM>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     case MOD_ALIBI:
-      launch_bwd_all<D, MOD_ALIBI>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
+      launch_all_nw<D, MOD_ALIBI>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs, do_rs, dq_rs, dk_rs, dv_rs);
       break;
     default:
       TORCH_CHECK(false, "attn_bwd: unknown mod ", mod);
@@ -646,9 +675,10 @@ std::vector<at::Tensor> attn_bwd_out(at::Tensor q, at::Tensor k, at::Tensor v, a
         drow.data_ptr<float>(), rows, D, do_rs, Hq);
   }
 
-  dim3 gq(cdiv(Sq, QPB), Hq, B);
-  dim3 gkv(cdiv(Skv, KPB), Hkv, B);
-  dim3 block(TPB);
+  const int qpb = 32 * bwd_nw();
+  dim3 gq(cdiv(Sq, qpb), Hq, B);
+  dim3 gkv(cdiv(Skv, qpb), Hkv, B);
+  dim3 block(bwd_nw() * WAVE);
   const float* sl = slopes.numel() > 0 ? slopes.data_ptr<float>() : nullptr;
   auto* qp = reinterpret_cast<const __hip_bfloat16*>(q.data_ptr());
   auto* kp = reinterpret_cast<const __hip_bfloat16*>(k.data_ptr());

@@ -344,6 +344,7 @@ void launch_fwd_cfg(dim3 grid, dim3 block, hipStream_t stream,
   else if (nw == 8 && kvb == 32) LAUNCH(8, 32);
   else if (nw == 4 && kvb == 64) LAUNCH(4, 64);
   else if (nw == 4 && kvb == 32) LAUNCH(4, 32);
+  else if (nw == 16 && kvb == 64) LAUNCH(16, 64);
   else TORCH_CHECK(false, "attn_fwd: unsupported NW/KVB ", nw, "/", kvb);
 #undef LAUNCH
 }
