@@ -129,6 +129,10 @@ class LoggingConfig:
     log_samples: bool = False
     log_samples_count: int = 3
     max_snapshots: int = 0  # >0: rotate old step checkpoints (reference train.py:79-80)
+    # WebSocket stats mesh (reference stats_server/stats_client):
+    # e.g. "ws://127.0.0.1:8765/ws" — each rank registers as rank<N> and
+    # streams per-step metrics.
+    stats_url: Optional[str] = None
 
 
 @dataclass

@@ -423,6 +423,24 @@ class Trainer:
         ckpt_interval = int(steps_cfg.get("checkpoint_interval", 0))
         val_interval = int(steps_cfg.get("validation_interval", 0))
 
+        # optional WebSocket stats mesh (reference stats_server/stats_client)
+        stats_client = None
+        stats_collector = None
+        if cfg.logging.stats_url:
+            try:
+                from ..utils.stats_client import StatsClient, WorkerMetricsCollector
+
+                stats_client = StatsClient(
+                    cfg.logging.stats_url, worker_id=f"rank{self.rank}",
+                    info={"run": cfg.name, "world_size": self.world_size},
+                )
+                stats_client.start()
+                stats_collector = WorkerMetricsCollector(
+                    f"rank{self.rank}", device=self.local_rank
+                )
+            except Exception as e:  # stats are best-effort
+                self.logger.log(f"stats mesh unavailable: {e}")
+
         self.current_step = 0
         if cfg.resume is not None:
             self.load_checkpoint(
@@ -438,6 +456,7 @@ class Trainer:
 
         self.model.train()
         start_time = time.time()
+        step_t0 = time.time()
         val_loss = None
         stop = False
         for step in range(self.start_step, self.total_steps):
@@ -449,6 +468,13 @@ class Trainer:
             loss_v = float(loss_g.item())
             ntok_v = int(ntok_g.item())
             self.total_tokens += ntok_v
+
+            if stats_collector is not None:
+                now = time.time()
+                stats_client.send_stats(
+                    stats_collector.update(step + 1, loss_v, ntok_v, now - step_t0)
+                )
+                step_t0 = now
 
             if val_interval and (step + 1) % val_interval == 0:
                 val_loss = self.validate()
@@ -496,4 +522,6 @@ class Trainer:
             f"Training complete: {self.total_tokens} tokens, final val_loss="
             f"{final_val if final_val is not None else float('nan')}"
         )
+        if stats_client is not None:
+            stats_client.stop()
         self.logger.close()

@@ -147,3 +147,54 @@ def test_metrics_collector():
     assert s2["tokens_total"] == 2000
     assert s1["tokens_per_sec"] == pytest.approx(2000.0)
     assert s2["tokens_per_sec_avg"] == pytest.approx(2000.0)
+
+
+@pytest.mark.timeout(120)
+def test_trainer_streams_to_stats_mesh(tmp_path):
+    """End-to-end: a Trainer with logging.stats_url streams per-step metrics
+    to a live StatsServer."""
+    import threading
+    from pathlib import Path
+
+    from mlx_cuda_distributed_pretraining_amd.core.config import Config
+    from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer
+    from mlx_cuda_distributed_pretraining_amd.utils.stats_server import StatsServer
+
+    server = StatsServer(port=18766)
+    loop_box = {}
+
+    def run_server():
+        async def main():
+            await server.start()
+            while not loop_box.get("stop"):
+                await asyncio.sleep(0.1)
+            await server.stop()
+
+        asyncio.run(main())
+
+    t = threading.Thread(target=run_server, daemon=True)
+    t.start()
+    time.sleep(0.5)
+
+    repo = Path(__file__).resolve().parent.parent
+    cfg = Config.from_yaml(repo / "configs" / "model-config-sample.yaml")
+    cfg.name = "stats-run"
+    cfg.overwrite = True
+    cfg.data.synthetic = True
+    cfg.training.hyperparameters["iters"] = 3
+    cfg.training.hyperparameters["batch_size"] = 2
+    cfg.data.preprocessing["max_context_size"] = 32
+    cfg.logging.steps = {"logging_interval": 1, "checkpoint_interval": 0,
+                         "validation_interval": 0}
+    cfg.logging.stats_url = "ws://127.0.0.1:18766/ws"
+    trainer = Trainer(cfg, runs_root=str(tmp_path / "runs"))
+    trainer.train()
+    for _ in range(50):
+        if len(server.history.get("rank0", [])) >= 3:
+            break
+        time.sleep(0.1)
+    loop_box["stop"] = True
+    t.join(5)
+    hist = server.history.get("rank0", [])
+    assert len(hist) >= 3
+    assert hist[-1]["step"] == 3 and "loss" in hist[-1]
