@@ -49,3 +49,45 @@ def test_all_baseline_configs_parse():
     ]:
         cfg = Config.from_yaml(str(REPO / "configs" / f"{name}.yaml"))
         assert cfg.name
+
+
+def test_config_extends_deep_merge(tmp_path):
+    """extends: base.yaml — child sections deep-merge over the base."""
+    from mlx_cuda_distributed_pretraining_amd.core.config import Config
+
+    (tmp_path / "base.yaml").write_text("""
+name: base
+data:
+  synthetic: true
+  preprocessing: {max_context_size: 512, chunk_overlap: 0}
+model:
+  architecture: llama
+  dimensions: {hidden_size: 64, intermediate_size: 128, num_layers: 2}
+training:
+  hyperparameters: {batch_size: 4, learning_rate: 1.0e-3, iters: 10}
+""")
+    (tmp_path / "child.yaml").write_text("""
+extends: base.yaml
+name: child
+training:
+  hyperparameters: {learning_rate: 5.0e-4}
+""")
+    cfg = Config.from_yaml(str(tmp_path / "child.yaml"))
+    assert cfg.name == "child"
+    # overridden leaf
+    assert cfg.training.hyperparameters["learning_rate"] == 5.0e-4
+    # inherited siblings survive the deep merge
+    assert cfg.training.hyperparameters["batch_size"] == 4
+    assert cfg.model.dimensions["hidden_size"] == 64
+    assert cfg.data.preprocessing["max_context_size"] == 512
+
+
+def test_config_extends_cycle_guard(tmp_path):
+    from mlx_cuda_distributed_pretraining_amd.core.config import Config
+
+    (tmp_path / "a.yaml").write_text("extends: b.yaml\nname: a\n")
+    (tmp_path / "b.yaml").write_text("extends: a.yaml\nname: b\n")
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError, match="extends"):
+        Config.from_yaml(str(tmp_path / "a.yaml"))
