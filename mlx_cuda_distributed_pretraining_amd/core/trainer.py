@@ -196,7 +196,9 @@ class Trainer:
         self.flat_space: Optional[FlatParamSpace] = None
         self.ddp: Optional[DataParallelGrads] = None
         if self.use_fused:
-            self.flat_space = FlatParamSpace(self.model)
+            # copy-mode grads: no autograd read-modify-write add per param,
+            # no per-step flat zero_ (FusedFlatAdamW reads flat_grad only)
+            self.flat_space = FlatParamSpace(self.model, grad_mode="copy")
             zero1 = cfg.system.zero_optimization_level >= 1
             if not zero1:
                 self.ddp = DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb)

@@ -130,3 +130,50 @@ def test_zero1_step_runs_and_syncs():
         p.join(120)
         assert p.exitcode == 0
     assert torch.isfinite(flat).all()
+
+
+def test_flat_copy_mode_matches_views_mode():
+    """grad_mode='copy' must produce the same flat_grad as 'views' (incl.
+    gradient-accumulation add on micro-step > 0)."""
+    from mlx_cuda_distributed_pretraining_amd.parallel.flat import FlatParamSpace
+
+    def run(mode):
+        torch.manual_seed(0)
+        model = Model(_tiny_args())
+        space = FlatParamSpace(model, grad_mode=mode)
+        batch = _make_batches()
+        for micro in range(2):  # two micro-steps: copy then accumulate
+            logits = model(batch[micro * 2 : micro * 2 + 2, :-1])
+            loss = torch.nn.functional.cross_entropy(
+                logits.reshape(-1, 67), batch[micro * 2 : micro * 2 + 2, 1:].reshape(-1)
+            )
+            loss.backward()
+        return space.flat_grad.clone()
+
+    g_views = run("views")
+    g_copy = run("copy")
+    assert torch.allclose(g_views, g_copy, atol=1e-6), (g_views - g_copy).abs().max()
+
+
+def test_flat_copy_mode_second_step_fresh():
+    """zero_grad() in copy mode resets the seen-set: the next step's grads
+    REPLACE (not accumulate onto) the previous step's."""
+    from mlx_cuda_distributed_pretraining_amd.parallel.flat import FlatParamSpace
+
+    torch.manual_seed(0)
+    model = Model(_tiny_args())
+    space = FlatParamSpace(model, grad_mode="copy")
+    batch = _make_batches()
+
+    def one_step():
+        space.zero_grad()
+        logits = model(batch[:2, :-1])
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, 67), batch[:2, 1:].reshape(-1)
+        )
+        loss.backward()
+        return space.flat_grad.clone()
+
+    g1 = one_step()
+    g2 = one_step()
+    assert torch.allclose(g1, g2, atol=1e-6)
