@@ -461,6 +461,28 @@ class Trainer:
         step_t0 = time.time()
         val_loss = None
         stop = False
+        try:
+            self._train_loop(cfg, log_interval, ckpt_interval, val_interval,
+                             start_time, step_t0, val_loss, stop,
+                             stats_client, stats_collector)
+        except Exception as e:
+            # Failure recovery: on any hot-loop failure
+            # (RCCL timeout/abort included — init_distributed sets a
+            # collective timeout), try to save an emergency checkpoint so
+            # the run can resume from the last completed step.
+            self.logger.log(f"FATAL at step {self.current_step}: {e!r}; "
+                            f"saving emergency checkpoint")
+            try:
+                self.save_checkpoint(f"emergency_{self.current_step}")
+            except Exception as e2:  # a dead process group must not mask e
+                self.logger.log(f"emergency checkpoint failed: {e2!r}")
+            if stats_client is not None:
+                stats_client.stop()
+            raise
+
+    def _train_loop(self, cfg, log_interval, ckpt_interval, val_interval,
+                    start_time, step_t0, val_loss, stop,
+                    stats_client, stats_collector) -> None:
         for step in range(self.start_step, self.total_steps):
             loss, ntok = self.train_step(step)
             self.current_step = step + 1  # completed steps

@@ -128,3 +128,33 @@ def test_epochs_mode(tmp_runs):
     trainer = Trainer(cfg, runs_root=tmp_runs)
     assert trainer.total_steps == trainer.steps_per_epoch * 2
     trainer.train()
+
+
+def test_emergency_checkpoint_on_failure(tmp_runs):
+    """Hot-loop failure -> emergency checkpoint saved, exception re-raised
+    (SURVEY.md §5.3 recovery story)."""
+    from pathlib import Path
+
+    cfg = small_cfg(tmp_runs)
+    cfg.name = "crash-run"
+    cfg.overwrite = True
+    cfg.data.synthetic = True
+    cfg.training.hyperparameters["iters"] = 5
+    trainer = Trainer(cfg, runs_root=tmp_runs)
+
+    calls = {"n": 0}
+    orig = trainer.train_step
+
+    def exploding(step):
+        calls["n"] += 1
+        if calls["n"] >= 3:
+            raise RuntimeError("simulated rank failure")
+        return orig(step)
+
+    trainer.train_step = exploding
+    import pytest as _pytest
+
+    with _pytest.raises(RuntimeError, match="simulated rank failure"):
+        trainer.train()
+    ckpts = list((Path(tmp_runs) / "crash-run" / "checkpoints").glob("step_emergency_*_model.safetensors"))
+    assert ckpts, "no emergency checkpoint written"
