@@ -53,11 +53,19 @@ def require_ext():
     return ext
 
 
-def use_hip(*tensors: torch.Tensor) -> bool:
-    """True when the op should dispatch to the HIP kernel path."""
+def use_hip(*tensors: torch.Tensor,
+            dtypes=(torch.bfloat16, torch.float32)) -> bool:
+    """True when the op should dispatch to the HIP kernel path.
+
+    ``dtypes``: dtypes the kernel supports — anything else (e.g. fp16 under
+    ``system.precision: float16``) runs the torch reference composition,
+    which is correct on GPU, just not the tuned path."""
     if not tensors:
         return False
     if not all(t.is_cuda for t in tensors if isinstance(t, torch.Tensor)):
+        return False
+    if any(t.dtype not in dtypes for t in tensors
+           if isinstance(t, torch.Tensor) and t.is_floating_point()):
         return False
     if os.environ.get("MCDP_FORCE_TORCH") == "1":
         return False

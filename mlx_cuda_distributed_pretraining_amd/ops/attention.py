@@ -113,7 +113,7 @@ class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal, scale, window, prefix_len, alibi_slopes):
         scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
-        if use_hip(q, k, v):
+        if use_hip(q, k, v, dtypes=(torch.bfloat16,)):
             ext = get_ext()
             mod = _mods_to_code(causal, window, prefix_len, alibi_slopes is not None)
             modarg = int(
@@ -253,7 +253,7 @@ def rope_flash_attention_qkv(
 
     HIP fast path (one fused autograd node, no grad cat); CPU falls back to
     the differentiable composition of the same ops."""
-    if use_hip(qkv):
+    if use_hip(qkv, dtypes=(torch.bfloat16,)):
         return _RopeAttnQKVFn.apply(qkv, cos, sin, n_heads, n_kv_heads, head_dim,
                                     traditional, causal, scale, window,
                                     prefix_len, alibi_slopes)
