@@ -40,6 +40,14 @@ std::vector<at::Tensor> attn_bwd_out(at::Tensor q, at::Tensor k, at::Tensor v, a
 // sampling.hip
 at::Tensor sample_token(at::Tensor logits, double temperature, double top_p, double min_p,
                         long seed);
+// decode.hip (hipGraph-capturable static decode)
+void rope_decode_(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
+                  at::Tensor pos);
+void kv_append_(at::Tensor k, at::Tensor v, at::Tensor kc, at::Tensor vc, at::Tensor pos);
+at::Tensor attn_decode(at::Tensor q, at::Tensor kc, at::Tensor vc, at::Tensor pos,
+                       at::Tensor part, double scale);
+void pos_incr_(at::Tensor pos);
+void write_token_(at::Tensor tok, at::Tensor ring, at::Tensor idx);
 // debug.hip
 at::Tensor mfma_tile_test(at::Tensor A, at::Tensor B);
 at::Tensor afrag_transform_test(at::Tensor M);
@@ -62,6 +70,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_out", &attn_bwd_out,
         "flash attention backward into strided out views (fused dQKV)");
   m.def("sample_token", &sample_token, "fused temperature/min-p sampling");
+  m.def("rope_decode_", &rope_decode_, "in-place RoPE at device position");
+  m.def("kv_append_", &kv_append_, "append k/v into the static cache at device position");
+  m.def("attn_decode", &attn_decode, "split-KV decode attention over the static cache");
+  m.def("pos_incr_", &pos_incr_, "device position += 1");
+  m.def("write_token_", &write_token_, "record token into the device ring");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");
 }

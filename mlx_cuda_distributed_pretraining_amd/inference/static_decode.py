@@ -1,0 +1,188 @@
+"""hipGraph-captured decode: static shapes, device-side position.
+
+The eager decode path launches ~400 kernels per token (22 layers x ~18 ops),
+which makes 1B decode launch-bound (~5.3 ms/token measured). Here the WHOLE
+decode step — embedding, all layers (RoPE at device position, KV append,
+split-KV decode attention, GEMMs), final norm, lm head, greedy argmax, token
+ring write, position increment — is captured ONCE into a hipGraph and
+replayed per token with zero host work.
+
+Design points (MI355X):
+  - the sequence length lives in a device int32 (`pos`); every decode kernel
+    reads it from memory, so one captured graph serves all positions,
+  - KV caches are preallocated [B, Lmax, Hkv, D] bf16 (288 GB HBM3E: a 1B
+    model's full-context cache is ~0.2 GB — capacity is never the issue),
+  - attention is a two-phase split-KV flash-decode (csrc/decode.hip):
+    chunk partials with own softmax stats, then a combine pass — fills the
+    chip even at batch 1,
+  - the generated token feeds back through a static input buffer inside the
+    graph, and lands in a device ring indexed by pos, so N tokens = N
+    graph replays + ONE host sync to read the ring.
+
+CPU (and non-bf16) fall back to an eager composition of the same ops —
+used by the unit tests as the numerics oracle.
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+
+from ..ops._ext import get_ext, use_hip
+
+
+class StaticLayerCache:
+    """Per-layer handle; Attention.forward delegates to ``attend``."""
+
+    def __init__(self, owner: "StaticDecodeState", idx: int):
+        self.owner = owner
+        self.idx = idx
+
+    # duck-typing marker checked by models.llama.Attention
+    static_decode = True
+
+    @property
+    def offset(self) -> int:
+        return int(self.owner.pos.item())
+
+    def attend(self, attn, qkv: torch.Tensor) -> torch.Tensor:
+        """qkv: [B, 1, (Hq+2*Hkv)*D] -> o [B, 1, Hq, D] (post-attention,
+        pre-wo). Applies RoPE at the device position, appends k/v to the
+        static cache, runs decode attention."""
+        ow = self.owner
+        B = qkv.shape[0]
+        Hq, Hkv, D = attn.n_heads, attn.n_kv_heads, attn.head_dim
+        q = qkv[..., : Hq * D].view(B, 1, Hq, D).contiguous()
+        k = qkv[..., Hq * D : (Hq + Hkv) * D].view(B, 1, Hkv, D).contiguous()
+        v = qkv[..., (Hq + Hkv) * D :].view(B, 1, Hkv, D).contiguous()
+        kc, vc = ow.k[self.idx], ow.v[self.idx]
+        trad = attn.args.rope_traditional
+
+        if use_hip(qkv, dtypes=(torch.bfloat16,)):
+            ext = get_ext()
+            ext.rope_decode_(q, ow.cos, ow.sin, trad, ow.pos)
+            ext.rope_decode_(k, ow.cos, ow.sin, trad, ow.pos)
+            ext.kv_append_(k, v, kc, vc, ow.pos)
+            return ext.attn_decode(q, kc, vc, ow.pos, ow.part, attn.scale)
+
+        # eager reference (CPU / non-bf16): same semantics
+        from ..ops.attention import attention_ref
+        from ..ops.rope import rope_ref
+
+        p = int(ow.pos.item())
+        q = rope_ref(q, ow.cos, ow.sin, trad, offset=p)
+        k = rope_ref(k, ow.cos, ow.sin, trad, offset=p)
+        kc[:, p : p + 1] = k
+        vc[:, p : p + 1] = v
+        return attention_ref(q, kc[:, : p + 1], vc[:, : p + 1], causal=True,
+                             scale=attn.scale)
+
+
+class StaticDecodeState:
+    """Preallocated caches + device position + per-layer handles."""
+
+    def __init__(self, model, batch: int = 1, max_len: int = 2048):
+        args = model.args
+        self.model = model
+        self.max_len = max_len
+        dev = next(model.parameters()).device
+        dtype = next(model.parameters()).dtype
+        L = len(model.layers)
+        Hq, Hkv, D = args.num_heads, args.num_kv_heads, args.head_dim
+        self.k = torch.zeros(L, batch, max_len, Hkv, D, device=dev, dtype=dtype)
+        self.v = torch.zeros_like(self.k)
+        self.pos = torch.zeros(1, dtype=torch.int32, device=dev)
+        nc = (max_len + 255) // 256
+        self.part = torch.empty(batch, Hq, nc, D + 2, dtype=torch.float32, device=dev)
+        cos, sin = model.layers[0].attention.rope_table.get(max_len, dev, 0)
+        self.cos, self.sin = cos.contiguous(), sin.contiguous()
+        self.layers: List[StaticLayerCache] = [StaticLayerCache(self, i) for i in range(L)]
+
+
+class GraphDecoder:
+    """Greedy decoder with (on GPU) hipGraph capture of the whole step."""
+
+    def __init__(self, model, batch: int = 1, max_len: int = 2048):
+        self.model = model
+        self.batch = batch
+        self.max_len = max_len
+        self.state = StaticDecodeState(model, batch, max_len)
+        dev = next(model.parameters()).device
+        self.dev = dev
+        self.on_gpu = dev.type == "cuda"
+        self.in_tok = torch.zeros(batch, 1, dtype=torch.long, device=dev)
+        self.ring = torch.zeros(max_len, batch, dtype=torch.long, device=dev)
+        self.graph = None
+
+    @torch.no_grad()
+    def prefill(self, prompt_tokens: torch.Tensor) -> None:
+        """prompt_tokens: [B, P]. Runs the normal (tiled-attention) path,
+        bulk-copies K/V into the static cache, sets pos=P, stages the first
+        generated token into the static input buffer."""
+        from ..models.llama import make_prompt_cache
+
+        model = self.model
+        P = prompt_tokens.shape[1]
+        assert P + 1 < self.max_len, "prompt too long for max_len"
+        cache = make_prompt_cache(model)
+        logits = model(prompt_tokens.to(self.dev), cache=cache)
+        for i, c in enumerate(cache):
+            self.state.k[i][:, :P] = c.k
+            self.state.v[i][:, :P] = c.v
+        self.state.pos.fill_(P)
+        first = logits[:, -1].argmax(-1)
+        self.in_tok.copy_(first.unsqueeze(1))
+        self.ring[P % self.max_len] = first
+
+    @torch.no_grad()
+    def _step(self) -> None:
+        logits = self.model(self.in_tok, cache=self.state.layers)
+        nxt = logits[:, -1].argmax(-1)
+        # advance pos, then record: the new token lives at ring[pos % cap]
+        if self.on_gpu:
+            ext = get_ext()
+            ext.pos_incr_(self.state.pos)
+            ext.write_token_(nxt, self.ring, self.state.pos)
+        else:
+            self.state.pos += 1
+            self.ring[int(self.state.pos.item()) % self.max_len] = nxt
+        self.in_tok.copy_(nxt.unsqueeze(1))
+
+    @torch.no_grad()
+    def capture(self) -> None:
+        """Capture one decode step into a hipGraph (GPU only)."""
+        if not self.on_gpu:
+            return
+        save_pos = self.state.pos.clone()
+        save_tok = self.in_tok.clone()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):  # warmup on a side stream
+            for _ in range(2):
+                self._step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.state.pos.copy_(save_pos)
+        self.in_tok.copy_(save_tok)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self._step()
+        # capture advanced nothing for real (replay does), but the capture
+        # itself ran once logically — reset state again
+        self.state.pos.copy_(save_pos)
+        self.in_tok.copy_(save_tok)
+
+    @torch.no_grad()
+    def decode(self, n_tokens: int) -> torch.Tensor:
+        """Generate n_tokens greedily. Returns [B, n_tokens] (device)."""
+        p0 = int(self.state.pos.item())
+        assert p0 + n_tokens < self.max_len, "decode would exceed max_len"
+        if self.graph is not None:
+            for _ in range(n_tokens):
+                self.graph.replay()
+        else:
+            for _ in range(n_tokens):
+                self._step()
+        idx = torch.arange(p0, p0 + n_tokens, device=self.dev) % self.max_len
+        out = self.ring[idx]  # [n, B]
+        return out.t().contiguous()

@@ -145,6 +145,14 @@ class Attention(nn.Module):
         B, S, _ = x.shape
         qkv = self.wqkv(x)
 
+        if getattr(cache, "static_decode", False):
+            # hipGraph-capturable decode step (inference/static_decode.py):
+            # RoPE at device position + static-cache append + split-KV
+            # decode attention, all shape-static
+            assert S == 1, "static decode processes one token at a time"
+            o = cache.attend(self, qkv)
+            return self.wo(o.reshape(B, S, -1))
+
         atype = self.args.attention_type
         if cache is None and atype != "simple":
             # training fast path: one fused autograd node over the fused QKV
