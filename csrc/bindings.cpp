@@ -48,6 +48,8 @@ at::Tensor attn_decode(at::Tensor q, at::Tensor kc, at::Tensor vc, at::Tensor po
                        at::Tensor part, double scale);
 void pos_incr_(at::Tensor pos);
 void write_token_(at::Tensor tok, at::Tensor ring, at::Tensor idx);
+// gemv.hip
+at::Tensor gemv_bf16(at::Tensor x, at::Tensor W);
 // debug.hip
 at::Tensor mfma_tile_test(at::Tensor A, at::Tensor B);
 at::Tensor afrag_transform_test(at::Tensor M);
@@ -75,6 +77,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode, "split-KV decode attention over the static cache");
   m.def("pos_incr_", &pos_incr_, "device position += 1");
   m.def("write_token_", &write_token_, "record token into the device ring");
+  m.def("gemv_bf16", &gemv_bf16, "bf16 GEMV (decode projections)");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");
 }

@@ -26,6 +26,7 @@ import torch
 import torch.nn as nn
 
 from ..ops.attention import attention_ref, flash_attention, rope_flash_attention_qkv
+from ..ops.gemv import FastLinear, linear_fast
 from ..ops.rmsnorm import RMSNorm
 from ..ops.rope import RopeTable, apply_rope
 from ..ops.swiglu import swiglu
@@ -130,8 +131,8 @@ class Attention(nn.Module):
         self.head_dim = args.head_dim
         self.scale = self.head_dim**-0.5
         qkv_out = (self.n_heads + 2 * self.n_kv_heads) * self.head_dim
-        self.wqkv = nn.Linear(args.hidden_size, qkv_out, bias=args.attention_bias)
-        self.wo = nn.Linear(self.n_heads * self.head_dim, args.hidden_size, bias=args.attention_bias)
+        self.wqkv = FastLinear(args.hidden_size, qkv_out, bias=args.attention_bias)
+        self.wo = FastLinear(self.n_heads * self.head_dim, args.hidden_size, bias=args.attention_bias)
         self.rope_table = rope_table
         if args.use_alibi:
             slopes = torch.tensor(
@@ -210,8 +211,8 @@ class MLP(nn.Module):
 
     def __init__(self, args: ModelArgs):
         super().__init__()
-        self.w_gate_up = nn.Linear(args.hidden_size, 2 * args.intermediate_size, bias=args.mlp_bias)
-        self.w_down = nn.Linear(args.intermediate_size, args.hidden_size, bias=args.mlp_bias)
+        self.w_gate_up = FastLinear(args.hidden_size, 2 * args.intermediate_size, bias=args.mlp_bias)
+        self.w_down = FastLinear(args.intermediate_size, args.hidden_size, bias=args.mlp_bias)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.w_down(swiglu(self.w_gate_up(x)))
@@ -255,7 +256,7 @@ class Model(nn.Module):
         )
         self.norm = RMSNorm(args.hidden_size, args.rms_norm_eps)
         if not args.tie_word_embeddings:
-            self.output = nn.Linear(args.hidden_size, args.vocab_size, bias=False)
+            self.output = FastLinear(args.hidden_size, args.vocab_size, bias=False)
         self.apply(self._init_weights)
 
     def _init_weights(self, module: nn.Module) -> None:
@@ -274,7 +275,7 @@ class Model(nn.Module):
             x = layer(x, cache[i] if cache is not None else None)
         x = self.norm(x)
         if self.args.tie_word_embeddings:
-            logits = x @ self.tok_embeddings.weight.t()
+            logits = linear_fast(x, self.tok_embeddings.weight)
         else:
             logits = self.output(x)
         if self.args.logit_scale:

@@ -332,3 +332,15 @@ def test_attn_decode_kernel_numerics(ext):
         want = attention_ref(q, kc[:, :length], vc[:, :length], causal=True)
         assert torch.allclose(o.float(), want.float(), atol=3e-2, rtol=3e-2), \
             (length, (o.float() - want.float()).abs().max())
+
+
+def test_gemv_matches_matmul(ext):
+    torch.manual_seed(0)
+    for B, N, K in [(1, 2560, 2048), (1, 2048, 5632), (2, 512, 768), (1, 32003, 2048)]:
+        x = torch.randn(B, 1, K, device=dev(), dtype=torch.bfloat16)
+        W = torch.randn(N, K, device=dev(), dtype=torch.bfloat16)
+        y = ext.gemv_bf16(x, W)
+        want = torch.nn.functional.linear(x, W)
+        assert y.shape == want.shape
+        assert torch.allclose(y.float(), want.float(), atol=2e-1, rtol=2e-2), \
+            (N, K, (y.float() - want.float()).abs().max())
