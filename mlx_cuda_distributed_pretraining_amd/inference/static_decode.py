@@ -53,18 +53,30 @@ class StaticLayerCache:
         ow = self.owner
         B = qkv.shape[0]
         Hq, Hkv, D = attn.n_heads, attn.n_kv_heads, attn.head_dim
-        q = qkv[..., : Hq * D].view(B, 1, Hq, D).contiguous()
-        k = qkv[..., Hq * D : (Hq + Hkv) * D].view(B, 1, Hkv, D).contiguous()
-        v = qkv[..., (Hq + Hkv) * D :].view(B, 1, Hkv, D).contiguous()
         kc, vc = ow.k[self.idx], ow.v[self.idx]
         trad = attn.args.rope_traditional
 
         if use_hip(qkv, dtypes=(torch.bfloat16,)):
             ext = get_ext()
-            ext.rope_decode_(q, ow.cos, ow.sin, trad, ow.pos)
-            ext.rope_decode_(k, ow.cos, ow.sin, trad, ow.pos)
+            qkv = qkv.contiguous()
+            qk = qkv[..., : (Hq + Hkv) * D].view(B, 1, Hq + Hkv, D)
+            if qk.is_contiguous():  # B == 1: q and k are ADJACENT in the fused
+                # projection — one in-place RoPE launch covers both
+                ext.rope_decode_(qk, ow.cos, ow.sin, trad, ow.pos)
+                q = qkv[..., : Hq * D].view(B, 1, Hq, D)
+                k = qkv[..., Hq * D : (Hq + Hkv) * D].view(B, 1, Hkv, D).contiguous()
+            else:
+                q = qkv[..., : Hq * D].view(B, 1, Hq, D).contiguous()
+                k = qkv[..., Hq * D : (Hq + Hkv) * D].view(B, 1, Hkv, D).contiguous()
+                ext.rope_decode_(q, ow.cos, ow.sin, trad, ow.pos)
+                ext.rope_decode_(k, ow.cos, ow.sin, trad, ow.pos)
+            v = qkv[..., (Hq + Hkv) * D :].view(B, 1, Hkv, D).contiguous()
             ext.kv_append_(k, v, kc, vc, ow.pos)
             return ext.attn_decode(q, kc, vc, ow.pos, ow.part, attn.scale)
+
+        q = qkv[..., : Hq * D].view(B, 1, Hq, D).contiguous()
+        k = qkv[..., Hq * D : (Hq + Hkv) * D].view(B, 1, Hkv, D).contiguous()
+        v = qkv[..., (Hq + Hkv) * D :].view(B, 1, Hkv, D).contiguous()
 
         # eager reference (CPU / non-bf16): same semantics
         from ..ops.attention import attention_ref
