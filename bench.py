@@ -56,6 +56,11 @@ def main() -> None:
         cfg.training.hyperparameters["batch_size"] = args.batch_size
     if args.seq_len:
         cfg.data.preprocessing["max_context_size"] = args.seq_len
+        # long-context runs: don't let the trainer truncate to the config's
+        # max_position_embeddings (the tiled attention is O(S) in memory)
+        attn = cfg.model.attention or {}
+        if attn.get("max_position_embeddings") and attn["max_position_embeddings"] < args.seq_len:
+            attn["max_position_embeddings"] = args.seq_len
     cfg.training.hyperparameters["iters"] = args.steps + args.warmup + 1
 
     import tempfile

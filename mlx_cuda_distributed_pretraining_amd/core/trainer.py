@@ -402,6 +402,21 @@ class Trainer:
                 w = csv.writer(f)
                 w.writerow(["lr", "loss"])
                 w.writerows(records)
+            try:  # loss-vs-lr plot (reference core/training.py:717-746)
+                import matplotlib
+
+                matplotlib.use("Agg")
+                import matplotlib.pyplot as plt
+
+                fig, ax = plt.subplots(figsize=(8, 5))
+                ax.plot([r[0] for r in records], [r[1] for r in records], "o-")
+                ax.axvline(best_lr, color="r", ls="--", label=f"suggested {best_lr:.2e}")
+                ax.set_xscale("log"); ax.set_xlabel("learning rate"); ax.set_ylabel("loss")
+                ax.legend(); ax.grid(alpha=0.3)
+                fig.savefig(self.run_dir / "lr_finder.png", dpi=120)
+                plt.close(fig)
+            except Exception:
+                pass
             self.logger.log(f"LR finder suggestion: {best_lr:.3e}")
         return {"suggested_lr": best_lr, "records": records}
 
