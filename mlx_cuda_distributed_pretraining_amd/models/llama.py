@@ -54,6 +54,7 @@ class ModelArgs:
     attention_window: Optional[int] = None  # sliding-window size (flex)
     attention_prefix_len: Optional[int] = None  # prefix-LM split (flex)
     use_alibi: bool = False
+    fp8: bool = False  # opt-in e4m3 GEMMs for the block projections (ops/fp8.py)
     # MoE knobs exist in the reference config (models/llama.py:40-41) but no
     # MoE layer is implemented there; kept for config parity.
     num_local_experts: int = 0
@@ -89,6 +90,7 @@ class ModelArgs:
             mlp_bias=bool(misc.get("mlp_bias", False)),
             tie_word_embeddings=bool(misc.get("tie_word_embeddings", True)),
             logit_scale=misc.get("logit_scale"),
+            fp8=bool(misc.get("fp8", False)),
             attention_type=str(attn.get("type", "flash")),
             attention_window=attn.get("window"),
             attention_prefix_len=attn.get("prefix_len"),
@@ -133,6 +135,7 @@ class Attention(nn.Module):
         qkv_out = (self.n_heads + 2 * self.n_kv_heads) * self.head_dim
         self.wqkv = FastLinear(args.hidden_size, qkv_out, bias=args.attention_bias)
         self.wo = FastLinear(self.n_heads * self.head_dim, args.hidden_size, bias=args.attention_bias)
+        self.wqkv.fp8 = self.wo.fp8 = args.fp8
         self.rope_table = rope_table
         if args.use_alibi:
             slopes = torch.tensor(
@@ -213,6 +216,7 @@ class MLP(nn.Module):
         super().__init__()
         self.w_gate_up = FastLinear(args.hidden_size, 2 * args.intermediate_size, bias=args.mlp_bias)
         self.w_down = FastLinear(args.intermediate_size, args.hidden_size, bias=args.mlp_bias)
+        self.w_gate_up.fp8 = self.w_down.fp8 = args.fp8
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.w_down(swiglu(self.w_gate_up(x)))

@@ -27,7 +27,22 @@ def linear_fast(x: torch.Tensor, weight: torch.Tensor,
 
 
 class FastLinear(torch.nn.Linear):
-    """nn.Linear that routes tiny-M inference matmuls to the GEMV kernel."""
+    """nn.Linear that routes tiny-M inference matmuls to the GEMV kernel and
+    (when ``fp8`` is set by the model builder) big training matmuls to the
+    e4m3 _scaled_mm path (ops/fp8.py)."""
+
+    fp8: bool = False
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if (
+            self.fp8
+            and self.bias is None
+            and torch.is_grad_enabled()
+            and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and x.shape[:-1].numel() >= 16
+        ):
+            from .fp8 import fp8_linear
+
+            return fp8_linear(x, self.weight)
         return linear_fast(x, self.weight, self.bias)
