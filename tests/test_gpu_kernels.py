@@ -354,7 +354,7 @@ def test_fp8_linear_gpu_matches_bf16(ext):
     torch.manual_seed(0)
     M, K, N = 512, 2048, 2560
     x = torch.randn(M, K, device=dev(), dtype=torch.bfloat16, requires_grad=True)
-    w = torch.randn(N, K, device=dev(), dtype=torch.bfloat16, requires_grad=True) * 0.02
+    w = (torch.randn(N, K, device=dev(), dtype=torch.bfloat16) * 0.02).requires_grad_(True)
     y = fp8_linear(x, w)
     dy = torch.randn_like(y)
     y.backward(dy)
