@@ -50,6 +50,8 @@ void pos_incr_(at::Tensor pos);
 void write_token_(at::Tensor tok, at::Tensor ring, at::Tensor idx);
 // gemv.hip
 at::Tensor gemv_bf16(at::Tensor x, at::Tensor W);
+// fp8_quant.hip
+std::vector<at::Tensor> fp8_quantize(at::Tensor x, bool transpose);
 // debug.hip
 at::Tensor mfma_tile_test(at::Tensor A, at::Tensor B);
 at::Tensor afrag_transform_test(at::Tensor M);
@@ -78,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pos_incr_", &pos_incr_, "device position += 1");
   m.def("write_token_", &write_token_, "record token into the device ring");
   m.def("gemv_bf16", &gemv_bf16, "bf16 GEMV (decode projections)");
+  m.def("fp8_quantize", &fp8_quantize, "fused bf16 -> e4m3 quantize (codes, scale)");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");
 }

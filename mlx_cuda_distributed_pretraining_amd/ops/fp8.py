@@ -43,27 +43,38 @@ def fp8_available() -> bool:
     return hasattr(torch, "_scaled_mm") and torch.cuda.is_available()
 
 
+def _quant(t: torch.Tensor, transpose: bool = False):
+    # fused HIP quantize (csrc/fp8_quant.hip) on GPU; torch fallback on CPU
+    from ._ext import get_ext
+
+    ext = get_ext()
+    if ext is not None and t.is_cuda:
+        return ext.fp8_quantize(t, transpose)
+    return quantize_e4m3(t.t().contiguous() if transpose else t)
+
+
 class _Fp8LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, weight: torch.Tensor):
         # x: [..., K] bf16; weight: [N, K] bf16
         shape = x.shape
         x2 = x.reshape(-1, shape[-1])
-        x8, sx = quantize_e4m3(x2)
-        w8, sw = quantize_e4m3(weight)
+        x8, sx = _quant(x2)
+        w8, sw = _quant(weight)
         y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
                              out_dtype=x.dtype)
-        ctx.save_for_backward(x2, w8, sw)
+        ctx.save_for_backward(x2, weight)
         return y.reshape(*shape[:-1], weight.shape[0])
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
-        x2, w8, sw = ctx.saved_tensors
-        N = w8.shape[0]
-        dy2 = dy.reshape(-1, N)
-        dy8, sd = quantize_e4m3(dy2)
-        # dx = dy @ W : b must be [N, K] column-major = (W8^T row-major).t()
-        w8t = w8.t().contiguous()
+        x2, weight = ctx.saved_tensors
+        N = weight.shape[0]
+        dy2 = dy.reshape(-1, N).contiguous()
+        dy8, sd = _quant(dy2)
+        # dx = dy @ W : b must be [N, K] column-major; the transposed-quantize
+        # kernel emits W^T [K, N] row-major, whose .t() is that col-major view
+        w8t, sw = _quant(weight, transpose=True)
         dx = torch._scaled_mm(dy8, w8t.t(), scale_a=sd, scale_b=sw,
                               out_dtype=dy.dtype)
         # wgrad in bf16 (outlier-sensitive)
