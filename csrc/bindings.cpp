@@ -50,6 +50,8 @@ void pos_incr_(at::Tensor pos);
 void write_token_(at::Tensor tok, at::Tensor ring, at::Tensor idx);
 // gemv.hip
 at::Tensor gemv_bf16(at::Tensor x, at::Tensor W);
+at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double eps,
+                   at::Tensor res);
 // fp8_quant.hip
 std::vector<at::Tensor> fp8_quantize(at::Tensor x, bool transpose);
 // debug.hip
@@ -80,6 +82,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pos_incr_", &pos_incr_, "device position += 1");
   m.def("write_token_", &write_token_, "record token into the device ring");
   m.def("gemv_bf16", &gemv_bf16, "bf16 GEMV (decode projections)");
+  m.def("gemv_ex", &gemv_ex,
+        "fused decode GEMV (rmsnorm/swiglu staging, residual epilogue)");
   m.def("fp8_quantize", &fp8_quantize, "fused bf16 -> e4m3 quantize (codes, scale)");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");

@@ -241,7 +241,35 @@ class TransformerBlock(nn.Module):
         x = x + self.mlp(self.mlp_norm(x))
         return x
 
+    def _static_decode(self, x: torch.Tensor, cache) -> torch.Tensor:
+        """hipGraph decode layer: fused GEMV chain (csrc/gemv.hip gemv_ex) —
+        norm1 folds into the QKV GEMV staging, the first residual add into
+        the wo epilogue, norm2 into gate-up staging, SwiGLU + the second add
+        into the down projection."""
+        from ..ops._ext import get_ext
+
+        ext = get_ext()
+        attn = self.attention
+        none = x.new_empty(0)
+        qkv = ext.gemv_ex(x, attn.wqkv.weight, 1, self.attention_norm.weight,
+                          self.attention_norm.eps, none)
+        o = cache.attend(attn, qkv.unsqueeze(1) if qkv.dim() == 2 else qkv)
+        B = x.shape[0]
+        o2 = o.reshape(B, -1)
+        h = ext.gemv_ex(o2, attn.wo.weight, 0, none, 0.0, x.reshape(B, -1))
+        gu = ext.gemv_ex(h, self.mlp.w_gate_up.weight, 1, self.mlp_norm.weight,
+                         self.mlp_norm.eps, none)
+        out = ext.gemv_ex(gu, self.mlp.w_down.weight, 2, none, 0.0, h)
+        return out.view_as(x)
+
     def forward(self, x: torch.Tensor, cache: Optional[KVCache] = None) -> torch.Tensor:
+        if (
+            getattr(cache, "static_decode", False)
+            and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and x.shape[0] * x.shape[1] == 1
+        ):
+            return self._static_decode(x.reshape(1, -1), cache).view_as(x)
         if self._checkpoint and self.training and cache is None:
             return torch.utils.checkpoint.checkpoint(
                 self._inner, x, cache, use_reentrant=False
@@ -277,6 +305,25 @@ class Model(nn.Module):
         x = self.tok_embeddings(tokens)
         for i, layer in enumerate(self.layers):
             x = layer(x, cache[i] if cache is not None else None)
+        if (
+            cache is not None
+            and getattr(cache[0], "static_decode", False)
+            and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and x.shape[0] * x.shape[1] == 1
+        ):
+            # decode epilogue: final RMSNorm folded into the lm-head GEMV
+            from ..ops._ext import get_ext
+
+            w = (self.tok_embeddings.weight if self.args.tie_word_embeddings
+                 else self.output.weight)
+            logits = get_ext().gemv_ex(
+                x.reshape(1, -1), w, 1, self.norm.weight, self.norm.eps,
+                x.new_empty(0),
+            ).view(x.shape[0], x.shape[1], -1)
+            if self.args.logit_scale:
+                logits = logits * self.args.logit_scale
+            return logits
         x = self.norm(x)
         if self.args.tie_word_embeddings:
             logits = linear_fast(x, self.tok_embeddings.weight)

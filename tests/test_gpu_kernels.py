@@ -397,3 +397,37 @@ def test_fp8_model_trains_on_gpu(ext):
         losses.append(float(loss.detach()))
     assert all(x == x and x < 20 for x in losses), losses
     assert losses[-1] <= losses[0] + 0.5, losses
+
+
+def test_gemv_ex_modes(ext):
+    """Fused GEMV staging modes vs the op compositions."""
+    torch.manual_seed(0)
+    K, N = 2048, 2560
+    x = torch.randn(1, K, device=dev(), dtype=torch.bfloat16)
+    W = torch.randn(N, K, device=dev(), dtype=torch.bfloat16) * 0.02
+    nw = torch.randn(K, device=dev(), dtype=torch.bfloat16)
+    res = torch.randn(1, N, device=dev(), dtype=torch.bfloat16)
+    none = x.new_empty(0)
+
+    def relerr(a, b):
+        return (a.float() - b.float()).norm() / b.float().norm()
+
+    # plain + residual epilogue
+    y = ext.gemv_ex(x, W, 0, none, 0.0, res)
+    want = torch.nn.functional.linear(x, W) + res
+    assert relerr(y, want) < 2e-2, relerr(y, want)
+
+    # rmsnorm staging
+    y = ext.gemv_ex(x, W, 1, nw, 1e-5, none)
+    xn = (x.float() * torch.rsqrt(x.float().pow(2).mean(-1, True) + 1e-5)
+          * nw.float()).to(torch.bfloat16)
+    want = torch.nn.functional.linear(xn, W)
+    assert relerr(y, want) < 2e-2, relerr(y, want)
+
+    # swiglu staging
+    gu = torch.randn(1, 2 * K, device=dev(), dtype=torch.bfloat16)
+    y = ext.gemv_ex(gu, W, 2, none, 0.0, none)
+    g, u = gu.float().split(K, dim=-1)
+    want = torch.nn.functional.linear(
+        (torch.nn.functional.silu(g) * u).to(torch.bfloat16), W)
+    assert relerr(y, want) < 2e-2, relerr(y, want)
