@@ -82,13 +82,6 @@ __device__ __forceinline__ void q_range_for_kv(int kv0, int kpb, int q_off, int 
 }
 
 // ---------------- dQ kernel (block per q-tile of QPB rows) ----------------
-// KVB2=64 kv rows per barrier, processed as two 32-row subtiles. Only K is
-// LDS-staged (row-major swizzled + transposed K^T image); V's A-fragments
-// are read DIRECTLY from global memory — all NW waves (x2 blocks/CU) read
-// the same tile rows back-to-back, so they come from L2. Halving the LDS
-// footprint is what lets the 64-row tile double-buffer at 2 blocks/CU
-// (48 KB/buffer with a V image would force 1 block/CU — measured 3-5x
-// slower in the NW=16 experiment, profiles/attn_ladder.md).
 template <int D, int MOD, int NW>
 __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
@@ -99,12 +92,11 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
     long q_rs, long k_rs, long v_rs, long do_rs, long dq_rs) {
   constexpr int TPB = NW * WAVE;
   constexpr int QPB = 32 * NW;
-  constexpr int KVB2 = 64;                    // kv rows per barrier
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
-  constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K row-major image
-  constexpr int TROW = 64;                    // K^T image row (elements; swzt())
-  constexpr int TILE = KVB2 * D + D * TROW;   // K rm + K^T
+  constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K/V row-major images
+  constexpr int TROW = 64;                     // K^T image row (elements; swzt())
+  constexpr int TILE = KVB * D * 2 + D * TROW; // K rm + V rm + K^T
 
   __shared__ __hip_bfloat16 smem[2 * TILE];
 
@@ -139,7 +131,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
   const float Lq2 = (q_valid ? lse[((long)b * Hq + hq) * Sq + qrow] : INFINITY) * LOG2E;
   const float Dq = q_valid ? drow[((long)b * Sq + qrow) * Hq + hq] : 0.f;
 
-  // kv range
+  // kv range (same as forward)
   const int blk_qpos_lo = qtile * QPB + q_off;
   const int blk_qpos_hi = blk_qpos_lo + QPB - 1;
   int kv_lo = 0, kv_hi = Skv;
@@ -147,16 +139,16 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
     kv_hi = min(Skv, blk_qpos_hi + 1);
   } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
     kv_hi = min(Skv, blk_qpos_hi + 1);
-    kv_lo = max(0, blk_qpos_lo - modarg + 1) & ~(KVB2 - 1);
+    kv_lo = max(0, blk_qpos_lo - modarg + 1) & ~(KVB - 1);
   } else if constexpr (MOD == MOD_PREFIX_LM) {
     kv_hi = min(Skv, max(blk_qpos_hi + 1, modarg));
   }
 
-  // K staging: chunks of (2 kv rows x 8 d), written row-major swizzled +
-  // pair-transposed into the K^T image
-  constexpr int CH_TOT = (KVB2 / 2) * (D / 8);
+  // T14 staging: chunks of (2 kv rows x 8 d). K chunks write row-major twice
+  // (plain + pair-transposed); V chunks write row-major.
+  constexpr int CH_TOT = (KVB / 2) * (D / 8);  // chunks per tile
   constexpr int NCH = (CH_TOT + TPB - 1) / TPB;
-  uint4 kreg[NCH][2];
+  uint4 kreg[NCH][2], vreg[NCH][2];
 
   auto stage_load = [&](int kv0) {
 #pragma unroll
@@ -164,25 +156,31 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
       const int u = tid + c * TPB;
       const int row = (u / (D / 8)) * 2;
       const int d0 = (u % (D / 8)) * 8;
-      if (row >= KVB2) continue;
+      if (row >= KVB) continue;
       const bool ok0 = kv0 + row < Skv;
       const bool ok1 = kv0 + row + 1 < Skv;
       const long kb = ((long)b * Skv + kv0 + row) * k_rs + (long)hkv * D + d0;
+      const long vb = ((long)b * Skv + kv0 + row) * v_rs + (long)hkv * D + d0;
       kreg[c][0] = ok0 ? *reinterpret_cast<const uint4*>(k + kb) : uint4{0, 0, 0, 0};
       kreg[c][1] = ok1 ? *reinterpret_cast<const uint4*>(k + kb + k_rs) : uint4{0, 0, 0, 0};
+      vreg[c][0] = ok0 ? *reinterpret_cast<const uint4*>(v + vb) : uint4{0, 0, 0, 0};
+      vreg[c][1] = ok1 ? *reinterpret_cast<const uint4*>(v + vb + v_rs) : uint4{0, 0, 0, 0};
     }
   };
   auto stage_write = [&](int bufsel) {
     __hip_bfloat16* k_lds = smem + bufsel * TILE;
-    __hip_bfloat16* kt_lds = k_lds + KVB2 * D;
+    __hip_bfloat16* v_lds = k_lds + KVB * D;
+    __hip_bfloat16* kt_lds = v_lds + KVB * D;
 #pragma unroll
     for (int c = 0; c < NCH; ++c) {
       const int u = tid + c * TPB;
       const int row = (u / (D / 8)) * 2;
       const int d0 = (u % (D / 8)) * 8;
-      if (row >= KVB2) continue;
+      if (row >= KVB) continue;
       *reinterpret_cast<uint4*>(k_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = kreg[c][0];
       *reinterpret_cast<uint4*>(k_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = kreg[c][1];
+      *reinterpret_cast<uint4*>(v_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = vreg[c][0];
+      *reinterpret_cast<uint4*>(v_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = vreg[c][1];
       Bf16x8U k0, k1;
       *reinterpret_cast<uint4*>(k0.s) = kreg[c][0];
       *reinterpret_cast<uint4*>(k1.s) = kreg[c][1];
@@ -206,99 +204,84 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
   __syncthreads();
 
   int buf = 0;
-  for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB2) {
-    const bool has_next = kv0 + KVB2 < kv_hi;
-    if (has_next) stage_load(kv0 + KVB2);
+  for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
+    const bool has_next = kv0 + KVB < kv_hi;
+    if (has_next) stage_load(kv0 + KVB);
 
     const __hip_bfloat16* k_lds = smem + buf * TILE;
-    const __hip_bfloat16* kt_lds = k_lds + KVB2 * D;
+    const __hip_bfloat16* v_lds = k_lds + KVB * D;
+    const __hip_bfloat16* kt_lds = v_lds + KVB * D;
 
+    // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
+    f32x16 st = {}, dpt = {};
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-      const int k0s = kv0 + sub * 32;
-      if (k0s >= kv_hi) break;
-      const int klq = sub * 32 + lq;
-      // V A-fragments straight from global (L2-resident: every wave of both
-      // resident blocks reads the same 32xD tile)
-      const bool kv_ok = k0s + lq < Skv;
-      const long vrow = ((long)b * Skv + (kv_ok ? k0s + lq : 0)) * v_rs + (long)hkv * D + hi * 8;
-      bf16x8 vf[DBLK];
-#pragma unroll
-      for (int dblk = 0; dblk < DBLK; ++dblk) {
-        Bf16x8U vu;
-        *reinterpret_cast<uint4*>(vu.s) =
-            kv_ok ? *reinterpret_cast<const uint4*>(v + vrow + dblk * 16) : uint4{0, 0, 0, 0};
-        vf[dblk] = vu.v;
-      }
+    for (int dblk = 0; dblk < DBLK; ++dblk) {
+      const int col = (dblk * 16 + hi * 8) ^ ((lq & KSWZ) << 3);
+      Bf16x8U kf, vf;
+      *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(k_lds + lq * D + col);
+      *reinterpret_cast<uint4*>(vf.s) = *reinterpret_cast<const uint4*>(v_lds + lq * D + col);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
+      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf.v, dof[dblk], dpt, 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
 
-      // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
-      f32x16 st = {}, dpt = {};
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int dblk = 0; dblk < DBLK; ++dblk) {
-        const int col = (dblk * 16 + hi * 8) ^ ((klq & KSWZ) << 3);
-        Bf16x8U kf;
-        *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(k_lds + klq * D + col);
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf[dblk], dof[dblk], dpt, 0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
-
-      // interior tiles: every k row of this subtile kept for this wave's q
-      bool full = k0s + 32 <= Skv;
-      if constexpr (MOD == MOD_CAUSAL) {
-        full = full && (k0s + 31 <= q0w + q_off);
-      } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
-        full = full && (k0s + 31 <= q0w + q_off) &&
-               ((q0w + 31 + q_off) - k0s < modarg);
-      } else if constexpr (MOD == MOD_PREFIX_LM) {
-        full = full && ((k0s + 31 <= q0w + q_off) || (k0s + 32 <= modarg));
-      } else if constexpr (MOD == MOD_ALIBI) {
-        full = false;
-      }
-
-      float ds[16];
-      if (full) {
-#pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          const float p = __builtin_amdgcn_exp2f(st[reg] * scale2 - Lq2);
-          ds[reg] = p * (dpt[reg] - Dq) * scale;
-        }
-      } else {
-#pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          const int k_pos = k0s + acc_row(reg, hi);
-          const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
-          float s2 = st[reg] * scale2;
-          if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
-          const float p = keep ? __builtin_amdgcn_exp2f(s2 - Lq2) : 0.f;
-          ds[reg] = p * (dpt[reg] - Dq) * scale;
-        }
-      }
-
-      bf16x8 da0, da1;
-      acc_to_afrag(ds, da0, da1);  // -> dS[32q x 16k] A-fragments
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int dc = 0; dc < DCOL; ++dc) {
-        f32x16 acc;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) acc[r] = dq_acc[dc][r];
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          const int drw = dc * 32 + lq;
-          Bf16x8U kb;  // B = K[16k x 32d] from the K^T image (this subtile's cols)
-          *reinterpret_cast<uint4*>(kb.s) = *reinterpret_cast<const uint4*>(
-              kt_lds + drw * TROW + ((sub * 32 + ks * 16 + hi * 8) ^ swzt(drw)));
-          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb.v, acc, 0, 0, 0);
-        }
-#pragma unroll
-        for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
-      }
-      __builtin_amdgcn_s_setprio(0);
+    // interior tiles: every k row of this tile is kept for every q row of
+    // this wave (wave-uniform branch)
+    bool full = kv0 + KVB <= Skv;
+    if constexpr (MOD == MOD_CAUSAL) {
+      full = full && (kv0 + KVB - 1 <= q0w + q_off);
+    } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+      full = full && (kv0 + KVB - 1 <= q0w + q_off) &&
+             ((q0w + 31 + q_off) - kv0 < modarg);
+    } else if constexpr (MOD == MOD_PREFIX_LM) {
+      full = full && ((kv0 + KVB - 1 <= q0w + q_off) || (kv0 + KVB <= modarg));
+    } else if constexpr (MOD == MOD_ALIBI) {
+      full = false;
     }
 
+    float ds[16];
+    if (full) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const float p = __builtin_amdgcn_exp2f(st[reg] * scale2 - Lq2);
+        ds[reg] = p * (dpt[reg] - Dq) * scale;
+      }
+    } else {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int k_pos = kv0 + acc_row(reg, hi);
+        const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+        float s2 = st[reg] * scale2;
+        if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
+        const float p = keep ? __builtin_amdgcn_exp2f(s2 - Lq2) : 0.f;
+        ds[reg] = p * (dpt[reg] - Dq) * scale;
+      }
+    }
+
+    bf16x8 da0, da1;
+    acc_to_afrag(ds, da0, da1);  // -> dS[32q x 16k] A-fragments
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dc = 0; dc < DCOL; ++dc) {
+      f32x16 acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[r] = dq_acc[dc][r];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int drw = dc * 32 + lq;
+        Bf16x8U kb;  // B = K[16k x 32d] from K^T image
+        *reinterpret_cast<uint4*>(kb.s) = *reinterpret_cast<const uint4*>(
+            kt_lds + drw * TROW + ((ks * 16 + hi * 8) ^ swzt(drw)));
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb.v, acc, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
+    }
+    __builtin_amdgcn_s_setprio(0);
+
     // write tile t+1 LAST: the whole iteration hides the global-load flight
+    // (buf^1 was last read before the previous barrier)
     if (has_next) stage_write(buf ^ 1);
 
     __syncthreads();
