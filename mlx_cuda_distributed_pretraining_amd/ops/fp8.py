@@ -77,13 +77,8 @@ class _Fp8LinearFn(torch.autograd.Function):
         w8t, sw = _quant(weight, transpose=True)
         dx = torch._scaled_mm(dy8, w8t.t(), scale_a=sd, scale_b=sw,
                               out_dtype=dy.dtype)
-        # wgrad in fp8 as well: dW [N,K] = dy^T [N,M] @ x [M,K].
-        # a must be row-major -> transposed-quantize dy; b must be
-        # col-major -> transposed-quantize x and view .t()
-        dy8t, sd2 = _quant(dy2, transpose=True)
-        x8t, sx = _quant(x2, transpose=True)
-        dw = torch._scaled_mm(dy8t, x8t.t(), scale_a=sd2, scale_b=sx,
-                              out_dtype=dy.dtype)
+        # wgrad in bf16 (outlier-sensitive)
+        dw = dy2.t() @ x2
         return dx.reshape(*dy.shape[:-1], x2.shape[-1]), dw
 
 
