@@ -40,6 +40,8 @@ std::vector<at::Tensor> attn_bwd_out(at::Tensor q, at::Tensor k, at::Tensor v, a
 // sampling.hip
 at::Tensor sample_token(at::Tensor logits, double temperature, double top_p, double min_p,
                         long seed);
+void sample_token_dev(at::Tensor logits, at::Tensor out, double temperature,
+                      double min_p, long seed, at::Tensor pos);
 // decode.hip (hipGraph-capturable static decode)
 void rope_decode_(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
                   at::Tensor pos);
@@ -76,6 +78,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_out", &attn_bwd_out,
         "flash attention backward into strided out views (fused dQKV)");
   m.def("sample_token", &sample_token, "fused temperature/min-p sampling");
+  m.def("sample_token_dev", &sample_token_dev,
+        "graph-capturable sampling (device position salt)");
   m.def("rope_decode_", &rope_decode_, "in-place RoPE at device position");
   m.def("kv_append_", &kv_append_, "append k/v into the static cache at device position");
   m.def("attn_decode", &attn_decode, "split-KV decode attention over the static cache");

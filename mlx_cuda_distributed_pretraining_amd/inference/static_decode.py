@@ -113,19 +113,40 @@ class StaticDecodeState:
 
 
 class GraphDecoder:
-    """Greedy decoder with (on GPU) hipGraph capture of the whole step."""
+    """Decoder with (on GPU) hipGraph capture of the whole step.
 
-    def __init__(self, model, batch: int = 1, max_len: int = 2048):
+    ``temperature=0`` decodes greedily; otherwise a graph-capturable
+    Gumbel-max sampling kernel draws tokens (fresh randomness per replay
+    via the device position salt)."""
+
+    def __init__(self, model, batch: int = 1, max_len: int = 2048,
+                 temperature: float = 0.0, min_p: float = 0.0, seed: int = 0):
         self.model = model
         self.batch = batch
         self.max_len = max_len
+        self.temperature = float(temperature)
+        self.min_p = float(min_p)
+        self.seed = int(seed)
         self.state = StaticDecodeState(model, batch, max_len)
         dev = next(model.parameters()).device
         self.dev = dev
         self.on_gpu = dev.type == "cuda"
         self.in_tok = torch.zeros(batch, 1, dtype=torch.long, device=dev)
         self.ring = torch.zeros(max_len, batch, dtype=torch.long, device=dev)
+        self._nxt = torch.zeros(batch, dtype=torch.long, device=dev)
         self.graph = None
+
+    def _pick(self, logits: torch.Tensor) -> torch.Tensor:
+        # logits: [B, V]
+        if self.temperature <= 0.0:
+            return logits.argmax(-1)
+        if self.on_gpu and logits.dtype == torch.bfloat16:
+            get_ext().sample_token_dev(logits.contiguous(), self._nxt,
+                                       self.temperature, self.min_p,
+                                       self.seed, self.state.pos)
+            return self._nxt
+        probs = torch.softmax(logits.float() / self.temperature, -1)
+        return torch.multinomial(probs, 1).squeeze(-1)
 
     @torch.no_grad()
     def prefill(self, prompt_tokens: torch.Tensor) -> None:
@@ -143,14 +164,14 @@ class GraphDecoder:
             self.state.k[i][:, :P] = c.k
             self.state.v[i][:, :P] = c.v
         self.state.pos.fill_(P)
-        first = logits[:, -1].argmax(-1)
+        first = self._pick(logits[:, -1])
         self.in_tok.copy_(first.unsqueeze(1))
         self.ring[P % self.max_len] = first
 
     @torch.no_grad()
     def _step(self) -> None:
         logits = self.model(self.in_tok, cache=self.state.layers)
-        nxt = logits[:, -1].argmax(-1)
+        nxt = self._pick(logits[:, -1])
         # advance pos, then record: the new token lives at ring[pos % cap]
         if self.on_gpu:
             ext = get_ext()

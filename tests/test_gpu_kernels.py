@@ -431,3 +431,21 @@ def test_gemv_ex_modes(ext):
     want = torch.nn.functional.linear(
         (torch.nn.functional.silu(g) * u).to(torch.bfloat16), W)
     assert relerr(y, want) < 2e-2, relerr(y, want)
+
+
+def test_sampled_graph_decode(ext):
+    """Temperature sampling inside the captured graph: tokens vary across
+    positions (device-salt RNG) and stay in-vocab."""
+    from mlx_cuda_distributed_pretraining_amd.inference.static_decode import GraphDecoder
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+
+    torch.manual_seed(0)
+    args = ModelArgs(hidden_size=256, intermediate_size=512, num_layers=2,
+                     num_heads=4, num_kv_heads=2, head_dim=128, vocab_size=503)
+    model = Model(args).to(dev(), torch.bfloat16).eval()
+    d = GraphDecoder(model, batch=1, max_len=256, temperature=1.0, seed=7)
+    d.prefill(torch.tensor([[3, 5, 7, 11]], device=dev()))
+    d.capture()
+    out = d.decode(32)[0]
+    assert ((out >= 0) & (out < 503)).all()
+    assert len(set(out.tolist())) > 4  # sampling, not a constant loop
