@@ -1,0 +1,85 @@
+#!/usr/bin/env python3
+"""End-to-end learning demonstration without network data: train the 124M
+config on a generated PATTERNED corpus (byte-level tokenizer). Unlike the
+uniform-random synthetic bench data (whose loss floor is ln(vocab)), this
+corpus has heavy n-gram structure, so the loss falls far below the unigram
+floor — evidence that the whole stack (kernels, optimizer, data path)
+actually LEARNS. Writes the loss curve PNG next to the run.
+
+Usage: python scripts/demo_learnable_run.py [--steps 200] [--out runs-demo]
+"""
+import argparse
+import json
+import random
+import sys
+import tempfile
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parents[1]
+sys.path.insert(0, str(REPO))
+
+from mlx_cuda_distributed_pretraining_amd.core.config import Config  # noqa: E402
+from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer  # noqa: E402
+
+SUBJECTS = ["the rover", "a signal", "the array", "our probe", "the lander",
+            "a beacon", "the relay", "this sensor"]
+VERBS = ["measures", "transmits", "records", "detects", "amplifies",
+         "calibrates", "tracks", "samples"]
+OBJECTS = ["the magnetic field", "a dust storm", "the thermal gradient",
+           "an ion stream", "the carrier wave", "a pressure drop",
+           "the solar flux", "a faint echo"]
+TAILS = ["near the crater rim.", "during the long night.", "at full power.",
+         "before the window closes.", "across the basin.", "in low orbit."]
+
+
+def make_corpus(path: Path, n_docs: int = 4000, seed: int = 0) -> None:
+    rng = random.Random(seed)
+    with open(path, "w") as f:
+        for _ in range(n_docs):
+            sents = [" ".join([rng.choice(SUBJECTS), rng.choice(VERBS),
+                               rng.choice(OBJECTS), rng.choice(TAILS)])
+                     for _ in range(rng.randint(4, 10))]
+            f.write(json.dumps({"text": " ".join(sents)}) + "\n")
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--out", type=str, default=None)
+    a = p.parse_args()
+
+    tmp = Path(tempfile.mkdtemp(prefix="demo_corpus_"))
+    corpus = tmp / "corpus.jsonl"
+    val = tmp / "val.jsonl"
+    make_corpus(corpus, 4000, seed=0)
+    make_corpus(val, 200, seed=1)
+
+    cfg = Config.from_yaml(REPO / "configs" / "model-config-124m.yaml")
+    cfg.name = "demo-learnable-124m"
+    cfg.overwrite = True
+    cfg.data.synthetic = False
+    cfg.data.input_file = str(corpus)
+    cfg.data.validation_file = str(val)
+    cfg.data.tokenizer = {"normal_vocab_size": 256,
+                          "special_tokens": {"pad": "<pad>", "bos": "<bos>", "eos": "<eos>"}}
+    cfg.data.preprocessing["max_context_size"] = 512
+    cfg.training.hyperparameters.update({"iters": a.steps, "batch_size": 16,
+                                         "learning_rate": 6e-4})
+    cfg.training.scheduler = {"type": "cosine_with_warmup", "warmup_steps": 20,
+                              "min_lr_ratio": 0.1}
+    cfg.logging.steps = {"logging_interval": 10, "checkpoint_interval": 0,
+                         "validation_interval": a.steps // 2}
+    runs_root = a.out or str(tmp / "runs")
+    trainer = Trainer(cfg, runs_root=runs_root)
+    trainer.train()
+
+    run_dir = Path(runs_root) / cfg.name
+    from mlx_cuda_distributed_pretraining_amd.utils.plotting import plot_run
+
+    plot_run(run_dir)
+    print(f"run dir: {run_dir}")
+    print((run_dir / "log.txt").read_text().splitlines()[-3:])
+
+
+if __name__ == "__main__":
+    main()
