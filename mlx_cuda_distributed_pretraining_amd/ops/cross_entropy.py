@@ -34,7 +34,7 @@ class _FusedCEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits: torch.Tensor, targets: torch.Tensor, ignore_index: int):
         n, v = logits.shape
-        if use_hip(logits):
+        if use_hip(logits) and logits.shape[-1] % 8 == 0:  # kernel needs V%8==0
             ext = get_ext()
             loss_sum, ntok, lse = ext.ce_fwd(logits.contiguous(), targets.contiguous(), ignore_index)
             ctx.save_for_backward(logits, targets, lse, ntok)

@@ -60,7 +60,8 @@ def main() -> None:
     cfg.data.synthetic = False
     cfg.data.input_file = str(corpus)
     cfg.data.validation_file = str(val)
-    cfg.data.tokenizer = {"normal_vocab_size": 256,
+    # 261 + 3 specials = 264 (the fused-CE kernel wants vocab % 8 == 0)
+    cfg.data.tokenizer = {"normal_vocab_size": 261,
                           "special_tokens": {"pad": "<pad>", "bos": "<bos>", "eos": "<eos>"}}
     cfg.data.preprocessing["max_context_size"] = 512
     cfg.training.hyperparameters.update({"iters": a.steps, "batch_size": 16,
@@ -76,7 +77,10 @@ def main() -> None:
     run_dir = Path(runs_root) / cfg.name
     from mlx_cuda_distributed_pretraining_amd.utils.plotting import plot_run
 
-    plot_run(run_dir)
+    try:
+        plot_run(run_dir)
+    except ValueError:
+        pass  # too few steps to have log lines
     print(f"run dir: {run_dir}")
     print((run_dir / "log.txt").read_text().splitlines()[-3:])
 
