@@ -57,71 +57,87 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
 #define GMODE_RMSNORM 1
 #define GMODE_SWIGLU 2
 
-template <int MODE>
+template <int MODE, int BB>
 __global__ __launch_bounds__(256) void gemv_ex_kernel(
     const __hip_bfloat16* __restrict__ W, const __hip_bfloat16* __restrict__ x,
     const __hip_bfloat16* __restrict__ nw, const __hip_bfloat16* __restrict__ res,
-    __hip_bfloat16* __restrict__ y, int N, int K, int Kin, float eps) {
-  extern __shared__ __hip_bfloat16 x_lds[];
+    __hip_bfloat16* __restrict__ y, int N, int K, int Kin, float eps, int B) {
+  // BB = compile-time batch: all BB x-vectors staged in LDS, W streamed ONCE
+  // with BB accumulators per output row (the whole point of batching — a
+  // per-row grid.y launch re-reads W B times).
+  extern __shared__ __hip_bfloat16 x_lds[];  // [BB][K]
   __shared__ float scratch[256 / WAVE];
-  const int b = blockIdx.y;
   const int tid = threadIdx.x;
-  const __hip_bfloat16* xb = x + (long)b * Kin;
 
-  if constexpr (MODE == GMODE_RMSNORM) {
-    float ss = 0.f;
-    for (int i = tid; i < K / 8; i += 256) {
-      U4 u; u.u = reinterpret_cast<const uint4*>(xb)[i];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) { float f = bf16_bits_to_f32(u.s[j]); ss += f * f; }
-    }
-    ss = block_reduce_sum<256>(ss, scratch);
-    const float r = rsqrtf(ss / K + eps);
-    for (int i = tid; i < K / 8; i += 256) {
-      U4 u, w8, o;
-      u.u = reinterpret_cast<const uint4*>(xb)[i];
-      w8.u = reinterpret_cast<const uint4*>(nw)[i];
+  for (int b = 0; b < BB; ++b) {
+    if (b >= B) break;
+    const __hip_bfloat16* xb = x + (long)b * Kin;
+    __hip_bfloat16* xl = x_lds + (long)b * K;
+    if constexpr (MODE == GMODE_RMSNORM) {
+      float ss = 0.f;
+      for (int i = tid; i < K / 8; i += 256) {
+        U4 u; u.u = reinterpret_cast<const uint4*>(xb)[i];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        o.s[j] = f32_to_bf16_bits(bf16_bits_to_f32(u.s[j]) * r * bf16_bits_to_f32(w8.s[j]));
-      reinterpret_cast<uint4*>(x_lds)[i] = o.u;
-    }
-  } else if constexpr (MODE == GMODE_SWIGLU) {
-    // xb = [gate | up] of length Kin = 2K; stage silu(gate)*up
-    for (int i = tid; i < K / 8; i += 256) {
-      U4 g, u, o;
-      g.u = reinterpret_cast<const uint4*>(xb)[i];
-      u.u = reinterpret_cast<const uint4*>(xb + K)[i];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float gf = bf16_bits_to_f32(g.s[j]);
-        const float uf = bf16_bits_to_f32(u.s[j]);
-        o.s[j] = f32_to_bf16_bits(gf / (1.f + __expf(-gf)) * uf);
+        for (int j = 0; j < 8; ++j) { float f = bf16_bits_to_f32(u.s[j]); ss += f * f; }
       }
-      reinterpret_cast<uint4*>(x_lds)[i] = o.u;
+      ss = block_reduce_sum<256>(ss, scratch);
+      const float r = rsqrtf(ss / K + eps);
+      for (int i = tid; i < K / 8; i += 256) {
+        U4 u, w8, o;
+        u.u = reinterpret_cast<const uint4*>(xb)[i];
+        w8.u = reinterpret_cast<const uint4*>(nw)[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o.s[j] = f32_to_bf16_bits(bf16_bits_to_f32(u.s[j]) * r * bf16_bits_to_f32(w8.s[j]));
+        reinterpret_cast<uint4*>(xl)[i] = o.u;
+      }
+    } else if constexpr (MODE == GMODE_SWIGLU) {
+      for (int i = tid; i < K / 8; i += 256) {
+        U4 g, u, o;
+        g.u = reinterpret_cast<const uint4*>(xb)[i];
+        u.u = reinterpret_cast<const uint4*>(xb + K)[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float gf = bf16_bits_to_f32(g.s[j]);
+          const float uf = bf16_bits_to_f32(u.s[j]);
+          o.s[j] = f32_to_bf16_bits(gf / (1.f + __expf(-gf)) * uf);
+        }
+        reinterpret_cast<uint4*>(xl)[i] = o.u;
+      }
+    } else {
+      for (int i = tid; i < K / 8; i += 256)
+        reinterpret_cast<uint4*>(xl)[i] = reinterpret_cast<const uint4*>(xb)[i];
     }
-  } else {
-    for (int i = tid; i < K / 8; i += 256)
-      reinterpret_cast<uint4*>(x_lds)[i] = reinterpret_cast<const uint4*>(xb)[i];
   }
   __syncthreads();
 
   const int wid = tid / WAVE, lane = tid % WAVE;
   for (int row = blockIdx.x * 4 + wid; row < N; row += gridDim.x * 4) {
     const __hip_bfloat16* wr = W + (long)row * K;
-    float acc = 0.f;
-    for (int k = lane * 8; k + 8 <= K; k += WAVE * 8) {
-      U4 wv, xv;
-      wv.u = *reinterpret_cast<const uint4*>(wr + k);
-      xv.u = *reinterpret_cast<const uint4*>(x_lds + k);
+    float acc[BB];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        acc += bf16_bits_to_f32(wv.s[j]) * bf16_bits_to_f32(xv.s[j]);
+    for (int b = 0; b < BB; ++b) acc[b] = 0.f;
+    for (int k = lane * 8; k + 8 <= K; k += WAVE * 8) {
+      U4 wv;
+      wv.u = *reinterpret_cast<const uint4*>(wr + k);
+#pragma unroll
+      for (int b = 0; b < BB; ++b) {
+        U4 xv;
+        xv.u = *reinterpret_cast<const uint4*>(x_lds + (long)b * K + k);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[b] += bf16_bits_to_f32(wv.s[j]) * bf16_bits_to_f32(xv.s[j]);
+      }
     }
-    acc = wave_reduce_sum(acc);
-    if (lane == 0) {
-      if (res != nullptr) acc += to_f32(res[(long)b * N + row]);
-      from_f32(&y[(long)b * N + row], acc);
+#pragma unroll
+    for (int b = 0; b < BB; ++b) {
+      if (b >= B) break;
+      float a = wave_reduce_sum(acc[b]);
+      if (lane == 0) {
+        if (res != nullptr) a += to_f32(res[(long)b * N + row]);
+        from_f32(&y[(long)b * N + row], a);
+      }
     }
   }
 }
@@ -131,7 +147,7 @@ __global__ __launch_bounds__(256) void gemv_ex_kernel(
 at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double eps,
                    at::Tensor res) {
   // x: [B, Kin]; W: [N, K]; mode: 0 plain (Kin==K), 1 rmsnorm (Kin==K,
-  // nw [K]), 2 swiglu (Kin==2K). res: [B, N] or empty.
+  // nw [K]), 2 swiglu (Kin==2K). res: [B, N] or empty. B <= 8.
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && W.is_contiguous());
   const int K = W.size(1), N = W.size(0);
   auto xc = x.contiguous().view({-1, x.size(-1)});
@@ -139,13 +155,15 @@ at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double 
   TORCH_CHECK(K % 8 == 0, "gemv_ex: K must be a multiple of 8");
   TORCH_CHECK((mode == 2 && Kin == 2 * K) || (mode != 2 && Kin == K), "gemv_ex: bad Kin");
   const int B = xc.size(0);
+  TORCH_CHECK(B <= 8, "gemv_ex: B must be <= 8");
   auto sizes = x.sizes().vec();
   sizes.back() = N;
   auto y = at::empty({B, N}, x.options());
   auto stream = at::cuda::getCurrentHIPStream();
   const int grid_x = std::min(cdiv(N, 4), 2048);
-  const size_t lds = (size_t)K * sizeof(__hip_bfloat16);
-  TORCH_CHECK(lds <= 160 * 1024, "gemv_ex: K too large for LDS staging");
+  const int BB = B <= 1 ? 1 : (B <= 2 ? 2 : (B <= 4 ? 4 : 8));
+  const size_t lds = (size_t)BB * K * sizeof(__hip_bfloat16);
+  TORCH_CHECK(lds <= 160 * 1024, "gemv_ex: B*K too large for LDS staging");
   auto* Wp = reinterpret_cast<const __hip_bfloat16*>(W.data_ptr());
   auto* xp = reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr());
   auto* np = mode == 1 ? reinterpret_cast<const __hip_bfloat16*>(nw.contiguous().data_ptr())
@@ -153,13 +171,23 @@ at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double 
   auto* rp = res.numel() > 0 ? reinterpret_cast<const __hip_bfloat16*>(res.contiguous().data_ptr())
                              : nullptr;
   auto* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
-  dim3 g(grid_x, B);
-  switch (mode) {
-    case 0: gemv_ex_kernel<GMODE_PLAIN><<<g, 256, lds, stream>>>(Wp, xp, np, rp, yp, N, K, Kin, (float)eps); break;
-    case 1: gemv_ex_kernel<GMODE_RMSNORM><<<g, 256, lds, stream>>>(Wp, xp, np, rp, yp, N, K, Kin, (float)eps); break;
-    case 2: gemv_ex_kernel<GMODE_SWIGLU><<<g, 256, lds, stream>>>(Wp, xp, np, rp, yp, N, K, Kin, (float)eps); break;
-    default: TORCH_CHECK(false, "gemv_ex: bad mode");
+#define GLAUNCH(M_, B_) \
+  gemv_ex_kernel<M_, B_><<<grid_x, 256, lds, stream>>>(Wp, xp, np, rp, yp, N, K, Kin, (float)eps, B)
+#define GMODE_SWITCH(B_)                                   \
+  switch (mode) {                                          \
+    case 0: GLAUNCH(GMODE_PLAIN, B_); break;               \
+    case 1: GLAUNCH(GMODE_RMSNORM, B_); break;             \
+    case 2: GLAUNCH(GMODE_SWIGLU, B_); break;              \
+    default: TORCH_CHECK(false, "gemv_ex: bad mode");      \
   }
+  switch (BB) {
+    case 1: GMODE_SWITCH(1); break;
+    case 2: GMODE_SWITCH(2); break;
+    case 4: GMODE_SWITCH(4); break;
+    case 8: GMODE_SWITCH(8); break;
+  }
+#undef GMODE_SWITCH
+#undef GLAUNCH
   return y.view(sizes);
 }
 

@@ -267,9 +267,10 @@ class TransformerBlock(nn.Module):
             getattr(cache, "static_decode", False)
             and x.is_cuda
             and x.dtype == torch.bfloat16
-            and x.shape[0] * x.shape[1] == 1
+            and x.shape[1] == 1
+            and x.shape[0] <= 8  # batched GEMV limit (LDS staging)
         ):
-            return self._static_decode(x.reshape(1, -1), cache).view_as(x)
+            return self._static_decode(x[:, 0, :], cache).unsqueeze(1)
         if self._checkpoint and self.training and cache is None:
             return torch.utils.checkpoint.checkpoint(
                 self._inner, x, cache, use_reentrant=False
@@ -310,7 +311,8 @@ class Model(nn.Module):
             and getattr(cache[0], "static_decode", False)
             and x.is_cuda
             and x.dtype == torch.bfloat16
-            and x.shape[0] * x.shape[1] == 1
+            and x.shape[1] == 1
+            and x.shape[0] <= 8
         ):
             # decode epilogue: final RMSNorm folded into the lm-head GEMV
             from ..ops._ext import get_ext
@@ -318,9 +320,9 @@ class Model(nn.Module):
             w = (self.tok_embeddings.weight if self.args.tie_word_embeddings
                  else self.output.weight)
             logits = get_ext().gemv_ex(
-                x.reshape(1, -1), w, 1, self.norm.weight, self.norm.eps,
+                x[:, 0, :], w, 1, self.norm.weight, self.norm.eps,
                 x.new_empty(0),
-            ).view(x.shape[0], x.shape[1], -1)
+            ).view(x.shape[0], 1, -1)
             if self.args.logit_scale:
                 logits = logits * self.args.logit_scale
             return logits
