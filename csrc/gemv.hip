@@ -161,18 +161,25 @@ at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double 
   auto y = at::empty({B, N}, x.options());
   auto stream = at::cuda::getCurrentHIPStream();
   const int grid_x = std::min(cdiv(N, 4), 2048);
-  const int BB = B <= 1 ? 1 : (B <= 2 ? 2 : (B <= 4 ? 4 : 8));
-  const size_t lds = (size_t)BB * K * sizeof(__hip_bfloat16);
-  TORCH_CHECK(lds <= 160 * 1024, "gemv_ex: B*K too large for LDS staging");
   auto* Wp = reinterpret_cast<const __hip_bfloat16*>(W.data_ptr());
-  auto* xp = reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr());
   auto* np = mode == 1 ? reinterpret_cast<const __hip_bfloat16*>(nw.contiguous().data_ptr())
                        : nullptr;
-  auto* rp = res.numel() > 0 ? reinterpret_cast<const __hip_bfloat16*>(res.contiguous().data_ptr())
-                             : nullptr;
-  auto* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
+  auto rc = res.numel() > 0 ? res.contiguous() : res;
+  // BB caps at 4: beyond that the inner loop turns LDS/VALU-bound (measured
+  // at B=8); rows 5..8 run as two <=4-row chunks (W streamed once per chunk)
+  for (int b0 = 0; b0 < B; b0 += 4) {
+    const int Bc = std::min(4, B - b0);
+    const int BB = Bc <= 1 ? 1 : (Bc <= 2 ? 2 : 4);
+    const size_t lds = (size_t)BB * K * sizeof(__hip_bfloat16);
+    TORCH_CHECK(lds <= 160 * 1024, "gemv_ex: K too large for LDS staging");
+    auto* xp = reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr()) + (long)b0 * Kin;
+    auto* rp = rc.numel() > 0
+                   ? reinterpret_cast<const __hip_bfloat16*>(rc.data_ptr()) + (long)b0 * N
+                   : nullptr;
+    auto* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr()) + (long)b0 * N;
+    const int Bv = Bc;
 #define GLAUNCH(M_, B_) \
-  gemv_ex_kernel<M_, B_><<<grid_x, 256, lds, stream>>>(Wp, xp, np, rp, yp, N, K, Kin, (float)eps, B)
+  gemv_ex_kernel<M_, B_><<<grid_x, 256, lds, stream>>>(Wp, xp, np, rp, yp, N, K, Kin, (float)eps, Bv)
 #define GMODE_SWITCH(B_)                                   \
   switch (mode) {                                          \
     case 0: GLAUNCH(GMODE_PLAIN, B_); break;               \
@@ -180,11 +187,11 @@ at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double 
     case 2: GLAUNCH(GMODE_SWIGLU, B_); break;              \
     default: TORCH_CHECK(false, "gemv_ex: bad mode");      \
   }
-  switch (BB) {
-    case 1: GMODE_SWITCH(1); break;
-    case 2: GMODE_SWITCH(2); break;
-    case 4: GMODE_SWITCH(4); break;
-    case 8: GMODE_SWITCH(8); break;
+    switch (BB) {
+      case 1: GMODE_SWITCH(1); break;
+      case 2: GMODE_SWITCH(2); break;
+      case 4: GMODE_SWITCH(4); break;
+    }
   }
 #undef GMODE_SWITCH
 #undef GLAUNCH
