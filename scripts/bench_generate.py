@@ -29,6 +29,8 @@ def main():
     p.add_argument("--decode-tokens", type=int, default=128)
     p.add_argument("--kv-bits", type=int, default=None)
     p.add_argument("--batch", type=int, default=1)
+    p.add_argument("--eager", action="store_true",
+                   help="eager KV-cache loop instead of the hipGraph decoder")
     args = p.parse_args()
 
     on_gpu = torch.cuda.is_available()
@@ -43,6 +45,32 @@ def main():
     model = Model(margs).to(dev, dtype).eval()
 
     toks = torch.randint(0, vocab, (args.batch, args.prompt_len), device=dev)
+
+    if not args.eager and on_gpu and args.kv_bits is None and args.batch <= 8:
+        # hipGraph-captured decoder (the serving path)
+        from mlx_cuda_distributed_pretraining_amd.inference.static_decode import (
+            GraphDecoder,
+        )
+
+        dec = GraphDecoder(model, batch=args.batch,
+                           max_len=args.prompt_len + args.decode_tokens + 8)
+        t0 = time.perf_counter()
+        dec.prefill(toks)
+        torch.cuda.synchronize()
+        t_prefill = time.perf_counter() - t0
+        dec.capture()
+        dec.decode(8)  # warm replays
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        dec.decode(args.decode_tokens - 8)
+        torch.cuda.synchronize()
+        t_decode = time.perf_counter() - t0
+        n_pre = args.batch * args.prompt_len
+        n_dec = args.batch * (args.decode_tokens - 8)
+        print(f"prefill: {n_pre} toks in {t_prefill*1e3:.1f} ms = {n_pre/t_prefill:.0f} tok/s")
+        print(f"decode (hipGraph): {n_dec} toks in {t_decode*1e3:.1f} ms = "
+              f"{n_dec/t_decode:.1f} tok/s ({1e3*t_decode/max(n_dec//args.batch,1):.2f} ms/step)")
+        return
 
     def sync():
         if on_gpu:
