@@ -11,7 +11,6 @@ trainer before logging.
 from __future__ import annotations
 
 import math
-import sys
 import time
 from pathlib import Path
 from typing import Any, Dict, Optional

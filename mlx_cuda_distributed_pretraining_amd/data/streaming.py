@@ -16,9 +16,7 @@ from __future__ import annotations
 
 import gzip
 import json
-import os
 import shutil
-import time
 from pathlib import Path
 from typing import Iterable, Iterator, List, Optional, Sequence
 

@@ -130,8 +130,6 @@ def beam_search(
     topv, topi = logits.reshape(-1).topk(beam_width)
     beams = []
     for v, i in zip(topv.tolist(), topi.tolist()):
-        import copy
-
         c = make_prompt_cache(model)
         # replay prompt per beam (simple; beams share prompt prefix cost once here)
         model(y, cache=c)

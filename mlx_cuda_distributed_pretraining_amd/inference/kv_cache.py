@@ -15,7 +15,7 @@ a further optimization, not required for parity).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 

@@ -24,8 +24,7 @@ used by the unit tests as the numerics oracle.
 """
 from __future__ import annotations
 
-import math
-from typing import List, Optional
+from typing import List
 
 import torch
 

@@ -16,7 +16,7 @@ zero transpose copies on the hot path.)
 from __future__ import annotations
 
 import math
-from typing import Callable, Optional, Tuple
+from typing import Callable, Optional
 
 import torch
 

@@ -17,7 +17,6 @@ config); the BASELINE headline remains bf16.
 """
 from __future__ import annotations
 
-from typing import Optional
 
 import torch
 

@@ -8,7 +8,6 @@ reduce-scatter grads -> update own shard -> all-gather params, all over RCCL.
 """
 from __future__ import annotations
 
-import math
 from typing import Optional
 
 import torch

@@ -10,7 +10,7 @@ Line format (Logger.format_metrics, core/logger.py):
 from __future__ import annotations
 
 import re
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from pathlib import Path
 from typing import List, Optional
 
