@@ -46,6 +46,8 @@ def main() -> None:
     p = argparse.ArgumentParser()
     p.add_argument("--steps", type=int, default=200)
     p.add_argument("--out", type=str, default=None)
+    p.add_argument("--fp8", action="store_true", help="opt-in e4m3 GEMM path")
+    p.add_argument("--name", type=str, default=None)
     a = p.parse_args()
 
     tmp = Path(tempfile.mkdtemp(prefix="demo_corpus_"))
@@ -55,7 +57,9 @@ def main() -> None:
     make_corpus(val, 200, seed=1)
 
     cfg = Config.from_yaml(REPO / "configs" / "model-config-124m.yaml")
-    cfg.name = "demo-learnable-124m"
+    cfg.name = a.name or ("demo-learnable-124m-fp8" if a.fp8 else "demo-learnable-124m")
+    if a.fp8:
+        cfg.model.misc = dict(cfg.model.misc or {}, fp8=True)
     cfg.overwrite = True
     cfg.data.synthetic = False
     cfg.data.input_file = str(corpus)
