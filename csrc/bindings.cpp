@@ -50,6 +50,10 @@ at::Tensor attn_decode(at::Tensor q, at::Tensor kc, at::Tensor vc, at::Tensor po
                        at::Tensor part, double scale);
 void pos_incr_(at::Tensor pos);
 void write_token_(at::Tensor tok, at::Tensor ring, at::Tensor idx);
+void kv_append_q8_(at::Tensor k, at::Tensor v, at::Tensor kc, at::Tensor ksz,
+                   at::Tensor vc, at::Tensor vsz, at::Tensor pos);
+at::Tensor attn_decode_q8(at::Tensor q, at::Tensor kc, at::Tensor ksz, at::Tensor vc,
+                          at::Tensor vsz, at::Tensor pos, at::Tensor part, double scale);
 // gemv.hip
 at::Tensor gemv_bf16(at::Tensor x, at::Tensor W);
 at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double eps,
@@ -85,6 +89,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode, "split-KV decode attention over the static cache");
   m.def("pos_incr_", &pos_incr_, "device position += 1");
   m.def("write_token_", &write_token_, "record token into the device ring");
+  m.def("kv_append_q8_", &kv_append_q8_, "quantize+append k/v into the int8 static cache");
+  m.def("attn_decode_q8", &attn_decode_q8, "decode attention over the int8 cache (fused dequant)");
   m.def("gemv_bf16", &gemv_bf16, "bf16 GEMV (decode projections)");
   m.def("gemv_ex", &gemv_ex,
         "fused decode GEMV (rmsnorm/swiglu staging, residual epilogue)");

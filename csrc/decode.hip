@@ -208,6 +208,126 @@ __global__ void attn_decode_combine_kernel(const float* __restrict__ part,
   }
 }
 
+
+// ---- int8-quantized KV decode (K13: quantized cache at kernel speed) ----
+// Codes: uint8 [B, Lmax, Hkv, D]; per-64-elem group (scale, zero) float2
+// [B, Lmax, Hkv, D/64]. Dequant fused into the attention dot loops — the
+// eager torch dequant path rebuilt the full bf16 cache every step.
+
+__global__ void kv_append_q8_kernel(const __hip_bfloat16* __restrict__ k,
+                                    const __hip_bfloat16* __restrict__ v,
+                                    unsigned char* __restrict__ kc, float2* __restrict__ ksz,
+                                    unsigned char* __restrict__ vc, float2* __restrict__ vsz,
+                                    const int* __restrict__ pos,
+                                    int B, int Lmax, int Hkv, int D) {
+  // one wave per (b, h, tensor, group-of-64): lanes 0..63 cover the group
+  const int G = 64;
+  const int groups = D / G;
+  const long total = (long)B * Hkv * 2 * groups;
+  const long gi = (blockIdx.x * (long)blockDim.x + threadIdx.x) / WAVE;
+  if (gi >= total) return;
+  const int lane = threadIdx.x % WAVE;
+  const int grp = (int)(gi % groups);
+  const int t = (int)((gi / groups) % 2);
+  const int h = (int)((gi / (2 * groups)) % Hkv);
+  const int b = (int)(gi / (2 * groups * Hkv));
+  const int p = *pos;
+  const __hip_bfloat16* src = (t == 0 ? k : v) + ((long)b * Hkv + h) * D + grp * G;
+  const float x = to_f32(src[lane]);
+  float lo = x, hi = x;
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    lo = fminf(lo, __shfl_xor(lo, off));
+    hi = fmaxf(hi, __shfl_xor(hi, off));
+  }
+  const float scale = fmaxf((hi - lo) / 255.f, 1e-8f);
+  const int code = (int)rintf((x - lo) / scale);
+  const long row = ((long)b * Lmax + p) * Hkv + h;
+  unsigned char* cdst = (t == 0 ? kc : vc) + row * D + grp * G;
+  cdst[lane] = (unsigned char)min(max(code, 0), 255);
+  if (lane == 0) (t == 0 ? ksz : vsz)[row * (D / G) + grp] = float2{scale, lo};
+}
+
+template <int D, int CHUNK>
+__global__ __launch_bounds__(256) void attn_decode_partial_q8_kernel(
+    const __hip_bfloat16* __restrict__ q,
+    const unsigned char* __restrict__ kc, const float2* __restrict__ ksz,
+    const unsigned char* __restrict__ vc, const float2* __restrict__ vsz,
+    const int* __restrict__ pos, float* __restrict__ part,
+    int B, int Lmax, int Hq, int Hkv, float scale2) {
+  const int chunk = blockIdx.x, hq = blockIdx.y, b = blockIdx.z;
+  const int NC = gridDim.x;
+  const int hkv = hq / (Hq / Hkv);
+  const int len = *pos + 1;
+  const int k0 = chunk * CHUNK;
+  float* out = part + (((long)b * Hq + hq) * NC + chunk) * (D + 2);
+  if (k0 >= len) {
+    if (threadIdx.x == 0) { out[D] = -INFINITY; out[D + 1] = 0.f; }
+    return;
+  }
+  const int kend = min(len, k0 + CHUNK);
+  __shared__ float scratch[256 / WAVE];
+  __shared__ float s_row[CHUNK];
+  const int tid = threadIdx.x;
+  for (int r = tid; r < CHUNK; r += 256) s_row[r] = -INFINITY;
+  __syncthreads();
+
+  const __hip_bfloat16* qp = q + ((long)b * Hq + hq) * D;
+  const int lpr = D / 8;
+  const int rpw = WAVE / lpr;
+  const int wid = tid / WAVE, lane = tid % WAVE;
+  const int sub = lane / lpr;
+  const int dl = (lane % lpr) * 8;
+  U4 qv;
+  qv.u = *reinterpret_cast<const uint4*>(qp + dl);
+
+  for (int r0 = k0 + wid * rpw; r0 < kend; r0 += 4 * rpw) {
+    const int krow = r0 + sub;
+    float acc = 0.f;
+    if (krow < kend) {
+      const long row = ((long)b * Lmax + krow) * Hkv + hkv;
+      const float2 sz = ksz[row * (D / 64) + dl / 64];
+      const unsigned char* kr = kc + row * D + dl;
+      uint2 codes = *reinterpret_cast<const uint2*>(kr);
+      const unsigned char* cb = reinterpret_cast<const unsigned char*>(&codes);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc += bf16_bits_to_f32(qv.s[j]) * (cb[j] * sz.x + sz.y);
+    }
+    for (int off = lpr / 2; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
+    if (krow < kend && (lane % lpr) == 0) s_row[krow - k0] = acc * scale2;
+  }
+  __syncthreads();
+
+  float m = -INFINITY;
+  for (int r = tid; r < CHUNK; r += 256) m = fmaxf(m, s_row[r]);
+  m = block_reduce_max<256>(m, scratch);
+  float l = 0.f;
+  for (int r = tid; r < CHUNK; r += 256) {
+    if (s_row[r] != -INFINITY) {
+      const float p = __builtin_amdgcn_exp2f(s_row[r] - m);
+      s_row[r] = p;
+      l += p;
+    } else {
+      s_row[r] = 0.f;
+    }
+  }
+  l = block_reduce_sum<256>(l, scratch);
+
+  const int nrows = kend - k0;
+  for (int d = tid; d < D; d += 256) {
+    float acc = 0.f;
+    const int grp = d / 64;
+    for (int r = 0; r < nrows; ++r) {
+      const long row = ((long)b * Lmax + k0 + r) * Hkv + hkv;
+      const float2 sz = vsz[row * (D / 64) + grp];
+      acc += s_row[r] * (vc[row * D + d] * sz.x + sz.y);
+    }
+    out[d] = acc;
+  }
+  if (tid == 0) { out[D] = m; out[D + 1] = l; }
+}
+
 __global__ void pos_incr_kernel(int* pos) {
   if (threadIdx.x == 0 && blockIdx.x == 0) *pos += 1;
 }
@@ -284,6 +404,53 @@ at::Tensor attn_decode(at::Tensor q, at::Tensor kc, at::Tensor vc, at::Tensor po
   } else {
     TORCH_CHECK(false, "attn_decode: head_dim must be 64 or 128");
   }
+  return o;
+}
+
+
+void kv_append_q8_(at::Tensor k, at::Tensor v, at::Tensor kc, at::Tensor ksz,
+                   at::Tensor vc, at::Tensor vsz, at::Tensor pos) {
+  const int B = kc.size(0), Lmax = kc.size(1), Hkv = kc.size(2), D = kc.size(3);
+  TORCH_CHECK(D % 64 == 0, "kv_append_q8: D must be a multiple of 64");
+  auto stream = at::cuda::getCurrentHIPStream();
+  const long waves = (long)B * Hkv * 2 * (D / 64);
+  kv_append_q8_kernel<<<cdiv(waves * WAVE, 256), 256, 0, stream>>>(
+      reinterpret_cast<const __hip_bfloat16*>(k.contiguous().data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(v.contiguous().data_ptr()),
+      kc.data_ptr<unsigned char>(), reinterpret_cast<float2*>(ksz.data_ptr<float>()),
+      vc.data_ptr<unsigned char>(), reinterpret_cast<float2*>(vsz.data_ptr<float>()),
+      pos.data_ptr<int>(), B, Lmax, Hkv, D);
+}
+
+at::Tensor attn_decode_q8(at::Tensor q, at::Tensor kc, at::Tensor ksz, at::Tensor vc,
+                          at::Tensor vsz, at::Tensor pos, at::Tensor part, double scale) {
+  constexpr int CHUNK = 256;
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  const int B = kc.size(0), Lmax = kc.size(1), Hkv = kc.size(2), D = kc.size(3);
+  const int Hq = q.size(2);
+  const int NC = cdiv(Lmax, CHUNK);
+  TORCH_CHECK(NC <= 64, "attn_decode_q8: Lmax too large");
+  auto o = at::empty({B, 1, Hq, D}, q.options());
+  auto stream = at::cuda::getCurrentHIPStream();
+  const float scale2 = (float)scale * LOG2E;
+  dim3 gA(NC, Hq, B);
+  auto* qp = reinterpret_cast<const __hip_bfloat16*>(q.contiguous().data_ptr());
+  auto* op = reinterpret_cast<__hip_bfloat16*>(o.data_ptr());
+#define Q8LAUNCH(D_)                                                                     \
+  do {                                                                                   \
+    attn_decode_partial_q8_kernel<D_, CHUNK><<<gA, 256, 0, stream>>>(                    \
+        qp, kc.data_ptr<unsigned char>(),                                                \
+        reinterpret_cast<const float2*>(ksz.data_ptr<float>()),                          \
+        vc.data_ptr<unsigned char>(),                                                    \
+        reinterpret_cast<const float2*>(vsz.data_ptr<float>()), pos.data_ptr<int>(),     \
+        part.data_ptr<float>(), B, Lmax, Hq, Hkv, scale2);                               \
+    attn_decode_combine_kernel<D_><<<dim3(Hq, B), 128, 0, stream>>>(                     \
+        part.data_ptr<float>(), op, pos.data_ptr<int>(), NC, Hq, CHUNK);                 \
+  } while (0)
+  if (D == 128) Q8LAUNCH(128);
+  else if (D == 64) Q8LAUNCH(64);
+  else TORCH_CHECK(false, "attn_decode_q8: head_dim must be 64 or 128");
+#undef Q8LAUNCH
   return o;
 }
 

@@ -70,6 +70,12 @@ class StaticLayerCache:
                 ext.rope_decode_(q, ow.cos, ow.sin, trad, ow.pos)
                 ext.rope_decode_(k, ow.cos, ow.sin, trad, ow.pos)
             v = qkv[..., (Hq + Hkv) * D :].view(B, 1, Hkv, D).contiguous()
+            if ow.kv_bits == 8:
+                ksz = ow.ksz[self.idx].view(B, -1)
+                vsz = ow.vsz[self.idx].view(B, -1)
+                ext.kv_append_q8_(k, v, kc, ksz, vc, vsz, ow.pos)
+                return ext.attn_decode_q8(q, kc, ksz, vc, vsz, ow.pos, ow.part,
+                                          attn.scale)
             ext.kv_append_(k, v, kc, vc, ow.pos)
             return ext.attn_decode(q, kc, vc, ow.pos, ow.part, attn.scale)
 
@@ -91,18 +97,35 @@ class StaticLayerCache:
 
 
 class StaticDecodeState:
-    """Preallocated caches + device position + per-layer handles."""
+    """Preallocated caches + device position + per-layer handles.
 
-    def __init__(self, model, batch: int = 1, max_len: int = 2048):
+    ``kv_bits=8``: int8 cache with per-64-element-group (scale, zero) —
+    dequant FUSED into the decode attention kernel (attn_decode_q8), ~1.8x
+    KV memory saving at kernel speed (the eager quantized cache rebuilt the
+    full bf16 tensors every step)."""
+
+    def __init__(self, model, batch: int = 1, max_len: int = 2048,
+                 kv_bits: int = None):
         args = model.args
         self.model = model
         self.max_len = max_len
+        self.kv_bits = kv_bits
         dev = next(model.parameters()).device
         dtype = next(model.parameters()).dtype
         L = len(model.layers)
         Hq, Hkv, D = args.num_heads, args.num_kv_heads, args.head_dim
-        self.k = torch.zeros(L, batch, max_len, Hkv, D, device=dev, dtype=dtype)
-        self.v = torch.zeros_like(self.k)
+        if kv_bits == 8:
+            assert D % 64 == 0, "int8 KV needs head_dim % 64 == 0"
+            self.k = torch.zeros(L, batch, max_len, Hkv, D, device=dev, dtype=torch.uint8)
+            self.v = torch.zeros_like(self.k)
+            # per-group (scale, zero) as float pairs
+            self.ksz = torch.zeros(L, batch, max_len, Hkv, D // 64, 2,
+                                   device=dev, dtype=torch.float32)
+            self.vsz = torch.zeros_like(self.ksz)
+        else:
+            assert kv_bits is None, "kv_bits must be None or 8"
+            self.k = torch.zeros(L, batch, max_len, Hkv, D, device=dev, dtype=dtype)
+            self.v = torch.zeros_like(self.k)
         self.pos = torch.zeros(1, dtype=torch.int32, device=dev)
         nc = (max_len + 255) // 256
         self.part = torch.empty(batch, Hq, nc, D + 2, dtype=torch.float32, device=dev)
@@ -119,14 +142,15 @@ class GraphDecoder:
     via the device position salt)."""
 
     def __init__(self, model, batch: int = 1, max_len: int = 2048,
-                 temperature: float = 0.0, min_p: float = 0.0, seed: int = 0):
+                 temperature: float = 0.0, min_p: float = 0.0, seed: int = 0,
+                 kv_bits: int = None):
         self.model = model
         self.batch = batch
         self.max_len = max_len
         self.temperature = float(temperature)
         self.min_p = float(min_p)
         self.seed = int(seed)
-        self.state = StaticDecodeState(model, batch, max_len)
+        self.state = StaticDecodeState(model, batch, max_len, kv_bits=kv_bits)
         dev = next(model.parameters()).device
         self.dev = dev
         self.on_gpu = dev.type == "cuda"
@@ -160,12 +184,31 @@ class GraphDecoder:
         cache = make_prompt_cache(model)
         logits = model(prompt_tokens.to(self.dev), cache=cache)
         for i, c in enumerate(cache):
-            self.state.k[i][:, :P] = c.k
-            self.state.v[i][:, :P] = c.v
+            if self.state.kv_bits == 8:
+                self._bulk_quantize(i, c.k, c.v, P)
+            else:
+                self.state.k[i][:, :P] = c.k
+                self.state.v[i][:, :P] = c.v
         self.state.pos.fill_(P)
         first = self._pick(logits[:, -1])
         self.in_tok.copy_(first.unsqueeze(1))
         self.ring[P % self.max_len] = first
+
+    @torch.no_grad()
+    def _bulk_quantize(self, layer: int, k: torch.Tensor, v: torch.Tensor, P: int) -> None:
+        """Quantize prompt K/V into the int8 cache (same 64-group scheme as
+        the kv_append_q8 kernel)."""
+        for t, codes, sz in ((k, self.state.k[layer], self.state.ksz[layer]),
+                             (v, self.state.v[layer], self.state.vsz[layer])):
+            B, _, H, D = t.shape
+            g = t.float().reshape(B, P, H, D // 64, 64)
+            lo = g.amin(-1)
+            hi = g.amax(-1)
+            scale = ((hi - lo) / 255.0).clamp_min(1e-8)
+            q = ((g - lo.unsqueeze(-1)) / scale.unsqueeze(-1)).round().clamp(0, 255)
+            codes[:, :P] = q.reshape(B, P, H, D).to(torch.uint8)
+            sz[:, :P, :, :, 0] = scale
+            sz[:, :P, :, :, 1] = lo
 
     @torch.no_grad()
     def _step(self) -> None:
