@@ -90,6 +90,27 @@ class StaticLayerCache:
         p = int(ow.pos.item())
         q = rope_ref(q, ow.cos, ow.sin, trad, offset=p)
         k = rope_ref(k, ow.cos, ow.sin, trad, offset=p)
+        if ow.kv_bits == 8:
+            # mirror of the kv_append_q8/attn_decode_q8 kernels in torch
+            def quant_row(t, codes, sz):
+                g = t.float().reshape(B, 1, Hkv, D // 64, 64)
+                lo = g.amin(-1)
+                s = ((g.amax(-1) - lo) / 255.0).clamp_min(1e-8)
+                codes[:, p : p + 1] = ((g - lo.unsqueeze(-1)) / s.unsqueeze(-1))                     .round().clamp(0, 255).reshape(B, 1, Hkv, D).to(torch.uint8)
+                sz[:, p : p + 1, :, :, 0] = s
+                sz[:, p : p + 1, :, :, 1] = lo
+
+            def dequant(codes, sz, n):
+                g = codes[:, :n].float().reshape(B, n, Hkv, D // 64, 64)
+                s = sz[:, :n, :, :, 0].unsqueeze(-1)
+                lo = sz[:, :n, :, :, 1].unsqueeze(-1)
+                return (g * s + lo).reshape(B, n, Hkv, D).to(q.dtype)
+
+            quant_row(k, kc, ow.ksz[self.idx])
+            quant_row(v, vc, ow.vsz[self.idx])
+            kd = dequant(kc, ow.ksz[self.idx], p + 1)
+            vd = dequant(vc, ow.vsz[self.idx], p + 1)
+            return attention_ref(q, kd, vd, causal=True, scale=attn.scale)
         kc[:, p : p + 1] = k
         vc[:, p : p + 1] = v
         return attention_ref(q, kc[:, : p + 1], vc[:, : p + 1], causal=True,

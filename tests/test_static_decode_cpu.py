@@ -57,3 +57,22 @@ def test_static_decode_pos_tracking():
     assert int(dec.state.pos.item()) == 7
     # cache rows 0..6 populated (prompt 3 + 4 decoded appends)
     assert dec.state.k[0][:, :7].abs().sum() > 0
+
+
+def test_int8_kv_cpu_reference_path():
+    """kv_bits=8 eager reference: decodes stay in-vocab and mostly agree
+    with the fp path (int8 perturbs logits slightly)."""
+    torch.manual_seed(0)
+    args = ModelArgs(hidden_size=128, intermediate_size=256, num_layers=2,
+                     num_heads=2, num_kv_heads=2, head_dim=64, vocab_size=97)
+    model = Model(args).eval()
+    prompt = [3, 17, 41, 5]
+    ref = GraphDecoder(model, batch=1, max_len=64)
+    ref.prefill(torch.tensor([prompt]))
+    a = ref.decode(10)[0].tolist()
+    q8 = GraphDecoder(model, batch=1, max_len=64, kv_bits=8)
+    q8.prefill(torch.tensor([prompt]))
+    b = q8.decode(10)[0].tolist()
+    assert all(0 <= t < 97 for t in b)
+    agree = sum(x == y for x, y in zip(a, b)) / len(a)
+    assert agree >= 0.6, (agree, a, b)
