@@ -112,46 +112,31 @@ __global__ __launch_bounds__(256) void gemv_ex_kernel(
   }
   __syncthreads();
 
-  // row loop: each wave owns RU consecutive rows — RU W-loads in flight per
-  // lane (the naive 1-row loop measured 82% WAIT_ANY), and each x_lds read
-  // is shared across the RU rows.
-  constexpr int RU = 4;
   const int wid = tid / WAVE, lane = tid % WAVE;
-  for (int row0 = (blockIdx.x * 4 + wid) * RU; row0 < N; row0 += gridDim.x * 4 * RU) {
-    float acc[RU][BB];
+  for (int row = blockIdx.x * 4 + wid; row < N; row += gridDim.x * 4) {
+    const __hip_bfloat16* wr = W + (long)row * K;
+    float acc[BB];
 #pragma unroll
-    for (int u = 0; u < RU; ++u)
-#pragma unroll
-      for (int b = 0; b < BB; ++b) acc[u][b] = 0.f;
+    for (int b = 0; b < BB; ++b) acc[b] = 0.f;
     for (int k = lane * 8; k + 8 <= K; k += WAVE * 8) {
-      U4 wv[RU];
-#pragma unroll
-      for (int u = 0; u < RU; ++u)
-        wv[u].u = (row0 + u < N)
-                      ? *reinterpret_cast<const uint4*>(W + (long)(row0 + u) * K + k)
-                      : uint4{0, 0, 0, 0};
+      U4 wv;
+      wv.u = *reinterpret_cast<const uint4*>(wr + k);
 #pragma unroll
       for (int b = 0; b < BB; ++b) {
         U4 xv;
         xv.u = *reinterpret_cast<const uint4*>(x_lds + (long)b * K + k);
 #pragma unroll
-        for (int u = 0; u < RU; ++u)
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            acc[u][b] += bf16_bits_to_f32(wv[u].s[j]) * bf16_bits_to_f32(xv.s[j]);
+        for (int j = 0; j < 8; ++j)
+          acc[b] += bf16_bits_to_f32(wv.s[j]) * bf16_bits_to_f32(xv.s[j]);
       }
     }
 #pragma unroll
-    for (int u = 0; u < RU; ++u) {
-      if (row0 + u >= N) break;
-#pragma unroll
-      for (int b = 0; b < BB; ++b) {
-        if (b >= B) break;
-        float a = wave_reduce_sum(acc[u][b]);
-        if (lane == 0) {
-          if (res != nullptr) a += to_f32(res[(long)b * N + row0 + u]);
-          from_f32(&y[(long)b * N + row0 + u], a);
-        }
+    for (int b = 0; b < BB; ++b) {
+      if (b >= B) break;
+      float a = wave_reduce_sum(acc[b]);
+      if (lane == 0) {
+        if (res != nullptr) a += to_f32(res[(long)b * N + row]);
+        from_f32(&y[(long)b * N + row], a);
       }
     }
   }
@@ -175,7 +160,7 @@ at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double 
   sizes.back() = N;
   auto y = at::empty({B, N}, x.options());
   auto stream = at::cuda::getCurrentHIPStream();
-  const int grid_x = std::min(cdiv(N, 16), 2048);  // 4 waves x RU=4 rows
+  const int grid_x = std::min(cdiv(N, 4), 2048);
   auto* Wp = reinterpret_cast<const __hip_bfloat16*>(W.data_ptr());
   auto* np = mode == 1 ? reinterpret_cast<const __hip_bfloat16*>(nw.contiguous().data_ptr())
                        : nullptr;
