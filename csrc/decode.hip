@@ -278,6 +278,8 @@ __global__ __launch_bounds__(256) void attn_decode_partial_q8_kernel(
     const unsigned char* __restrict__ vc, const float2* __restrict__ vsz,
     const int* __restrict__ pos, float* __restrict__ part,
     int B, int Lmax, int Hq, int Hkv, float scale2) {
+  // same MLP-pipelined structure as the bf16 kernel (UNR loads in flight),
+  // with the int8 group dequant fused into both passes
   const int chunk = blockIdx.x, hq = blockIdx.y, b = blockIdx.z;
   const int NC = gridDim.x;
   const int hkv = hq / (Hq / Hkv);
@@ -296,6 +298,13 @@ __global__ __launch_bounds__(256) void attn_decode_partial_q8_kernel(
   __syncthreads();
 
   const __hip_bfloat16* qp = q + ((long)b * Hq + hq) * D;
+  const long rs = (long)Hkv * D;
+  const unsigned char* kbase = kc + ((long)b * Lmax) * rs + (long)hkv * D;
+  const unsigned char* vbase = vc + ((long)b * Lmax) * rs + (long)hkv * D;
+  const long szrs = (long)Hkv * (D / 64);
+  const float2* kszb = ksz + ((long)b * Lmax) * szrs + (long)hkv * (D / 64);
+  const float2* vszb = vsz + ((long)b * Lmax) * szrs + (long)hkv * (D / 64);
+
   const int lpr = D / 8;
   const int rpw = WAVE / lpr;
   const int wid = tid / WAVE, lane = tid % WAVE;
@@ -304,21 +313,29 @@ __global__ __launch_bounds__(256) void attn_decode_partial_q8_kernel(
   U4 qv;
   qv.u = *reinterpret_cast<const uint4*>(qp + dl);
 
-  for (int r0 = k0 + wid * rpw; r0 < kend; r0 += 4 * rpw) {
-    const int krow = r0 + sub;
-    float acc = 0.f;
-    if (krow < kend) {
-      const long row = ((long)b * Lmax + krow) * Hkv + hkv;
-      const float2 sz = ksz[row * (D / 64) + dl / 64];
-      const unsigned char* kr = kc + row * D + dl;
-      uint2 codes = *reinterpret_cast<const uint2*>(kr);
-      const unsigned char* cb = reinterpret_cast<const unsigned char*>(&codes);
+  constexpr int UNR = 4;
+  for (int r0 = k0 + wid * rpw; r0 < kend; r0 += 4 * rpw * UNR) {
+    uint2 codes[UNR];
+    float2 sz[UNR];
+    int krow[UNR];
+#pragma unroll
+    for (int u = 0; u < UNR; ++u) {
+      krow[u] = r0 + u * 4 * rpw + sub;
+      const bool ok = krow[u] < kend;
+      codes[u] = ok ? *reinterpret_cast<const uint2*>(kbase + (long)krow[u] * rs + dl)
+                    : uint2{0, 0};
+      sz[u] = ok ? kszb[(long)krow[u] * szrs + dl / 64] : float2{0.f, 0.f};
+    }
+#pragma unroll
+    for (int u = 0; u < UNR; ++u) {
+      const unsigned char* cb = reinterpret_cast<const unsigned char*>(&codes[u]);
+      float acc = 0.f;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        acc += bf16_bits_to_f32(qv.s[j]) * (cb[j] * sz.x + sz.y);
+        acc += bf16_bits_to_f32(qv.s[j]) * (cb[j] * sz[u].x + sz[u].y);
+      for (int off = lpr / 2; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
+      if (krow[u] < kend && (lane % lpr) == 0) s_row[krow[u] - k0] = acc * scale2;
     }
-    for (int off = lpr / 2; off > 0; off >>= 1) acc += __shfl_xor(acc, off);
-    if (krow < kend && (lane % lpr) == 0) s_row[krow - k0] = acc * scale2;
   }
   __syncthreads();
 
@@ -338,14 +355,35 @@ __global__ __launch_bounds__(256) void attn_decode_partial_q8_kernel(
   l = block_reduce_sum<256>(l, scratch);
 
   const int nrows = kend - k0;
-  for (int d = tid; d < D; d += 256) {
-    float acc = 0.f;
-    const int grp = d / 64;
-    for (int r = 0; r < nrows; ++r) {
-      const long row = ((long)b * Lmax + k0 + r) * Hkv + hkv;
-      const float2 sz = vsz[row * (D / 64) + grp];
-      acc += s_row[r] * (vc[row * D + d] * sz.x + sz.y);
+  const int d = tid % D;
+  const int ph = tid / D;
+  const int nph = 256 / D;
+  const int grp = d / 64;
+  float acc = 0.f;
+  {
+    int r = ph;
+    for (; r + 8 * nph <= nrows; r += 8 * nph) {
+      float vals[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const long row = (long)(k0 + r + u * nph);
+        const float2 s = vszb[row * szrs + grp];
+        vals[u] = vbase[row * rs + d] * s.x + s.y;
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u) acc += s_row[r + u * nph] * vals[u];
     }
+    for (; r < nrows; r += nph) {
+      const long row = (long)(k0 + r);
+      const float2 s = vszb[row * szrs + grp];
+      acc += s_row[r] * (vbase[row * rs + d] * s.x + s.y);
+    }
+  }
+  __syncthreads();
+  if (ph > 0) s_row[(ph - 1) * D + d] = acc;
+  __syncthreads();
+  if (ph == 0) {
+    for (int p = 1; p < nph; ++p) acc += s_row[(p - 1) * D + d];
     out[d] = acc;
   }
   if (tid == 0) { out[D] = m; out[D + 1] = l; }
