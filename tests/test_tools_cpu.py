@@ -131,3 +131,25 @@ def test_model_cli_list_details_load(tmp_path, capsys):
     cli.cmd_generate("hello")
     out = capsys.readouterr().out
     assert "tok/s" in out
+
+
+def test_export_loads_in_hf_transformers(tmp_path):
+    """The exported directory must load through HF transformers' own Llama
+    and produce logits matching our model — the real interop check behind
+    the mlx-lm/HF export format."""
+    transformers = pytest.importorskip("transformers")
+    from convert_to_mlx_lm import convert_run
+
+    trainer, run_dir = _tiny_run(tmp_path, name="tiny-hf")
+    out = convert_run(run_dir, tmp_path / "hf_export")
+
+    # config.json advertises bf16; load fp32 for an exact numerics compare
+    hf_model = transformers.AutoModelForCausalLM.from_pretrained(
+        str(out), torch_dtype=torch.float32)
+    hf_model.eval()
+    toks = torch.randint(3, 50, (1, 16))
+    with torch.no_grad():
+        ours = trainer.model(toks).float()
+        theirs = hf_model(toks).logits.float()
+    assert torch.allclose(ours, theirs, atol=2e-3, rtol=2e-3), \
+        (ours - theirs).abs().max()

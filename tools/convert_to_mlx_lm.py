@@ -86,7 +86,7 @@ def hf_config_from(config: Config, args: ModelArgs) -> Dict:
         "num_key_value_heads": args.num_kv_heads,
         "head_dim": args.head_dim,
         "vocab_size": args.vocab_size,
-        "max_position_embeddings": args.max_position_embeddings,
+        "max_position_embeddings": int(args.max_position_embeddings or 2048),
         "rms_norm_eps": args.rms_norm_eps,
         "rope_theta": args.rope_theta,
         "tie_word_embeddings": args.tie_word_embeddings,
@@ -139,7 +139,7 @@ def convert_run(run_dir: str | Path, out_path: str | Path, checkpoint: str = "fi
         "pad_token": special.get("pad", "<pad>"),
         "add_bos_token": True,
         "add_eos_token": False,
-        "model_max_length": args.max_position_embeddings,
+        "model_max_length": int(args.max_position_embeddings or 2048),
     }
     (out / "tokenizer_config.json").write_text(json.dumps(tokenizer_config, indent=2))
 
