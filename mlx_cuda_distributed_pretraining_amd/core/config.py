@@ -124,6 +124,10 @@ class LoggingConfig:
     wandb_project: Optional[str] = None
     wandb_entity: Optional[str] = None
     log_memory_usage: bool = False
+    # per-step phase breakdown (data/compute/optimizer ms) appended to the
+    # log line; measured with CUDA events, synchronized
+    # only at logging steps
+    log_step_breakdown: bool = False
     log_gradient_norm: bool = False
     log_parameter_norm: bool = False
     log_samples: bool = False

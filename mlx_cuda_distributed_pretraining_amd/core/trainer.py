@@ -248,8 +248,16 @@ class Trainer:
 
     def train_step(self, step: int) -> tuple:
         """One optimizer step (with gradient accumulation). Returns
-        (loss_detached, ntokens)."""
+        (loss_detached, ntokens). When ``logging.log_step_breakdown`` is on,
+        fills self.last_breakdown = (data_ms, compute_ms, optim_ms) — CUDA
+        events, resolved lazily (synchronizes only when read at log time)."""
         lr = self.lr_schedule(step)
+        breakdown = bool(self.config.logging.log_step_breakdown)
+        on_gpu = self.device.type == "cuda"
+        if breakdown and on_gpu:
+            ev = [torch.cuda.Event(enable_timing=True) for _ in range(3)]
+        t_host0 = time.perf_counter()
+        t_data = 0.0
         if self.ddp is not None:
             self.ddp.require_reduce = False
         total_loss = None
@@ -261,9 +269,13 @@ class Trainer:
         for micro in range(self.grad_accum_steps):
             if self.ddp is not None and micro == self.grad_accum_steps - 1:
                 self.ddp.require_reduce = True
+            t_d0 = time.perf_counter()
             batch = self.data_manager.generate_batch(
                 step * self.grad_accum_steps + micro
             ).to(self.device, non_blocking=True)
+            t_data += time.perf_counter() - t_d0
+            if breakdown and on_gpu and micro == 0:
+                ev[0].record()
             inputs, targets = batch[:, :-1], batch[:, 1:]
             loss, ntok = self.compute_loss(inputs, targets)
             (loss / self.grad_accum_steps).backward()
@@ -272,6 +284,8 @@ class Trainer:
             total_tok = ntok if total_tok is None else total_tok + ntok
         if self.ddp is not None:
             self.ddp.finalize()
+        if breakdown and on_gpu:
+            ev[1].record()
 
         if isinstance(self.optimizer, FusedFlatAdamW):
             self.optimizer.step(lr=lr)
@@ -289,7 +303,27 @@ class Trainer:
                     for group in o.param_groups:
                         group["lr"] = lr
             self.optimizer.step()
+        if breakdown:
+            if on_gpu:
+                ev[2].record()
+                self.last_breakdown = (t_data, ev, t_host0)
+            else:
+                self.last_breakdown = (t_data, None, t_host0)
         return total_loss / self.grad_accum_steps, total_tok
+
+    def format_breakdown(self) -> str:
+        """Resolve the last step's phase timings (synchronizes the GPU)."""
+        if not getattr(self, "last_breakdown", None):
+            return ""
+        t_data, ev, t_host0 = self.last_breakdown
+        if ev is not None:
+            ev[2].synchronize()
+            compute_ms = ev[0].elapsed_time(ev[1])
+            optim_ms = ev[1].elapsed_time(ev[2])
+        else:
+            total = (time.perf_counter() - t_host0) * 1000
+            compute_ms, optim_ms = total - t_data * 1000, 0.0
+        return f" | data_ms={t_data*1e3:.1f} | compute_ms={compute_ms:.1f} | optim_ms={optim_ms:.1f}"
 
     # ------------------------------------------------------------------
     @torch.no_grad()
@@ -539,6 +573,8 @@ class Trainer:
                     * self.grad_accum_steps
                     * self.world_size,
                 )
+                if cfg.logging.log_step_breakdown:
+                    line += self.format_breakdown()
                 if cfg.logging.log_gradient_norm and self.flat_space is not None:
                     gnorm = float(self.flat_space.flat_grad.float().norm().item())
                     line += f" | grad_norm={gnorm:.3e}"
