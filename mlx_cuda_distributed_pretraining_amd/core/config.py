@@ -157,7 +157,6 @@ class SystemConfig:
     # MI355X extensions
     distributed_backend: str = "nccl"  # "nccl" is RCCL on ROCm
     bucket_mb: int = 50
-    compile_step: bool = False  # capture the train step in a hipGraph
 
 
 @dataclass
