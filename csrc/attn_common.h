@@ -58,6 +58,31 @@ __device__ __forceinline__ void acc_to_afrag(const float (&p)[16], bf16x8& f0, b
   }
 }
 
+// Cross-half (lane ^ 32) exchange WITHOUT ds_bpermute: __shfl_xor(x, 32)
+// compiles to ds_bpermute + an lgkmcnt(0) drain, which ALSO waits for any
+// prefetched ds_reads in flight (lgkmcnt counts both) — a serializer in the
+// online-softmax critical path. v_permlane32_swap_b32 is a VALU op:
+// permlane32_swap(x, x) returns r0 = {x.lo | x.lo}, r1 = {x.hi | x.hi}
+// (each lane's r0/r1 hold the low/high-half values of its column), so
+// max(r0, r1) / r0 + r1 IS the cross-half reduction in every lane.
+__device__ __forceinline__ float cross32_max(float x) {
+  union { float f; unsigned u; } a, r0, r1;
+  a.f = x;
+  auto r = __builtin_amdgcn_permlane32_swap(a.u, a.u, false, false);
+  r0.u = r[0];
+  r1.u = r[1];
+  return fmaxf(r0.f, r1.f);
+}
+
+__device__ __forceinline__ float cross32_sum(float x) {
+  union { float f; unsigned u; } a, r0, r1;
+  a.f = x;
+  auto r = __builtin_amdgcn_permlane32_swap(a.u, a.u, false, false);
+  r0.u = r[0];
+  r1.u = r[1];
+  return r0.f + r1.f;
+}
+
 // row index held by (reg, hi) in a 32x32 accumulator
 __device__ __forceinline__ int acc_row(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;

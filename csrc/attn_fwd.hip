@@ -235,7 +235,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
           tmax = fmaxf(tmax, p[kt][reg]);
         }
     }
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 32));
+    tmax = cross32_max(tmax);
     // defer-max (guide T13): while no lane's tile max exceeds the running
     // max by more than THR (base-2), keep the old max and SKIP the O/l
     // rescale entirely — P is then bounded by 2^THR, which the fp32
@@ -253,7 +253,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
         p[kt][reg] = __builtin_amdgcn_exp2f(p[kt][reg] - mc);
         psum += p[kt][reg];
       }
-    psum += __shfl_xor(psum, 32);
+    psum = cross32_sum(psum);
     if (defer) {
       l += psum;
     } else {
