@@ -94,11 +94,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
   constexpr int QPB = 32 * NW;
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
-  constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K/V row-major images
-  constexpr int TROW = 64;                     // K^T image row (elements; swzt())
-  constexpr int TILE = KVB * D * 2 + D * TROW; // K rm + V rm + K^T
+  // K rm + V rm only: the dQ-accumulate B-fragments (K, k-strided) are read
+  // straight from the row-major K image with ds_read_b64_tr_b16 (tr16_frag),
+  // so no transposed K image is staged.
+  constexpr int TILE = KVB * D * 2;
 
-  __shared__ __hip_bfloat16 smem[2 * TILE];
+  __shared__ __align__(16) __hip_bfloat16 smem[2 * TILE];
 
   const int b = blockIdx.z, hq = blockIdx.y, qtile = blockIdx.x;
   const int hkv = hq / (Hq / Hkv);
@@ -144,8 +145,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
     kv_hi = min(Skv, max(blk_qpos_hi + 1, modarg));
   }
 
-  // T14 staging: chunks of (2 kv rows x 8 d). K chunks write row-major twice
-  // (plain + pair-transposed); V chunks write row-major.
+  // T14 staging: chunks of (2 kv rows x 8 d); K and V both write row-major.
   constexpr int CH_TOT = (KVB / 2) * (D / 8);  // chunks per tile
   constexpr int NCH = (CH_TOT + TPB - 1) / TPB;
   uint4 kreg[NCH][2], vreg[NCH][2];
@@ -170,26 +170,16 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
   auto stage_write = [&](int bufsel) {
     __hip_bfloat16* k_lds = smem + bufsel * TILE;
     __hip_bfloat16* v_lds = k_lds + KVB * D;
-    __hip_bfloat16* kt_lds = v_lds + KVB * D;
 #pragma unroll
     for (int c = 0; c < NCH; ++c) {
       const int u = tid + c * TPB;
       const int row = (u / (D / 8)) * 2;
       const int d0 = (u % (D / 8)) * 8;
       if (row >= KVB) continue;
-      *reinterpret_cast<uint4*>(k_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = kreg[c][0];
-      *reinterpret_cast<uint4*>(k_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = kreg[c][1];
-      *reinterpret_cast<uint4*>(v_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = vreg[c][0];
-      *reinterpret_cast<uint4*>(v_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = vreg[c][1];
-      Bf16x8U k0, k1;
-      *reinterpret_cast<uint4*>(k0.s) = kreg[c][0];
-      *reinterpret_cast<uint4*>(k1.s) = kreg[c][1];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int drw = d0 + j;
-        const uint pair = (uint)k0.s[j] | ((uint)k1.s[j] << 16);
-        *reinterpret_cast<uint*>(kt_lds + drw * TROW + (row ^ swzt(drw))) = pair;
-      }
+      *reinterpret_cast<uint4*>(k_lds + rm_swz<D>(row, d0)) = kreg[c][0];
+      *reinterpret_cast<uint4*>(k_lds + rm_swz<D>(row + 1, d0)) = kreg[c][1];
+      *reinterpret_cast<uint4*>(v_lds + rm_swz<D>(row, d0)) = vreg[c][0];
+      *reinterpret_cast<uint4*>(v_lds + rm_swz<D>(row + 1, d0)) = vreg[c][1];
     }
   };
 
@@ -210,17 +200,16 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
 
     const __hip_bfloat16* k_lds = smem + buf * TILE;
     const __hip_bfloat16* v_lds = k_lds + KVB * D;
-    const __hip_bfloat16* kt_lds = v_lds + KVB * D;
 
     // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
     f32x16 st = {}, dpt = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dblk = 0; dblk < DBLK; ++dblk) {
-      const int col = (dblk * 16 + hi * 8) ^ ((lq & KSWZ) << 3);
+      const int off = rm_swz<D>(lq, dblk * 16 + hi * 8);
       Bf16x8U kf, vf;
-      *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(k_lds + lq * D + col);
-      *reinterpret_cast<uint4*>(vf.s) = *reinterpret_cast<const uint4*>(v_lds + lq * D + col);
+      *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(k_lds + off);
+      *reinterpret_cast<uint4*>(vf.s) = *reinterpret_cast<const uint4*>(v_lds + off);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
       dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf.v, dof[dblk], dpt, 0, 0, 0);
     }
@@ -269,11 +258,9 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
       for (int r = 0; r < 16; ++r) acc[r] = dq_acc[dc][r];
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        const int drw = dc * 32 + lq;
-        Bf16x8U kb;  // B = K[16k x 32d] from K^T image
-        *reinterpret_cast<uint4*>(kb.s) = *reinterpret_cast<const uint4*>(
-            kt_lds + drw * TROW + ((ks * 16 + hi * 8) ^ swzt(drw)));
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb.v, acc, 0, 0, 0);
+        // B = K[16k x 32d]: tr_read from the row-major K image
+        bf16x8 kb = tr16_frag<D>(k_lds, ks * 16 + hi * 8, dc * 32, lane);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb, acc, 0, 0, 0);
       }
 #pragma unroll
       for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
@@ -315,12 +302,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   constexpr int KPB = 32 * NW;
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
-  constexpr int KSWZ = (D >= 128) ? 15 : 7;  // row-major q/do images
-  constexpr int TROW = 64;                   // X^T image row (elements; swzt())
-  // q-tile images: Q rm; X^T (X = Q for dK, dO for dV); dO rm (dK only)
-  constexpr int TILE = 32 * D + D * TROW + (WANT_DK ? 32 * D : 0);
+  // q-tile images: Q rm + dO rm. The accumulate B-fragments (Q for dK,
+  // dO for dV — k-strided) are tr16_frag reads from the row-major images,
+  // so no transposed copy is staged.
+  constexpr int TILE = 2 * 32 * D;
 
-  __shared__ __hip_bfloat16 smem[2 * TILE];
+  __shared__ __align__(16) __hip_bfloat16 smem[2 * TILE];
   __shared__ float stats_lds[2][2][32];  // [buf][L|D][qrow]
 
   const int b = blockIdx.z, hkv = blockIdx.y, kvtile = blockIdx.x;
@@ -383,32 +370,14 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   };
   auto stage_write = [&](int bufsel) {
     __hip_bfloat16* q_lds = smem + bufsel * TILE;
-    __hip_bfloat16* xt_lds = q_lds + 32 * D;
-    __hip_bfloat16* do_lds = xt_lds + D * TROW;  // only sized/used when WANT_DK
+    __hip_bfloat16* do_lds = q_lds + 32 * D;
     const int u = is_q_half ? tid : tid - CH_TOT;
     if (u >= CH_TOT) return;
     const int row = (u / (D / 8)) * 2;
     const int d0 = (u % (D / 8)) * 8;
-    // row-major images: Q always; dO only for dK
-    if (is_q_half) {
-      *reinterpret_cast<uint4*>(q_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = sreg[0];
-      *reinterpret_cast<uint4*>(q_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = sreg[1];
-    } else if constexpr (WANT_DK) {
-      *reinterpret_cast<uint4*>(do_lds + row * D + (d0 ^ ((row & KSWZ) << 3))) = sreg[0];
-      *reinterpret_cast<uint4*>(do_lds + (row + 1) * D + (d0 ^ (((row + 1) & KSWZ) << 3))) = sreg[1];
-    }
-    // transposed image: Q^T for dK, dO^T for dV
-    if (is_q_half == WANT_DK) {
-      Bf16x8U x0, x1;
-      *reinterpret_cast<uint4*>(x0.s) = sreg[0];
-      *reinterpret_cast<uint4*>(x1.s) = sreg[1];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int drw = d0 + j;
-        const uint pair = (uint)x0.s[j] | ((uint)x1.s[j] << 16);
-        *reinterpret_cast<uint*>(xt_lds + drw * TROW + (row ^ swzt(drw))) = pair;
-      }
-    }
+    __hip_bfloat16* dst = is_q_half ? q_lds : do_lds;
+    *reinterpret_cast<uint4*>(dst + rm_swz<D>(row, d0)) = sreg[0];
+    *reinterpret_cast<uint4*>(dst + rm_swz<D>(row + 1, d0)) = sreg[1];
   };
 
   // iterate (GQA head, q tile) pairs with a flat prefetch pipeline
@@ -449,8 +418,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
       }
 
       const __hip_bfloat16* q_lds = smem + buf * TILE;
-      const __hip_bfloat16* xt_lds = q_lds + 32 * D;
-      const __hip_bfloat16* do_lds = xt_lds + D * TROW;
+      const __hip_bfloat16* do_lds = q_lds + 32 * D;
 
       // S = mfma(Q, K^T): A=Q rm frags from LDS, B = register kf.
       // element (r=q_local, c=k_local)
@@ -458,13 +426,13 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int dblk = 0; dblk < DBLK; ++dblk) {
-        const int col = (dblk * 16 + hi * 8) ^ ((lk & KSWZ) << 3);
+        const int off = rm_swz<D>(lk, dblk * 16 + hi * 8);
         Bf16x8U qa;
-        *reinterpret_cast<uint4*>(qa.s) = *reinterpret_cast<const uint4*>(q_lds + lk * D + col);
+        *reinterpret_cast<uint4*>(qa.s) = *reinterpret_cast<const uint4*>(q_lds + off);
         s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa.v, kf[dblk], s_acc, 0, 0, 0);
         if constexpr (WANT_DK) {
           Bf16x8U da;
-          *reinterpret_cast<uint4*>(da.s) = *reinterpret_cast<const uint4*>(do_lds + lk * D + col);
+          *reinterpret_cast<uint4*>(da.s) = *reinterpret_cast<const uint4*>(do_lds + off);
           dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da.v, vf[dblk], dp_acc, 0, 0, 0);
         }
       }
@@ -526,11 +494,10 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
         for (int r = 0; r < 16; ++r) acc[r] = acc_out[dc][r];
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-          const int drw = dc * 32 + lk;
-          Bf16x8U xb;  // B = X[16q x 32d] from X^T image
-          *reinterpret_cast<uint4*>(xb.s) = *reinterpret_cast<const uint4*>(
-              xt_lds + drw * TROW + ((ks * 16 + hi * 8) ^ swzt(drw)));
-          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb.v, acc, 0, 0, 0);
+          // B = X[16q x 32d] (X = Q for dK, dO for dV): tr_read from rm image
+          bf16x8 xb = tr16_frag<D>(WANT_DK ? q_lds : do_lds, ks * 16 + hi * 8,
+                                   dc * 32, lane);
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb, acc, 0, 0, 0);
         }
 #pragma unroll
         for (int r = 0; r < 16; ++r) acc_out[dc][r] = acc[r];

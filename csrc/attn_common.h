@@ -88,6 +88,69 @@ __device__ __forceinline__ int acc_row(int reg, int hi) {
   return (reg & 3) + 8 * (reg >> 2) + 4 * hi;
 }
 
+// ---- ds_read_b64_tr_b16 (gfx950 hardware transpose read, guide T10) ----
+//
+// Semantics (probed on hardware, probe/tr16_probe.hip): per 16-lane group,
+// each lane reads 4 contiguous bf16 (8 B, address MUST be 8-B aligned:
+// misaligned returns the aligned address's data with no fault); the group's
+// 64 elements form a window and dest lane i slot j receives window[16j + i]
+// (window order = source lane i4 contributes window[4*i4 .. 4*i4+3]).
+//
+// Used here to read an MFMA B-fragment X[16 rows x 32 cols] DIRECTLY from a
+// row-major LDS image (no transposed copy): dest lane l wants
+// X[base + 8*(l>>5) + j][c = l&31]; per tr_read the window is a
+// [4 row][16 col] block, so source lane i4 = l&15 (col group gd = (l>>4)&1)
+// reads rows base + (i4>>2), 4 contiguous cols at 4*(i4&3).
+//
+// Bank structure: the image swizzle below makes BOTH the b128 row reads and
+// the tr_read gather conflict-free (simulated + PMC-checked):
+//   elem(q, d) at q*D + ((d & ~7) ^ (s_tr<D>(q) << 3)) + (d & 7)
+// Constraints satisfied: s_tr bijective on every 16-row window (b128 phase
+// covers all 64 banks) and s_tr>>1 distinct on every aligned 4-row window
+// (the [4 row][16 col] tr gather covers 16 distinct 8-B bank slots).
+template <int D>
+__device__ __forceinline__ int s_tr(int q) {
+  if constexpr (D >= 128) return ((q & 7) << 1) | ((q >> 3) & 1);
+  else return (((q >> 1) & 3) << 1) | ((q >> 3) & 1);
+}
+
+// address (in elements) of the 8-aligned chunk holding cols d8..d8+7 of row q
+template <int D>
+__device__ __forceinline__ int rm_swz(int q, int d8) {
+  return q * D + (d8 ^ (s_tr<D>(q) << 3));
+}
+
+// two tr_reads -> one bf16x8 B-operand fragment (rows qb..qb+7, col lane&31)
+template <int D>
+__device__ __forceinline__ bf16x8 tr16_frag(const __hip_bfloat16* img, int qb,
+                                            int dcol0, int lane) {
+  const int i4 = lane & 15, gd = (lane >> 4) & 1;
+  const int dch = dcol0 + gd * 16 + 4 * (i4 & 3);
+  const int q0 = qb + (i4 >> 2);
+  const unsigned o0 = (unsigned)(unsigned long long)(
+      img + q0 * D + ((dch & ~7) ^ (s_tr<D>(q0) << 3)) + (dch & 4));
+  const unsigned o1 = (unsigned)(unsigned long long)(
+      img + (q0 + 4) * D + ((dch & ~7) ^ (s_tr<D>(q0 + 4) << 3)) + (dch & 4));
+  uint2 a, b;
+  // "=&v" (early-clobber) is REQUIRED: ds_read is asynchronous, so without it
+  // the allocator may alias an output pair with the second read's address
+  // register, and the first read's writeback can corrupt that address if the
+  // wave stalls between the two issues (observed: non-deterministic scattered
+  // errors under multi-wave load; single-wave self-tests pass).
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(a), "=&v"(b)
+      : "v"(o0), "v"(o1));
+  Bf16x8U out;
+  out.u[0] = a.x;
+  out.u[1] = a.y;
+  out.u[2] = b.x;
+  out.u[3] = b.y;
+  return out.v;
+}
+
 // Swizzle (element units) for TRANSPOSED LDS images stored as [D][64] rows
 // (128 B rows). ds_write bank = (byte/4) % 32, so the row base (drow*32
 // dwords) contributes nothing — the spread must come from the column offset.

@@ -63,6 +63,7 @@ std::vector<at::Tensor> fp8_quantize(at::Tensor x, bool transpose);
 // debug.hip
 at::Tensor mfma_tile_test(at::Tensor A, at::Tensor B);
 at::Tensor afrag_transform_test(at::Tensor M);
+at::Tensor tr16_frag_test(at::Tensor X);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (y, rstd)");
@@ -97,4 +98,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fp8_quantize", &fp8_quantize, "fused bf16 -> e4m3 quantize (codes, scale)");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");
+  m.def("tr16_frag_test", &tr16_frag_test, "debug: ds_read_b64_tr_b16 B-fragment gather");
 }

@@ -47,7 +47,56 @@ __global__ void afrag_transform_kernel(const float* __restrict__ M, float* __res
   }
 }
 
+// Verify tr16_frag: stage X[32][D] into the rm_swz image exactly as the bwd
+// kernels do, then emit every (ks, dc) B-fragment; out[ks][dc][k8][c] should
+// equal X[ks*16 + k8][dc*32 + c] if the hardware transpose-read delivery
+// matches the probed semantics (probe/tr16_probe.hip).
+template <int D>
+__global__ void tr16_frag_kernel(const __hip_bfloat16* __restrict__ X,
+                                 float* __restrict__ out) {
+  __shared__ __align__(16) __hip_bfloat16 img[32 * D];
+  const int lane = threadIdx.x % WAVE;
+  const int tid = threadIdx.x;
+  // stage: thread u writes rows 2u/(D/8) pair, 8 cols
+  for (int u = tid; u < (32 / 2) * (D / 8); u += blockDim.x) {
+    const int row = (u / (D / 8)) * 2;
+    const int d0 = (u % (D / 8)) * 8;
+    *reinterpret_cast<uint4*>(img + rm_swz<D>(row, d0)) =
+        *reinterpret_cast<const uint4*>(X + row * D + d0);
+    *reinterpret_cast<uint4*>(img + rm_swz<D>(row + 1, d0)) =
+        *reinterpret_cast<const uint4*>(X + (row + 1) * D + d0);
+  }
+  __syncthreads();
+  if (tid >= WAVE) return;
+  const int hi = lane >> 5;
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks)
+    for (int dc = 0; dc < D / 32; ++dc) {
+      bf16x8 f = tr16_frag<D>(img, ks * 16 + hi * 8, dc * 32, lane);
+      Bf16x8U u;
+      u.v = f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        out[((ks * (D / 32) + dc) * 16 + hi * 8 + j) * 32 + (lane & 31)] = to_f32(u.h[j]);
+    }
+}
+
 }  // namespace
+
+at::Tensor tr16_frag_test(at::Tensor X) {
+  TORCH_CHECK(X.is_cuda() && X.scalar_type() == at::kBFloat16 && X.dim() == 2 && X.size(0) == 32);
+  const int D = X.size(1);
+  TORCH_CHECK(D == 64 || D == 128);
+  auto out = at::zeros({2 * (D / 32), 16, 32}, X.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  if (D == 64)
+    tr16_frag_kernel<64><<<1, 256, 0, stream>>>(
+        reinterpret_cast<const __hip_bfloat16*>(X.contiguous().data_ptr()), out.data_ptr<float>());
+  else
+    tr16_frag_kernel<128><<<1, 256, 0, stream>>>(
+        reinterpret_cast<const __hip_bfloat16*>(X.contiguous().data_ptr()), out.data_ptr<float>());
+  return out;
+}
 
 at::Tensor mfma_tile_test(at::Tensor A, at::Tensor B) {
   TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16);
