@@ -23,7 +23,10 @@
 
 namespace {
 
-constexpr int KVB = 32;        // kv tile per iteration (dQ kernel)
+// kv tile per dQ iteration: 64 rows staged per barrier (two 32-row halves
+// processed sequentially -> 48 MFMAs between barriers instead of 24; viable
+// at 32 KB/buffer now that the transposed K image is gone)
+constexpr int KVB = 64;
 constexpr float LOG2E = 1.4426950408889634f;
 
 // waves per block (8 or 16): 16-wave blocks halve staging traffic and
@@ -198,74 +201,81 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
     const bool has_next = kv0 + KVB < kv_hi;
     if (has_next) stage_load(kv0 + KVB);
 
-    const __hip_bfloat16* k_lds = smem + buf * TILE;
-    const __hip_bfloat16* v_lds = k_lds + KVB * D;
-
-    // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
-    f32x16 st = {}, dpt = {};
-    __builtin_amdgcn_s_setprio(1);
+    // two 32-row halves per staged tile (s_tr only uses low row bits, so the
+    // half-image at rows 32..63 is the same layout at a +32*D offset)
 #pragma unroll
-    for (int dblk = 0; dblk < DBLK; ++dblk) {
-      const int off = rm_swz<D>(lq, dblk * 16 + hi * 8);
-      Bf16x8U kf, vf;
-      *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(k_lds + off);
-      *reinterpret_cast<uint4*>(vf.s) = *reinterpret_cast<const uint4*>(v_lds + off);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
-      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf.v, dof[dblk], dpt, 0, 0, 0);
-    }
-    __builtin_amdgcn_s_setprio(0);
+    for (int hf = 0; hf < 2; ++hf) {
+      const int kvh = kv0 + hf * 32;
+      if (kvh >= kv_hi) break;  // block-uniform (kv_hi is block-level)
+      const __hip_bfloat16* k_lds = smem + buf * TILE + hf * 32 * D;
+      const __hip_bfloat16* v_lds = k_lds + KVB * D;
 
-    // interior tiles: every k row of this tile is kept for every q row of
-    // this wave (wave-uniform branch)
-    bool full = kv0 + KVB <= Skv;
-    if constexpr (MOD == MOD_CAUSAL) {
-      full = full && (kv0 + KVB - 1 <= q0w + q_off);
-    } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
-      full = full && (kv0 + KVB - 1 <= q0w + q_off) &&
-             ((q0w + 31 + q_off) - kv0 < modarg);
-    } else if constexpr (MOD == MOD_PREFIX_LM) {
-      full = full && ((kv0 + KVB - 1 <= q0w + q_off) || (kv0 + KVB <= modarg));
-    } else if constexpr (MOD == MOD_ALIBI) {
-      full = false;
-    }
-
-    float ds[16];
-    if (full) {
+      // S^T = mfma(K, Q); dP^T = mfma(V, dO) — both (r=k_local, c=q_local)
+      f32x16 st = {}, dpt = {};
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        const float p = __builtin_amdgcn_exp2f(st[reg] * scale2 - Lq2);
-        ds[reg] = p * (dpt[reg] - Dq) * scale;
+      for (int dblk = 0; dblk < DBLK; ++dblk) {
+        const int off = rm_swz<D>(lq, dblk * 16 + hi * 8);
+        Bf16x8U kf, vf;
+        *reinterpret_cast<uint4*>(kf.s) = *reinterpret_cast<const uint4*>(k_lds + off);
+        *reinterpret_cast<uint4*>(vf.s) = *reinterpret_cast<const uint4*>(v_lds + off);
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st, 0, 0, 0);
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf.v, dof[dblk], dpt, 0, 0, 0);
       }
-    } else {
-#pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        const int k_pos = kv0 + acc_row(reg, hi);
-        const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
-        float s2 = st[reg] * scale2;
-        if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
-        const float p = keep ? __builtin_amdgcn_exp2f(s2 - Lq2) : 0.f;
-        ds[reg] = p * (dpt[reg] - Dq) * scale;
-      }
-    }
+      __builtin_amdgcn_s_setprio(0);
 
-    bf16x8 da0, da1;
-    acc_to_afrag(ds, da0, da1);  // -> dS[32q x 16k] A-fragments
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int dc = 0; dc < DCOL; ++dc) {
-      f32x16 acc;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) acc[r] = dq_acc[dc][r];
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        // B = K[16k x 32d]: tr_read from the row-major K image
-        bf16x8 kb = tr16_frag<D>(k_lds, ks * 16 + hi * 8, dc * 32, lane);
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb, acc, 0, 0, 0);
+      // interior half-tiles: every k row kept for every q row of this wave
+      // (wave-uniform branch)
+      bool full = kvh + 32 <= Skv;
+      if constexpr (MOD == MOD_CAUSAL) {
+        full = full && (kvh + 31 <= q0w + q_off);
+      } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+        full = full && (kvh + 31 <= q0w + q_off) &&
+               ((q0w + 31 + q_off) - kvh < modarg);
+      } else if constexpr (MOD == MOD_PREFIX_LM) {
+        full = full && ((kvh + 31 <= q0w + q_off) || (kvh + 32 <= modarg));
+      } else if constexpr (MOD == MOD_ALIBI) {
+        full = false;
       }
+
+      float ds[16];
+      if (full) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
+        for (int reg = 0; reg < 16; ++reg) {
+          const float p = __builtin_amdgcn_exp2f(st[reg] * scale2 - Lq2);
+          ds[reg] = p * (dpt[reg] - Dq) * scale;
+        }
+      } else {
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          const int k_pos = kvh + acc_row(reg, hi);
+          const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+          float s2 = st[reg] * scale2;
+          if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
+          const float p = keep ? __builtin_amdgcn_exp2f(s2 - Lq2) : 0.f;
+          ds[reg] = p * (dpt[reg] - Dq) * scale;
+        }
+      }
+
+      bf16x8 da0, da1;
+      acc_to_afrag(ds, da0, da1);  // -> dS[32q x 16k] A-fragments
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int dc = 0; dc < DCOL; ++dc) {
+        f32x16 acc;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) acc[r] = dq_acc[dc][r];
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          // B = K[16k x 32d]: tr_read from the row-major K image
+          bf16x8 kb = tr16_frag<D>(k_lds, ks * 16 + hi * 8, dc * 32, lane);
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? da0 : da1, kb, acc, 0, 0, 0);
+        }
+#pragma unroll
+        for (int r = 0; r < 16; ++r) dq_acc[dc][r] = acc[r];
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
 
     // write tile t+1 LAST: the whole iteration hides the global-load flight
     // (buf^1 was last read before the previous barrier)
