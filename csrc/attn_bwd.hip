@@ -312,13 +312,15 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   constexpr int KPB = 32 * NW;
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
-  // q-tile images: Q rm + dO rm. The accumulate B-fragments (Q for dK,
-  // dO for dV — k-strided) are tr16_frag reads from the row-major images,
-  // so no transposed copy is staged.
-  constexpr int TILE = 2 * 32 * D;
+  // q-tile images: Q rm + dO rm, TWO (GQA head, q tile) iterations staged per
+  // barrier (sub-tiles s=0,1 at +s*32*D inside each 64-row tensor region).
+  // The accumulate B-fragments (Q for dK, dO for dV — k-strided) are
+  // tr16_frag reads from the row-major images, so no transposed copy is
+  // staged.
+  constexpr int TILE = 2 * 64 * D;
 
   __shared__ __align__(16) __hip_bfloat16 smem[2 * TILE];
-  __shared__ float stats_lds[2][2][32];  // [buf][L|D][qrow]
+  __shared__ float stats_lds[2][2][64];  // [buf][L|D][s*32 + qrow]
 
   const int b = blockIdx.z, hkv = blockIdx.y, kvtile = blockIdx.x;
   const int group = Hq / Hkv;
@@ -359,33 +361,39 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   int q_lo, q_hi;
   q_range_for_kv<MOD>(kvtile * KPB, KPB, q_off, Sq, modarg, q_lo, q_hi);
 
-  // T14 staging of one 32-row q tile: threads 0..255 own Q chunks
-  // (2 rows x 8 d), threads 256..511 own dO chunks.
-  constexpr int CH_TOT = (32 / 2) * (D / 8);  // 256 chunks per tensor
+  // sub-tiles per barrier: dV runs two (32 MFMAs/barrier, 220-238 VGPR);
+  // dK holds V fragments too and spills at NP=2, so it stays at one.
+  constexpr int NP = WANT_DK ? 1 : 2;
+
+  // T14 staging of two 32-row q tiles: threads 0..255 own Q chunks
+  // (2 rows x 8 d), threads 256..511 own dO chunks; each thread carries one
+  // chunk pair per sub-tile.
+  constexpr int CH_TOT = (32 / 2) * (D / 8);  // 256 chunks per tensor-tile
   const bool is_q_half = tid < CH_TOT;
-  uint4 sreg[2];
+  uint4 sreg[2];  // ONE sub-tile's chunk pair at a time (split-half staging:
+                  // load s -> compute s-1 -> write s; keeps peak VGPR down)
 
   auto stage_load = [&](int hq_, int q0) {
     const int u = is_q_half ? tid : tid - CH_TOT;
     if (u >= CH_TOT) return;
     const int row = (u / (D / 8)) * 2;
     const int d0 = (u % (D / 8)) * 8;
-    const bool ok0 = q0 + row < Sq;
-    const bool ok1 = q0 + row + 1 < Sq;
     const __hip_bfloat16* src = is_q_half ? q : dout;
     const long rs_ = is_q_half ? q_rs : do_rs;
+    const bool ok0 = q0 + row < Sq;
+    const bool ok1 = q0 + row + 1 < Sq;
     const long base = ((long)b * Sq + q0 + row) * rs_ + (long)hq_ * D + d0;
     sreg[0] = ok0 ? *reinterpret_cast<const uint4*>(src + base) : uint4{0, 0, 0, 0};
     sreg[1] = ok1 ? *reinterpret_cast<const uint4*>(src + base + rs_) : uint4{0, 0, 0, 0};
   };
-  auto stage_write = [&](int bufsel) {
+  auto stage_write = [&](int bufsel, int s) {
     __hip_bfloat16* q_lds = smem + bufsel * TILE;
-    __hip_bfloat16* do_lds = q_lds + 32 * D;
+    __hip_bfloat16* do_lds = q_lds + 64 * D;
     const int u = is_q_half ? tid : tid - CH_TOT;
     if (u >= CH_TOT) return;
     const int row = (u / (D / 8)) * 2;
     const int d0 = (u % (D / 8)) * 8;
-    __hip_bfloat16* dst = is_q_half ? q_lds : do_lds;
+    __hip_bfloat16* dst = (is_q_half ? q_lds : do_lds) + s * 32 * D;
     *reinterpret_cast<uint4*>(dst + rm_swz<D>(row, d0)) = sreg[0];
     *reinterpret_cast<uint4*>(dst + rm_swz<D>(row + 1, d0)) = sreg[1];
   };
@@ -400,124 +408,156 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
       hq_ = hkv * group + it / ntiles;
       q0 = q_lo + (it % ntiles) * 32;
     };
-    auto load_stats = [&](int bufsel, int hq_, int q0) {
+    auto load_stats = [&](int bufsel, int s, int hq_, int q0) {
       if (tid < 32) {
         const int qr = q0 + tid;
-        stats_lds[bufsel][0][tid] =
+        stats_lds[bufsel][0][s * 32 + tid] =
             ((qr < Sq) ? lse[((long)b * Hq + hq_) * Sq + qr] : INFINITY) * LOG2E;
-        stats_lds[bufsel][1][tid] =
+        stats_lds[bufsel][1][s * 32 + tid] =
             (qr < Sq) ? drow[((long)b * Sq + qr) * Hq + hq_] : 0.f;
       }
     };
-    int hq_, q0;
-    iter_to(0, hq_, q0);
-    stage_load(hq_, q0);
-    stage_write(0);
-    load_stats(0, hq_, q0);
+    int hq0_, q00;
+    iter_to(0, hq0_, q00);
+    stage_load(hq0_, q00);
+    stage_write(0, 0);
+    load_stats(0, 0, hq0_, q00);
+    if (NP == 2 && total_iters > 1) {
+      int hq1_, q01;
+      iter_to(1, hq1_, q01);
+      stage_load(hq1_, q01);
+      stage_write(0, 1);
+      load_stats(0, 1, hq1_, q01);
+    }
     __syncthreads();
 
     int buf = 0;
-    for (int it = 0; it < total_iters; ++it) {
-      iter_to(it, hq_, q0);
-      const float slope2 = (MOD == MOD_ALIBI) ? slopes[hq_] * LOG2E : 0.f;
-      const bool has_next = it + 1 < total_iters;
-      int nhq, nq0;
+    for (int it0 = 0; it0 < total_iters; it0 += NP) {
+      const int npair = min(NP, total_iters - it0);
+      const bool has_next = it0 + NP < total_iters;
+      const int nleft = total_iters - it0 - NP;  // sub-tiles in the next pair
+
+      // split-half staging: next pair's half h loads during THIS pair's
+      // half h compute and is written to buf^1 right after it (buf^1 was
+      // last read before the previous barrier, so the write is race-free)
       if (has_next) {
-        iter_to(it + 1, nhq, nq0);
+        int nhq, nq0;
+        iter_to(it0 + NP, nhq, nq0);
         stage_load(nhq, nq0);
       }
 
-      const __hip_bfloat16* q_lds = smem + buf * TILE;
-      const __hip_bfloat16* do_lds = q_lds + 32 * D;
-
-      // S = mfma(Q, K^T): A=Q rm frags from LDS, B = register kf.
-      // element (r=q_local, c=k_local)
-      f32x16 s_acc = {}, dp_acc = {};
-      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int dblk = 0; dblk < DBLK; ++dblk) {
-        const int off = rm_swz<D>(lk, dblk * 16 + hi * 8);
-        Bf16x8U qa;
-        *reinterpret_cast<uint4*>(qa.s) = *reinterpret_cast<const uint4*>(q_lds + off);
-        s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa.v, kf[dblk], s_acc, 0, 0, 0);
-        if constexpr (WANT_DK) {
-          Bf16x8U da;
-          *reinterpret_cast<uint4*>(da.s) = *reinterpret_cast<const uint4*>(do_lds + off);
-          dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da.v, vf[dblk], dp_acc, 0, 0, 0);
-        }
-      }
-      __builtin_amdgcn_s_setprio(0);
+      for (int s = 0; s < NP; ++s) {
+        if (s >= npair) break;  // block-uniform
+        int hq_, q0;
+        iter_to(it0 + s, hq_, q0);
+        const float slope2 = (MOD == MOD_ALIBI) ? slopes[hq_] * LOG2E : 0.f;
+        const __hip_bfloat16* q_lds = smem + buf * TILE + s * 32 * D;
+        const __hip_bfloat16* do_lds = smem + buf * TILE + 64 * D + s * 32 * D;
 
-      // interior (q tile entirely below this wave's k rows): mask-free
-      bool full = q0 + 31 < Sq;
-      if constexpr (MOD == MOD_CAUSAL) {
-        full = full && (q0 + q_off >= kv0w + 31);
-      } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
-        full = full && (q0 + q_off >= kv0w + 31) &&
-               ((q0 + 31 + q_off) - kv0w < modarg);
-      } else if constexpr (MOD == MOD_PREFIX_LM) {
-        // second clause wave-uniform: whole wave's k rows inside the prefix
-        full = full && ((q0 + q_off >= kv0w + 31) || (kv0w + 31 < modarg));
-      } else {
-        full = false;  // MOD_NONE boundary Sq checks + ALIBI slope
-      }
-
-      float pv[16];
-      if (full) {
+        // S = mfma(Q, K^T): A=Q rm frags from LDS, B = register kf.
+        // element (r=q_local, c=k_local)
+        f32x16 s_acc = {}, dp_acc = {};
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          const int qr = acc_row(reg, hi);
-          const float p = __builtin_amdgcn_exp2f(s_acc[reg] * scale2 - stats_lds[buf][0][qr]);
+        for (int dblk = 0; dblk < DBLK; ++dblk) {
+          const int off = rm_swz<D>(lk, dblk * 16 + hi * 8);
+          Bf16x8U qa;
+          *reinterpret_cast<uint4*>(qa.s) = *reinterpret_cast<const uint4*>(q_lds + off);
+          s_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qa.v, kf[dblk], s_acc, 0, 0, 0);
           if constexpr (WANT_DK) {
-            pv[reg] = p * (dp_acc[reg] - stats_lds[buf][1][qr]) * scale;  // dS
-          } else {
-            pv[reg] = p;
+            Bf16x8U da;
+            *reinterpret_cast<uint4*>(da.s) = *reinterpret_cast<const uint4*>(do_lds + off);
+            dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da.v, vf[dblk], dp_acc, 0, 0, 0);
           }
         }
-      } else {
+        __builtin_amdgcn_s_setprio(0);
+
+        // interior (q tile entirely below this wave's k rows): mask-free
+        bool full = q0 + 31 < Sq;
+        if constexpr (MOD == MOD_CAUSAL) {
+          full = full && (q0 + q_off >= kv0w + 31);
+        } else if constexpr (MOD == MOD_SLIDING_WINDOW) {
+          full = full && (q0 + q_off >= kv0w + 31) &&
+                 ((q0 + 31 + q_off) - kv0w < modarg);
+        } else if constexpr (MOD == MOD_PREFIX_LM) {
+          // second clause wave-uniform: whole wave's k rows inside the prefix
+          full = full && ((q0 + q_off >= kv0w + 31) || (kv0w + 31 < modarg));
+        } else {
+          full = false;  // MOD_NONE boundary Sq checks + ALIBI slope
+        }
+
+        float pv[16];
+        if (full) {
 #pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          const int qr = acc_row(reg, hi);
-          const int q_r = q0 + qr;
-          const int q_pos = q_r + q_off;
-          const int k_pos = krow;
-          const bool keep = k_valid && q_r < Sq && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
-          float s2 = s_acc[reg] * scale2;
-          if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
-          const float p = keep ? __builtin_amdgcn_exp2f(s2 - stats_lds[buf][0][qr]) : 0.f;
-          if constexpr (WANT_DK) {
-            pv[reg] = p * (dp_acc[reg] - stats_lds[buf][1][qr]) * scale;  // dS
-          } else {
-            pv[reg] = p;
+          for (int reg = 0; reg < 16; ++reg) {
+            const int qr = s * 32 + acc_row(reg, hi);
+            const float p = __builtin_amdgcn_exp2f(s_acc[reg] * scale2 - stats_lds[buf][0][qr]);
+            if constexpr (WANT_DK) {
+              pv[reg] = p * (dp_acc[reg] - stats_lds[buf][1][qr]) * scale;  // dS
+            } else {
+              pv[reg] = p;
+            }
+          }
+        } else {
+#pragma unroll
+          for (int reg = 0; reg < 16; ++reg) {
+            const int qr = acc_row(reg, hi);
+            const int q_r = q0 + qr;
+            const int q_pos = q_r + q_off;
+            const int k_pos = krow;
+            const bool keep = k_valid && q_r < Sq && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+            float s2 = s_acc[reg] * scale2;
+            if constexpr (MOD == MOD_ALIBI) s2 += slope2 * (k_pos - q_pos);
+            const float p =
+                keep ? __builtin_amdgcn_exp2f(s2 - stats_lds[buf][0][s * 32 + qr]) : 0.f;
+            if constexpr (WANT_DK) {
+              pv[reg] = p * (dp_acc[reg] - stats_lds[buf][1][s * 32 + qr]) * scale;  // dS
+            } else {
+              pv[reg] = p;
+            }
           }
         }
-      }
 
-      // transform: acc holds M[r=q][c=k]; A-frags of M^T = dS^T (dK) / P^T (dV)
-      bf16x8 a0, a1;
-      acc_to_afrag(pv, a0, a1);
-      __builtin_amdgcn_s_setprio(1);
+        // transform: acc holds M[r=q][c=k]; A-frags of M^T = dS^T (dK) / P^T (dV)
+        bf16x8 a0, a1;
+        acc_to_afrag(pv, a0, a1);
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int dc = 0; dc < DCOL; ++dc) {
-        f32x16 acc;
+        for (int dc = 0; dc < DCOL; ++dc) {
+          f32x16 acc;
 #pragma unroll
-        for (int r = 0; r < 16; ++r) acc[r] = acc_out[dc][r];
+          for (int r = 0; r < 16; ++r) acc[r] = acc_out[dc][r];
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks) {
-          // B = X[16q x 32d] (X = Q for dK, dO for dV): tr_read from rm image
-          bf16x8 xb = tr16_frag<D>(WANT_DK ? q_lds : do_lds, ks * 16 + hi * 8,
-                                   dc * 32, lane);
-          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb, acc, 0, 0, 0);
+          for (int ks = 0; ks < 2; ++ks) {
+            // B = X[16q x 32d] (X = Q for dK, dO for dV): tr_read from rm image
+            bf16x8 xb = tr16_frag<D>(WANT_DK ? q_lds : do_lds, ks * 16 + hi * 8,
+                                     dc * 32, lane);
+            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ks == 0 ? a0 : a1, xb, acc, 0, 0, 0);
+          }
+#pragma unroll
+          for (int r = 0; r < 16; ++r) acc_out[dc][r] = acc[r];
         }
-#pragma unroll
-        for (int r = 0; r < 16; ++r) acc_out[dc][r] = acc[r];
-      }
-      __builtin_amdgcn_s_setprio(0);
+        __builtin_amdgcn_s_setprio(0);
 
-      // write tile t+1 LAST: the whole iteration hides the global-load flight
-      if (has_next) {
-        stage_write(buf ^ 1);
-        load_stats(buf ^ 1, nhq, nq0);
+        // split-half staging epilogue: write the half that loaded during
+        // this half's compute, then start the next half's load
+        if (has_next) {
+          int nhq, nq0;
+          if (s == 0) {
+            iter_to(it0 + NP, nhq, nq0);
+            stage_write(buf ^ 1, 0);
+            load_stats(buf ^ 1, 0, nhq, nq0);
+            if (NP == 2 && nleft > 1) {
+              iter_to(it0 + NP + 1, nhq, nq0);
+              stage_load(nhq, nq0);
+            }
+          } else if (nleft > 1) {
+            iter_to(it0 + NP + 1, nhq, nq0);
+            stage_write(buf ^ 1, 1);
+            load_stats(buf ^ 1, 1, nhq, nq0);
+          }
+        }
       }
 
       __syncthreads();
