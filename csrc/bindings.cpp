@@ -4,6 +4,8 @@
 // rmsnorm.hip
 std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor rstd, at::Tensor dy);
+std::vector<at::Tensor> rmsnorm_fwd_res(at::Tensor x, at::Tensor res, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_bwd_add(at::Tensor x, at::Tensor w, at::Tensor rstd, at::Tensor dy, at::Tensor dadd);
 // rope.hip
 at::Tensor rope_fwd(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
                     long offset, bool conj);
@@ -68,6 +70,8 @@ at::Tensor tr16_frag_test(at::Tensor X);
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (y, rstd)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (dx, dw)");
+  m.def("rmsnorm_fwd_res", &rmsnorm_fwd_res, "fused residual-add + RMSNorm fwd (y, rstd, s)");
+  m.def("rmsnorm_bwd_add", &rmsnorm_bwd_add, "RMSNorm bwd with fused grad add (dx, dw)");
   m.def("rope_fwd", &rope_fwd, "RoPE apply (conj=true for backward)");
   m.def("rope_fwd_out", &rope_fwd_out, "RoPE apply into a strided out view");
   m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");

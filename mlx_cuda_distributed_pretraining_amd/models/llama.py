@@ -27,7 +27,7 @@ import torch.nn as nn
 
 from ..ops.attention import attention_ref, flash_attention, rope_flash_attention_qkv
 from ..ops.gemv import FastLinear, linear_fast
-from ..ops.rmsnorm import RMSNorm
+from ..ops.rmsnorm import RMSNorm, add_rms_norm
 from ..ops.rope import RopeTable, apply_rope
 from ..ops.swiglu import swiglu
 
@@ -241,6 +241,32 @@ class TransformerBlock(nn.Module):
         x = x + self.mlp(self.mlp_norm(x))
         return x
 
+    def _inner_pair(self, res: torch.Tensor, delta: Optional[torch.Tensor],
+                    cache: Optional[KVCache]):
+        """Residual-fused path: the incoming stream is (res, delta) with
+        x = res + delta; every residual add happens INSIDE the following
+        RMSNorm kernel (ops.rmsnorm.add_rms_norm -> csrc/rmsnorm.hip
+        HAS_RES), so no standalone elementwise-add kernels run. Returns the
+        next (res, delta) pair; the final add folds into the model's last
+        norm."""
+        if delta is None:
+            x = res
+            y1 = self.attention_norm(x)
+        else:
+            x, y1 = add_rms_norm(res, delta, self.attention_norm.weight,
+                                 self.attention_norm.eps)
+        a = self.attention(y1, cache)
+        h, y2 = add_rms_norm(x, a, self.mlp_norm.weight, self.mlp_norm.eps)
+        return h, self.mlp(y2)
+
+    def forward_pair(self, res: torch.Tensor, delta: Optional[torch.Tensor],
+                     cache: Optional[KVCache] = None):
+        if self._checkpoint and self.training and cache is None:
+            return torch.utils.checkpoint.checkpoint(
+                self._inner_pair, res, delta, cache, use_reentrant=False
+            )
+        return self._inner_pair(res, delta, cache)
+
     def _static_decode(self, x: torch.Tensor, cache) -> torch.Tensor:
         """hipGraph decode layer: fused GEMV chain (csrc/gemv.hip gemv_ex) —
         norm1 folds into the QKV GEMV staging, the first residual add into
@@ -304,6 +330,32 @@ class Model(nn.Module):
         self, tokens: torch.Tensor, cache: Optional[List[KVCache]] = None
     ) -> torch.Tensor:
         x = self.tok_embeddings(tokens)
+        static = (
+            cache is not None
+            and getattr(cache[0], "static_decode", False)
+            and x.is_cuda
+            and x.dtype == torch.bfloat16
+            and x.shape[1] == 1
+            and x.shape[0] <= 8
+        )
+        if not static:
+            # residual-fused layer chain: adds live inside the RMSNorm kernels
+            res, delta = x, None
+            for i, layer in enumerate(self.layers):
+                res, delta = layer.forward_pair(
+                    res, delta, cache[i] if cache is not None else None
+                )
+            if delta is None:
+                x = self.norm(res)
+            else:
+                _, x = add_rms_norm(res, delta, self.norm.weight, self.norm.eps)
+            if self.args.tie_word_embeddings:
+                logits = linear_fast(x, self.tok_embeddings.weight)
+            else:
+                logits = self.output(x)
+            if self.args.logit_scale:
+                logits = logits * self.args.logit_scale
+            return logits
         for i, layer in enumerate(self.layers):
             x = layer(x, cache[i] if cache is not None else None)
         if (

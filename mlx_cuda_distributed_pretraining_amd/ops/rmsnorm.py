@@ -74,3 +74,57 @@ class RMSNorm(torch.nn.Module):
 
     def extra_repr(self) -> str:
         return f"{self.weight.shape[0]}, eps={self.eps}"
+
+
+class _AddRMSNormFn(torch.autograd.Function):
+    """Fused residual: s = x + res; y = rmsnorm(s) * w in ONE kernel each way
+    (csrc/rmsnorm.hip HAS_RES / HAS_DADD) — the separate residual-add kernels
+    and their HBM round trips disappear. Returns (s, y); s is the residual
+    stream the next sublayer adds onto."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, res: torch.Tensor, weight: torch.Tensor, eps: float):
+        if use_hip(x, res, weight):
+            ext = get_ext()
+            y, rstd, s = ext.rmsnorm_fwd_res(
+                x.contiguous(), res.contiguous(), weight.contiguous(), eps
+            )
+            ctx.save_for_backward(s, weight, rstd)
+            ctx.hip = True
+            return s, y
+        s = x + res
+        sf = s.float()
+        rstd = torch.rsqrt(sf.pow(2).mean(-1) + eps)
+        y = (sf * rstd.unsqueeze(-1) * weight.float()).to(x.dtype)
+        ctx.save_for_backward(s, weight, rstd)
+        ctx.hip = False
+        return s, y
+
+    @staticmethod
+    def backward(ctx, ds: torch.Tensor, dy: torch.Tensor):
+        s, weight, rstd = ctx.saved_tensors
+        if ctx.hip:
+            ext = get_ext()
+            dadd = ds.contiguous() if ds is not None else s.new_empty(0)
+            dx, dw = ext.rmsnorm_bwd_add(
+                s.contiguous(), weight.contiguous(), rstd, dy.contiguous(), dadd
+            )
+            # s = x + res: both inputs get the same gradient tensor
+            return dx, dx, dw.to(weight.dtype), None
+        sf = s.float()
+        dyf = dy.float()
+        wf = weight.float()
+        r = rstd.unsqueeze(-1)
+        xhat = sf * r
+        dw = (dyf * xhat).reshape(-1, s.shape[-1]).sum(0)
+        dyw = dyf * wf
+        dx = (r * (dyw - xhat * (dyw * xhat).mean(-1, keepdim=True))).to(s.dtype)
+        if ds is not None:
+            dx = dx + ds
+        return dx, dx, dw.to(weight.dtype), None
+
+
+def add_rms_norm(x: torch.Tensor, res: torch.Tensor, weight: torch.Tensor,
+                 eps: float = 1e-5):
+    """(s, y) with s = x + res and y = rms_norm(s) * weight, fused."""
+    return _AddRMSNormFn.apply(x, res, weight, eps)

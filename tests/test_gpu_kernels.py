@@ -475,3 +475,28 @@ def test_int8_kv_decode(ext):
     agree = sum(a == b for a, b in zip(ref_out, q8_out)) / len(ref_out)
     assert agree >= 0.7, (agree, ref_out, q8_out)
     assert all(0 <= t < 503 for t in q8_out)
+
+
+def test_add_rmsnorm_gpu(ext):
+    from mlx_cuda_distributed_pretraining_amd.ops.rmsnorm import add_rms_norm
+
+    torch.manual_seed(0)
+    for rows, H in ((64, 2048), (33, 1024)):
+        x = torch.randn(rows, H, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+        r = torch.randn(rows, H, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+        w = torch.randn(H, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+        s, y = add_rms_norm(x, r, w, 1e-5)
+        (s.float().sin().sum() + y.float().sum()).backward()
+
+        x2 = x.detach().float().requires_grad_(True)
+        r2 = r.detach().float().requires_grad_(True)
+        w2 = w.detach().float().requires_grad_(True)
+        s2 = x2 + r2
+        y2 = s2 * torch.rsqrt(s2.pow(2).mean(-1, keepdim=True) + 1e-5) * w2
+        (s2.sin().sum() + y2.sum()).backward()
+
+        assert (s.float() - s2).abs().max() < 2e-2
+        assert (y.float() - y2).abs().max() < 8e-2
+        assert (x.grad.float() - x2.grad).abs().max() < 8e-2
+        assert (r.grad.float() - r2.grad).abs().max() < 8e-2
+        assert (w.grad.float() - w2.grad).abs().max() / w2.grad.abs().max() < 3e-2
