@@ -4,7 +4,7 @@
 // rmsnorm.hip
 std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
 std::vector<at::Tensor> rmsnorm_bwd(at::Tensor x, at::Tensor w, at::Tensor rstd, at::Tensor dy);
-std::vector<at::Tensor> rmsnorm_fwd_res(at::Tensor x, at::Tensor res, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_fwd_res(at::Tensor x, at::Tensor res, at::Tensor w, double eps, bool with_amax);
 std::vector<at::Tensor> rmsnorm_bwd_add(at::Tensor x, at::Tensor w, at::Tensor rstd, at::Tensor dy, at::Tensor dadd);
 // rope.hip
 at::Tensor rope_fwd(at::Tensor x, at::Tensor cost, at::Tensor sint, bool traditional,
@@ -13,6 +13,7 @@ at::Tensor rope_fwd_out(at::Tensor x, at::Tensor cost, at::Tensor sint, bool tra
                         long offset, bool conj, at::Tensor y);
 // swiglu.hip
 at::Tensor swiglu_fwd(at::Tensor gu);
+std::vector<at::Tensor> swiglu_fwd_amax(at::Tensor gu, bool with_amax);
 at::Tensor swiglu_bwd(at::Tensor gu, at::Tensor dy);
 // cross_entropy.hip
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets, long ignore_index);
@@ -62,6 +63,7 @@ at::Tensor gemv_ex(at::Tensor x, at::Tensor W, long mode, at::Tensor nw, double 
                    at::Tensor res);
 // fp8_quant.hip
 std::vector<at::Tensor> fp8_quantize(at::Tensor x, bool transpose);
+std::vector<at::Tensor> fp8_quantize_pre(at::Tensor x, at::Tensor amax_in, bool transpose);
 // debug.hip
 at::Tensor mfma_tile_test(at::Tensor A, at::Tensor B);
 at::Tensor afrag_transform_test(at::Tensor M);
@@ -70,11 +72,12 @@ at::Tensor tr16_frag_test(at::Tensor X);
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (y, rstd)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (dx, dw)");
-  m.def("rmsnorm_fwd_res", &rmsnorm_fwd_res, "fused residual-add + RMSNorm fwd (y, rstd, s)");
+  m.def("rmsnorm_fwd_res", &rmsnorm_fwd_res, "fused residual-add + RMSNorm fwd (y, rstd[, s][, amax])");
   m.def("rmsnorm_bwd_add", &rmsnorm_bwd_add, "RMSNorm bwd with fused grad add (dx, dw)");
   m.def("rope_fwd", &rope_fwd, "RoPE apply (conj=true for backward)");
   m.def("rope_fwd_out", &rope_fwd_out, "RoPE apply into a strided out view");
   m.def("swiglu_fwd", &swiglu_fwd, "SwiGLU forward");
+  m.def("swiglu_fwd_amax", &swiglu_fwd_amax, "swiglu fwd emitting |out| amax bits");
   m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (loss_sum, ntok, lse)");
   m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward (dlogits)");
@@ -100,6 +103,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv_ex", &gemv_ex,
         "fused decode GEMV (rmsnorm/swiglu staging, residual epilogue)");
   m.def("fp8_quantize", &fp8_quantize, "fused bf16 -> e4m3 quantize (codes, scale)");
+  m.def("fp8_quantize_pre", &fp8_quantize_pre, "e4m3 quantize with producer-supplied amax bits");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");
   m.def("tr16_frag_test", &tr16_frag_test, "debug: ds_read_b64_tr_b16 B-fragment gather");

@@ -46,6 +46,16 @@ __global__ void scale_from_amax_kernel(const unsigned* __restrict__ amax_bits,
   *scale = fmaxf(c.f / E4M3_MAX, 1e-12f);
 }
 
+// reduce a producer's per-block partial-max array -> scale (one tiny block)
+__global__ void scale_from_partials_kernel(const float* __restrict__ partials, long n,
+                                           float* __restrict__ scale) {
+  __shared__ float scratch[256 / WAVE];
+  float m = 0.f;
+  for (long i = threadIdx.x; i < n; i += 256) m = fmaxf(m, partials[i]);
+  m = block_reduce_max<256>(m, scratch);
+  if (threadIdx.x == 0) *scale = fmaxf(m / E4M3_MAX, 1e-12f);
+}
+
 // cast with the INVERSE scale read from device memory. 8 elems/thread.
 __global__ void cast_e4m3_kernel(const __hip_bfloat16* __restrict__ x,
                                  unsigned char* __restrict__ y, long n,
@@ -101,20 +111,30 @@ __global__ void cast_e4m3_t_kernel(const __hip_bfloat16* __restrict__ x,
 
 }  // namespace
 
-std::vector<at::Tensor> fp8_quantize(at::Tensor x, bool transpose) {
+// amax_in numel()==0 -> run the amax pass; otherwise amax_in is a PRODUCER
+// kernel's per-block partial-max float array (rmsnorm / swiglu emit one) and
+// the extra full read of x is skipped — the tiny reduce replaces it.
+std::vector<at::Tensor> fp8_quantize_pre(at::Tensor x, at::Tensor amax_in, bool transpose) {
   // x: bf16 [.., K] (2D for transpose). Returns (codes float8_e4m3fn, scale f32[1]).
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
   auto xc = x.contiguous();
   const long n = xc.numel();
   auto stream = at::cuda::getCurrentHIPStream();
-  auto amax_bits = at::zeros({1}, x.options().dtype(at::kInt));
+  const bool have_amax = amax_in.numel() > 0;
   auto scale = at::empty({}, x.options().dtype(at::kFloat));
   const long grid = std::min<long>(cdiv(n, 256 * 8), 2048);
   auto* xp = reinterpret_cast<const __hip_bfloat16*>(xc.data_ptr());
-  amax_kernel<<<grid, 256, 0, stream>>>(
-      xp, n, reinterpret_cast<unsigned*>(amax_bits.data_ptr<int>()));
-  scale_from_amax_kernel<<<1, 1, 0, stream>>>(
-      reinterpret_cast<unsigned*>(amax_bits.data_ptr<int>()), scale.data_ptr<float>());
+  if (have_amax) {
+    TORCH_CHECK(amax_in.scalar_type() == at::kFloat);
+    scale_from_partials_kernel<<<1, 256, 0, stream>>>(
+        amax_in.data_ptr<float>(), amax_in.numel(), scale.data_ptr<float>());
+  } else {
+    auto amax_bits = at::zeros({1}, x.options().dtype(at::kInt));
+    amax_kernel<<<grid, 256, 0, stream>>>(
+        xp, n, reinterpret_cast<unsigned*>(amax_bits.data_ptr<int>()));
+    scale_from_amax_kernel<<<1, 1, 0, stream>>>(
+        reinterpret_cast<unsigned*>(amax_bits.data_ptr<int>()), scale.data_ptr<float>());
+  }
   at::Tensor y;
   if (!transpose) {
     y = at::empty_like(xc, xc.options().dtype(at::kFloat8_e4m3fn));
@@ -129,4 +149,8 @@ std::vector<at::Tensor> fp8_quantize(at::Tensor x, bool transpose) {
         xp, reinterpret_cast<unsigned char*>(y.data_ptr()), N, K, scale.data_ptr<float>());
   }
   return {y, scale};
+}
+
+std::vector<at::Tensor> fp8_quantize(at::Tensor x, bool transpose) {
+  return fp8_quantize_pre(x, at::empty({0}, x.options().dtype(at::kFloat)), transpose);
 }

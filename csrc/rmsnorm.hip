@@ -12,11 +12,14 @@ namespace {
 // HAS_RES: s = x + res is computed in pass 1, written out (it IS the next
 // residual stream), and normalized — the separate residual-add kernel and
 // its extra HBM round trip disappear (SURVEY.md §2.6 fusion rule).
+// amax_partial (nullable): per-BLOCK max|y| written to amax_partial[blockIdx]
+// (plain store — an atomic per wave to one address serializes at 32k blocks);
+// fp8_quantize_pre reduces the partial array (skips its own amax pass).
 template <typename T, int BLOCK, bool HAS_RES>
 __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ res,
                                    T* __restrict__ sum_out, const T* __restrict__ w,
                                    T* __restrict__ y, float* __restrict__ rstd,
-                                   int H, float eps) {
+                                   int H, float eps, float* __restrict__ amax_partial) {
   __shared__ float scratch[BLOCK / WAVE];
   const long row = blockIdx.x;
   const T* xr = x + row * (long)H;
@@ -62,6 +65,7 @@ __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict_
   if (threadIdx.x == 0) rstd[row] = r;
 
   const T* src = HAS_RES ? sr : xr;
+  float am = 0.f;
   if constexpr (sizeof(T) == 2) {
     const int HV = H / 8;
     const uint4* xv = reinterpret_cast<const uint4*>(src);
@@ -70,15 +74,29 @@ __global__ void rmsnorm_fwd_kernel(const T* __restrict__ x, const T* __restrict_
     for (int i = threadIdx.x; i < HV; i += BLOCK) {
       U4 u, ww, o; u.u = xv[i]; ww.u = wv[i];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        o.s[j] = f32_to_bf16_bits(bf16_bits_to_f32(u.s[j]) * r * bf16_bits_to_f32(ww.s[j]));
+      for (int j = 0; j < 8; ++j) {
+        const float yf = bf16_bits_to_f32(u.s[j]) * r * bf16_bits_to_f32(ww.s[j]);
+        o.s[j] = f32_to_bf16_bits(yf);
+        am = fmaxf(am, fabsf(yf));
+      }
       yv[i] = o.u;
     }
-    for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK)
-      from_f32(&yr[i], to_f32(src[i]) * r * to_f32(w[i]));
+    for (int i = HV * 8 + threadIdx.x; i < H; i += BLOCK) {
+      const float yf = to_f32(src[i]) * r * to_f32(w[i]);
+      from_f32(&yr[i], yf);
+      am = fmaxf(am, fabsf(yf));
+    }
   } else {
-    for (int i = threadIdx.x; i < H; i += BLOCK)
-      from_f32(&yr[i], to_f32(src[i]) * r * to_f32(w[i]));
+    for (int i = threadIdx.x; i < H; i += BLOCK) {
+      const float yf = to_f32(src[i]) * r * to_f32(w[i]);
+      from_f32(&yr[i], yf);
+      am = fmaxf(am, fabsf(yf));
+    }
+  }
+  if (amax_partial) {
+    __syncthreads();  // scratch was read by every thread in pass 1's reduce
+    am = block_reduce_max<BLOCK>(am, scratch);
+    if (threadIdx.x == 0) amax_partial[row] = am;
   }
 }
 
@@ -169,7 +187,10 @@ __global__ void rmsnorm_bwd_kernel(const T* __restrict__ x, const T* __restrict_
 
 // res numel()==0 -> plain norm, returns {y, rstd}.
 // res present -> fused s = x + res: returns {y, rstd, s}.
-std::vector<at::Tensor> rmsnorm_fwd_res(at::Tensor x, at::Tensor res, at::Tensor w, double eps) {
+// with_amax: also return an int[1] tensor of monotonic float bits holding
+// max|y| (consumed by fp8_quantize_pre). Returns {y, rstd[, s][, amax]}.
+std::vector<at::Tensor> rmsnorm_fwd_res(at::Tensor x, at::Tensor res, at::Tensor w, double eps,
+                                        bool with_amax) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous());
   const bool has_res = res.numel() > 0;
   if (has_res) TORCH_CHECK(res.is_contiguous() && res.sizes() == x.sizes());
@@ -178,6 +199,9 @@ std::vector<at::Tensor> rmsnorm_fwd_res(at::Tensor x, at::Tensor res, at::Tensor
   auto y = at::empty_like(x);
   auto s = has_res ? at::empty_like(x) : at::empty({0}, x.options());
   auto rstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto amax = with_amax ? at::empty({rows}, x.options().dtype(at::kFloat))
+                        : at::empty({0}, x.options().dtype(at::kFloat));
+  float* ab = with_amax ? amax.data_ptr<float>() : nullptr;
   auto stream = at::cuda::getCurrentHIPStream();
   constexpr int BLOCK = 256;
   dim3 grid(rows);
@@ -188,22 +212,24 @@ std::vector<at::Tensor> rmsnorm_fwd_res(at::Tensor x, at::Tensor res, at::Tensor
         rmsnorm_fwd_kernel<T, BLOCK, true><<<grid, BLOCK, 0, stream>>>(
             reinterpret_cast<const T*>(x.data_ptr()), reinterpret_cast<const T*>(res.data_ptr()),
             reinterpret_cast<T*>(s.data_ptr()), reinterpret_cast<const T*>(w.data_ptr()),
-            reinterpret_cast<T*>(y.data_ptr()), rstd.data_ptr<float>(), H, (float)eps);
+            reinterpret_cast<T*>(y.data_ptr()), rstd.data_ptr<float>(), H, (float)eps, ab);
       else
         rmsnorm_fwd_kernel<T, BLOCK, false><<<grid, BLOCK, 0, stream>>>(
             reinterpret_cast<const T*>(x.data_ptr()), nullptr, nullptr,
             reinterpret_cast<const T*>(w.data_ptr()),
-            reinterpret_cast<T*>(y.data_ptr()), rstd.data_ptr<float>(), H, (float)eps);
+            reinterpret_cast<T*>(y.data_ptr()), rstd.data_ptr<float>(), H, (float)eps, ab);
     } else {
       TORCH_CHECK(false, "rmsnorm: unsupported dtype");
     }
   });
-  if (has_res) return {y, rstd, s};
-  return {y, rstd};
+  std::vector<at::Tensor> out = {y, rstd};
+  if (has_res) out.push_back(s);
+  if (with_amax) out.push_back(amax);
+  return out;
 }
 
 std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
-  return rmsnorm_fwd_res(x, at::empty({0}, x.options()), w, eps);
+  return rmsnorm_fwd_res(x, at::empty({0}, x.options()), w, eps, false);
 }
 
 // dadd numel()==0 -> plain; otherwise dx += dadd (fused residual grad add).

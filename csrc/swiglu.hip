@@ -6,9 +6,13 @@
 
 namespace {
 
+// amax_partial (nullable): per-block max|out| (plain store, reduced later by
+// fp8_quantize_pre — no hot atomics; ops/fp8.py).
 template <typename T>
 __global__ void swiglu_fwd_kernel(const T* __restrict__ gu, T* __restrict__ out,
-                                  long rows, int I) {
+                                  long rows, int I, float* __restrict__ amax_partial) {
+  __shared__ float scratch[256 / WAVE];
+  float am = 0.f;
   const int IV = I / 8;
   for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < rows * (long)IV;
        idx += gridDim.x * (long)blockDim.x) {
@@ -25,16 +29,24 @@ __global__ void swiglu_fwd_kernel(const T* __restrict__ gu, T* __restrict__ out,
       for (int j = 0; j < 8; ++j) {
         const float gf = bf16_bits_to_f32(gv.s[j]);
         const float uf = bf16_bits_to_f32(uv.s[j]);
-        ov.s[j] = f32_to_bf16_bits(gf / (1.f + __expf(-gf)) * uf);
+        const float of = gf / (1.f + __expf(-gf)) * uf;
+        ov.s[j] = f32_to_bf16_bits(of);
+        am = fmaxf(am, fabsf(of));
       }
       *reinterpret_cast<uint4*>(o) = ov.u;
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const float gf = to_f32(g[j]);
-        from_f32(&o[j], gf / (1.f + __expf(-gf)) * to_f32(u[j]));
+        const float of = gf / (1.f + __expf(-gf)) * to_f32(u[j]);
+        from_f32(&o[j], of);
+        am = fmaxf(am, fabsf(of));
       }
     }
+  }
+  if (amax_partial) {
+    am = block_reduce_max<256>(am, scratch);
+    if (threadIdx.x == 0) amax_partial[blockIdx.x] = am;
   }
 }
 
@@ -84,7 +96,7 @@ __global__ void swiglu_bwd_kernel(const T* __restrict__ gu, const T* __restrict_
 
 }  // namespace
 
-at::Tensor swiglu_fwd(at::Tensor gu) {
+std::vector<at::Tensor> swiglu_fwd_amax(at::Tensor gu, bool with_amax) {
   TORCH_CHECK(gu.is_cuda() && gu.is_contiguous());
   const int I2 = gu.size(-1);
   TORCH_CHECK(I2 % 16 == 0, "swiglu: intermediate size must be a multiple of 8");
@@ -97,16 +109,23 @@ at::Tensor swiglu_fwd(at::Tensor gu) {
   const long total = rows * (I / 8);
   const int block = 256;
   const long grid = std::min<long>(cdiv(total, block), 2048);
+  auto amax = with_amax ? at::empty({grid}, gu.options().dtype(at::kFloat))
+                        : at::empty({0}, gu.options().dtype(at::kFloat));
   AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, gu.scalar_type(), "swiglu_fwd", [&] {
     using T = std::conditional_t<std::is_same_v<scalar_t, at::BFloat16>, __hip_bfloat16, float>;
     if constexpr (std::is_same_v<scalar_t, at::BFloat16> || std::is_same_v<scalar_t, float>) {
       swiglu_fwd_kernel<T><<<grid, block, 0, stream>>>(
-          reinterpret_cast<const T*>(gu.data_ptr()), reinterpret_cast<T*>(out.data_ptr()), rows, I);
+          reinterpret_cast<const T*>(gu.data_ptr()), reinterpret_cast<T*>(out.data_ptr()), rows, I,
+          with_amax ? amax.data_ptr<float>() : nullptr);
     } else {
       TORCH_CHECK(false, "swiglu: unsupported dtype");
     }
   });
-  return out;
+  return {out, amax};
+}
+
+at::Tensor swiglu_fwd(at::Tensor gu) {
+  return swiglu_fwd_amax(gu, false)[0];
 }
 
 at::Tensor swiglu_bwd(at::Tensor gu, at::Tensor dy) {

@@ -219,7 +219,8 @@ class MLP(nn.Module):
         self.w_gate_up.fp8 = self.w_down.fp8 = args.fp8
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.w_down(swiglu(self.w_gate_up(x)))
+        # emit_amax: w_down's fp8 quantizer reuses the swiglu-accumulated amax
+        return self.w_down(swiglu(self.w_gate_up(x), emit_amax=self.w_down.fp8))
 
 
 class TransformerBlock(nn.Module):
@@ -229,6 +230,9 @@ class TransformerBlock(nn.Module):
         self.attention = Attention(args, rope_table)
         self.mlp_norm = RMSNorm(args.hidden_size, args.rms_norm_eps)
         self.mlp = MLP(args)
+        # fp8 consumers (wqkv / w_gate_up) reuse the norms' amax accumulators
+        self.attention_norm.emit_amax = args.fp8
+        self.mlp_norm.emit_amax = args.fp8
         self._checkpoint = False
 
     def enable_checkpointing(self) -> None:
@@ -254,9 +258,11 @@ class TransformerBlock(nn.Module):
             y1 = self.attention_norm(x)
         else:
             x, y1 = add_rms_norm(res, delta, self.attention_norm.weight,
-                                 self.attention_norm.eps)
+                                 self.attention_norm.eps,
+                                 self.attention_norm.emit_amax)
         a = self.attention(y1, cache)
-        h, y2 = add_rms_norm(x, a, self.mlp_norm.weight, self.mlp_norm.eps)
+        h, y2 = add_rms_norm(x, a, self.mlp_norm.weight, self.mlp_norm.eps,
+                             self.mlp_norm.emit_amax)
         return h, self.mlp(y2)
 
     def forward_pair(self, res: torch.Tensor, delta: Optional[torch.Tensor],

@@ -42,23 +42,27 @@ def fp8_available() -> bool:
     return hasattr(torch, "_scaled_mm") and torch.cuda.is_available()
 
 
-def _quant(t: torch.Tensor, transpose: bool = False):
-    # fused HIP quantize (csrc/fp8_quant.hip) on GPU; torch fallback on CPU
+def _quant(t: torch.Tensor, transpose: bool = False, amax: torch.Tensor = None):
+    # fused HIP quantize (csrc/fp8_quant.hip) on GPU; torch fallback on CPU.
+    # amax: producer-accumulated |t|max float bits (rmsnorm/swiglu emit them)
+    # -> the quantizer's own amax read of t is skipped.
     from ._ext import get_ext
 
     ext = get_ext()
     if ext is not None and t.is_cuda:
+        if amax is not None:
+            return ext.fp8_quantize_pre(t, amax, transpose)
         return ext.fp8_quantize(t, transpose)
     return quantize_e4m3(t.t().contiguous() if transpose else t)
 
 
 class _Fp8LinearFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, weight: torch.Tensor):
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, x_amax: torch.Tensor = None):
         # x: [..., K] bf16; weight: [N, K] bf16
         shape = x.shape
         x2 = x.reshape(-1, shape[-1])
-        x8, sx = _quant(x2)
+        x8, sx = _quant(x2, amax=x_amax)
         w8, sw = _quant(weight)
         y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
                              out_dtype=x.dtype)
@@ -78,11 +82,13 @@ class _Fp8LinearFn(torch.autograd.Function):
                               out_dtype=dy.dtype)
         # wgrad in bf16 (outlier-sensitive)
         dw = dy2.t() @ x2
-        return dx.reshape(*dy.shape[:-1], x2.shape[-1]), dw
+        return dx.reshape(*dy.shape[:-1], x2.shape[-1]), dw, None
 
 
 def fp8_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
-    return _Fp8LinearFn.apply(x, weight)
+    # views drop python attributes, so the producer's amax rides in from the
+    # ORIGINAL activation tensor here
+    return _Fp8LinearFn.apply(x, weight, getattr(x, "_mcdp_amax", None))
 
 
 def fp8_linear_ref(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:

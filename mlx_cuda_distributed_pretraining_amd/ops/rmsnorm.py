@@ -21,10 +21,17 @@ def rms_norm_ref(x: torch.Tensor, weight: torch.Tensor, eps: float) -> torch.Ten
 
 class _RMSNormFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, eps: float):
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor, eps: float,
+                emit_amax: bool = False):
         if use_hip(x, weight):
             ext = get_ext()
-            y, rstd = ext.rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps)
+            if emit_amax:
+                y, rstd, amax = ext.rmsnorm_fwd_res(
+                    x.contiguous(), x.new_empty(0), weight.contiguous(), eps, True
+                )
+                y._mcdp_amax = amax  # fp8 quantizer skips its amax pass
+            else:
+                y, rstd = ext.rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps)
             ctx.save_for_backward(x, weight, rstd)
             ctx.eps = eps
             ctx.hip = True
@@ -46,7 +53,7 @@ class _RMSNormFn(torch.autograd.Function):
             dx, dw = ext.rmsnorm_bwd(
                 x.contiguous(), weight.contiguous(), rstd, dy.contiguous()
             )
-            return dx, dw.to(weight.dtype), None
+            return dx, dw.to(weight.dtype), None, None
         xf = x.float()
         dyf = dy.float()
         wf = weight.float()
@@ -56,11 +63,12 @@ class _RMSNormFn(torch.autograd.Function):
         # dx = r * (dy*w - xhat * mean(dy*w*xhat))
         dyw = dyf * wf
         dx = r * (dyw - xhat * (dyw * xhat).mean(-1, keepdim=True))
-        return dx.to(x.dtype), dw.to(weight.dtype), None
+        return dx.to(x.dtype), dw.to(weight.dtype), None, None
 
 
-def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
-    return _RMSNormFn.apply(x, weight, eps)
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5,
+             emit_amax: bool = False) -> torch.Tensor:
+    return _RMSNormFn.apply(x, weight, eps, emit_amax)
 
 
 class RMSNorm(torch.nn.Module):
@@ -68,9 +76,11 @@ class RMSNorm(torch.nn.Module):
         super().__init__()
         self.weight = torch.nn.Parameter(torch.ones(hidden_size, dtype=dtype))
         self.eps = eps
+        # set by the model when the fp8 path consumes this norm's output
+        self.emit_amax = False
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return rms_norm(x, self.weight, self.eps)
+        return rms_norm(x, self.weight, self.eps, self.emit_amax)
 
     def extra_repr(self) -> str:
         return f"{self.weight.shape[0]}, eps={self.eps}"
@@ -83,12 +93,16 @@ class _AddRMSNormFn(torch.autograd.Function):
     stream the next sublayer adds onto."""
 
     @staticmethod
-    def forward(ctx, x: torch.Tensor, res: torch.Tensor, weight: torch.Tensor, eps: float):
+    def forward(ctx, x: torch.Tensor, res: torch.Tensor, weight: torch.Tensor, eps: float,
+                emit_amax: bool = False):
         if use_hip(x, res, weight):
             ext = get_ext()
-            y, rstd, s = ext.rmsnorm_fwd_res(
-                x.contiguous(), res.contiguous(), weight.contiguous(), eps
+            out = ext.rmsnorm_fwd_res(
+                x.contiguous(), res.contiguous(), weight.contiguous(), eps, emit_amax
             )
+            y, rstd, s = out[0], out[1], out[2]
+            if emit_amax:
+                y._mcdp_amax = out[3]
             ctx.save_for_backward(s, weight, rstd)
             ctx.hip = True
             return s, y
@@ -110,7 +124,7 @@ class _AddRMSNormFn(torch.autograd.Function):
                 s.contiguous(), weight.contiguous(), rstd, dy.contiguous(), dadd
             )
             # s = x + res: both inputs get the same gradient tensor
-            return dx, dx, dw.to(weight.dtype), None
+            return dx, dx, dw.to(weight.dtype), None, None
         sf = s.float()
         dyf = dy.float()
         wf = weight.float()
@@ -121,10 +135,10 @@ class _AddRMSNormFn(torch.autograd.Function):
         dx = (r * (dyw - xhat * (dyw * xhat).mean(-1, keepdim=True))).to(s.dtype)
         if ds is not None:
             dx = dx + ds
-        return dx, dx, dw.to(weight.dtype), None
+        return dx, dx, dw.to(weight.dtype), None, None
 
 
 def add_rms_norm(x: torch.Tensor, res: torch.Tensor, weight: torch.Tensor,
-                 eps: float = 1e-5):
+                 eps: float = 1e-5, emit_amax: bool = False):
     """(s, y) with s = x + res and y = rms_norm(s) * weight, fused."""
-    return _AddRMSNormFn.apply(x, res, weight, eps)
+    return _AddRMSNormFn.apply(x, res, weight, eps, emit_amax)

@@ -25,9 +25,13 @@ def swiglu_ref(gate_up: torch.Tensor) -> torch.Tensor:
 
 class _SwiGLUFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, gate_up: torch.Tensor):
+    def forward(ctx, gate_up: torch.Tensor, emit_amax: bool = False):
         ctx.save_for_backward(gate_up)
         if use_hip(gate_up):
+            if emit_amax:
+                y, amax = get_ext().swiglu_fwd_amax(gate_up.contiguous(), True)
+                y._mcdp_amax = amax  # fp8 quantizer skips its amax pass
+                return y
             return get_ext().swiglu_fwd(gate_up.contiguous())
         return swiglu_ref(gate_up)
 
@@ -35,7 +39,7 @@ class _SwiGLUFn(torch.autograd.Function):
     def backward(ctx, dy: torch.Tensor):
         (gate_up,) = ctx.saved_tensors
         if use_hip(gate_up):
-            return get_ext().swiglu_bwd(gate_up.contiguous(), dy.contiguous())
+            return get_ext().swiglu_bwd(gate_up.contiguous(), dy.contiguous()), None
         i = gate_up.shape[-1] // 2
         g = gate_up[..., :i].float()
         u = gate_up[..., i:].float()
@@ -44,9 +48,10 @@ class _SwiGLUFn(torch.autograd.Function):
         silu_g = g * sg
         dgate = dyf * u * (sg * (1 + g * (1 - sg)))
         dup = dyf * silu_g
-        return torch.cat([dgate, dup], dim=-1).to(gate_up.dtype)
+        return torch.cat([dgate, dup], dim=-1).to(gate_up.dtype), None
 
 
-def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
-    """gate_up: [..., 2I] -> [..., I] = silu(gate_up[...,:I]) * gate_up[...,I:]"""
-    return _SwiGLUFn.apply(gate_up)
+def swiglu(gate_up: torch.Tensor, emit_amax: bool = False) -> torch.Tensor:
+    """gate_up: [..., 2I] -> [..., I] = silu(gate_up[...,:I]) * gate_up[...,I:].
+    emit_amax: attach ``_mcdp_amax`` (max|out| float bits) for the fp8 path."""
+    return _SwiGLUFn.apply(gate_up, emit_amax)
