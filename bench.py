@@ -117,7 +117,8 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": value / BASELINE_TOKS_PER_SEC,
-            "dtype": "bf16" if on_gpu else "fp32",
+            "dtype": (("bf16+fp8-gemm" if (cfg.model.misc or {}).get("fp8") else "bf16")
+                      if on_gpu else "fp32"),
             "data": "synthetic",
             "config": {
                 "model": model_name,
