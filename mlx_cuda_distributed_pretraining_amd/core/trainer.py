@@ -153,7 +153,13 @@ class Trainer:
         if self.config.model.architecture == "llama_standard":
             args.attention_type = "simple"
         self.model_args = args
-        self.model = Model(args).to(device=self.device, dtype=self.param_dtype)
+        # Build directly ON the training device: CPU weight init is
+        # single-threaded torch.normal_ and takes minutes at 7B scale, while
+        # the same init on the GPU is sub-second (288 GB HBM holds the whole
+        # model anyway).
+        with torch.device(self.device):
+            self.model = Model(args)
+        self.model = self.model.to(device=self.device, dtype=self.param_dtype)
         if self.config.system.gradient_checkpointing:
             ratio = self.config.system.gradient_checkpointing_ratio
             n = int(len(self.model.layers) * ratio)
