@@ -157,8 +157,11 @@ class Trainer:
         # single-threaded torch.normal_ and takes minutes at 7B scale, while
         # the same init on the GPU is sub-second (288 GB HBM holds the whole
         # model anyway).
-        with torch.device(self.device):
-            self.model = Model(args)
+        try:
+            with torch.device(self.device):
+                self.model = Model(args)
+        except Exception:
+            self.model = Model(args)  # CPU init fallback (slow but safe)
         self.model = self.model.to(device=self.device, dtype=self.param_dtype)
         if self.config.system.gradient_checkpointing:
             ratio = self.config.system.gradient_checkpointing_ratio
