@@ -170,6 +170,24 @@ def update_metadata(run_dir: Path, entry: Dict[str, Any]) -> None:
     meta_path.write_text(json.dumps(meta, indent=2))
 
 
+def latest_checkpoint(run_dir: str | Path) -> str | None:
+    """Base path (``.../checkpoints/step_<N>``) of the run's newest step
+    checkpoint, or None. Used by ``--auto-resume`` (elastic restarts):
+    numeric steps are preferred over 'final' (a finished run has nothing to
+    resume)."""
+    ckdir = Path(run_dir) / "checkpoints"
+    if not ckdir.is_dir():
+        return None
+    best = -1
+    for pth in ckdir.glob("step_*_state.json"):
+        m = re.match(r"step_(\d+)_state\.json", pth.name)
+        if m:
+            best = max(best, int(m.group(1)))
+    if best < 0:
+        return None
+    return str(ckdir / f"step_{best}")
+
+
 def rotate_snapshots(checkpoint_dir: Path, max_snapshots: int) -> None:
     """Keep only the newest ``max_snapshots`` step checkpoints (never 'final').
 

@@ -31,6 +31,11 @@ def build_arg_parser() -> argparse.ArgumentParser:
     p.add_argument("--tensorboard", action="store_true")
     p.add_argument("--overwrite", action="store_true")
     p.add_argument("--resume", type=str, default=None, help="checkpoint base path to resume from")
+    p.add_argument("--auto-resume", action="store_true",
+                   help="resume from this run's newest step checkpoint if one exists "
+                        "(closes the elastic-restart loop: torchrun --max-restarts N "
+                        "-m ...training --config X --auto-resume)")
+    p.add_argument("--runs-root", type=str, default="runs")
     return p
 
 
@@ -62,6 +67,18 @@ def apply_overrides(config: Config, args: argparse.Namespace) -> Config:
         from .config import ResumeConfig
 
         config.resume = ResumeConfig(checkpoint=args.resume)
+    elif getattr(args, "auto_resume", False):
+        from pathlib import Path
+
+        from .checkpoint import latest_checkpoint
+        from .config import ResumeConfig
+
+        latest = latest_checkpoint(Path(getattr(args, "runs_root", "runs")) / config.name)
+        if latest:
+            config.resume = ResumeConfig(checkpoint=latest)
+        else:
+            # crashed (or first start) before any checkpoint: start clean
+            config.overwrite = True
     return config
 
 
@@ -69,7 +86,7 @@ def main(argv=None) -> None:
     args = build_arg_parser().parse_args(argv)
     config = Config.from_yaml(args.config)
     config = apply_overrides(config, args)
-    trainer = Trainer(config)
+    trainer = Trainer(config, runs_root=args.runs_root)
     trainer.train()
 
 

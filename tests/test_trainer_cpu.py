@@ -158,3 +158,42 @@ def test_emergency_checkpoint_on_failure(tmp_runs):
         trainer.train()
     ckpts = list((Path(tmp_runs) / "crash-run" / "checkpoints").glob("step_emergency_*_model.safetensors"))
     assert ckpts, "no emergency checkpoint written"
+
+
+def test_auto_resume_end_to_end(tmp_path):
+    """--auto-resume closes the elastic-restart loop: run 4 iters with a
+    checkpoint at 2, then re-invoke with --auto-resume and more iters — the
+    second run must resume from step 2's checkpoint (not restart), and a
+    fresh name with --auto-resume must start clean."""
+    import yaml
+    from mlx_cuda_distributed_pretraining_amd.core import training as T
+    from mlx_cuda_distributed_pretraining_amd.core.checkpoint import latest_checkpoint
+
+    cfg = {
+        "name": "autoresume-test",
+        "data": {"synthetic": True, "synthetic_vocab_size": 64,
+                 "preprocessing": {"max_context_size": 32}},
+        "model": {"dimensions": {"hidden_size": 32, "intermediate_size": 64, "num_layers": 1},
+                  "attention": {"num_heads": 2, "num_kv_heads": 2,
+                                "max_position_embeddings": 64}},
+        "training": {"hyperparameters": {"iters": 4, "batch_size": 2, "learning_rate": 1e-3}},
+        "logging": {"steps": {"logging_interval": 0, "checkpoint_interval": 2,
+                              "validation_interval": 0}},
+        "system": {"device": "cpu"},
+    }
+    cfg_path = tmp_path / "cfg.yaml"
+    cfg_path.write_text(yaml.safe_dump(cfg))
+    runs = tmp_path / "runs"
+
+    # first start: no run dir -> auto-resume falls through to a clean start
+    T.main(["--config", str(cfg_path), "--auto-resume", "--runs-root", str(runs)])
+    latest = latest_checkpoint(runs / "autoresume-test")
+    assert latest is not None and latest.endswith("step_4")
+
+    # restart with more iters: must RESUME (run dir exists -> without resume
+    # the unique-name check would raise)
+    T.main(["--config", str(cfg_path), "--auto-resume", "--runs-root", str(runs),
+            "--iters", "6"])
+    assert latest_checkpoint(runs / "autoresume-test").endswith("step_6")
+    text = (runs / "autoresume-test" / "log.txt").read_text()
+    assert "Resumed from" in text
