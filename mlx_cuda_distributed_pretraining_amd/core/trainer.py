@@ -253,6 +253,10 @@ class Trainer:
             targets.reshape(-1),
             ignore_index=self.tokenizer.PAD_TOKEN,
         )
+        # MoE: add the router load-balance loss (models/llama.py MoE)
+        aux = getattr(self.model, "aux_loss", None)
+        if aux is not None and self.model.training:
+            loss = loss + self.model_args.router_aux_loss_coef * aux
         return loss, ntok
 
     def train_step(self, step: int) -> tuple:

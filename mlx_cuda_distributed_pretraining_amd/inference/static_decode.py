@@ -165,6 +165,12 @@ class GraphDecoder:
     def __init__(self, model, batch: int = 1, max_len: int = 2048,
                  temperature: float = 0.0, min_p: float = 0.0, seed: int = 0,
                  kv_bits: int = None):
+        if getattr(model.args, "num_local_experts", 0):
+            raise NotImplementedError(
+                "GraphDecoder: MoE models are not supported by the fused "
+                "static-decode path yet (token-dependent routing breaks the "
+                "fixed captured graph) — use eager generation"
+            )
         self.model = model
         self.batch = batch
         self.max_len = max_len

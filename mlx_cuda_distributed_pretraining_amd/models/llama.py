@@ -55,6 +55,7 @@ class ModelArgs:
     attention_prefix_len: Optional[int] = None  # prefix-LM split (flex)
     use_alibi: bool = False
     fp8: bool = False  # opt-in e4m3 GEMMs for the block projections (ops/fp8.py)
+    router_aux_loss_coef: float = 0.01  # MoE load-balance loss weight
     # MoE knobs exist in the reference config (models/llama.py:40-41) but no
     # MoE layer is implemented there; kept for config parity.
     num_local_experts: int = 0
@@ -91,6 +92,7 @@ class ModelArgs:
             tie_word_embeddings=bool(misc.get("tie_word_embeddings", True)),
             logit_scale=misc.get("logit_scale"),
             fp8=bool(misc.get("fp8", False)),
+            router_aux_loss_coef=float(misc.get("router_aux_loss_coef", 0.01)),
             attention_type=str(attn.get("type", "flash")),
             attention_window=attn.get("window"),
             attention_prefix_len=attn.get("prefix_len"),
@@ -223,13 +225,68 @@ class MLP(nn.Module):
         return self.w_down(swiglu(self.w_gate_up(x), emit_amax=self.w_down.fp8))
 
 
+class MoE(nn.Module):
+    """Mixture-of-experts FFN — BEYOND reference parity: the reference carries
+    the ``num_local_experts`` / ``num_experts_per_tok`` config knobs but
+    implements no MoE layer (/root/reference/models/llama.py:40-41,
+    SURVEY.md §2.2).
+
+    Design: per-expert fused gate+up / down weights held as single stacked
+    3-D parameters (one flat region each in the fused optimizer), softmax
+    top-k routing with renormalized gates, Switch-style load-balancing aux
+    loss exposed on ``self.aux_loss`` (the trainer adds
+    ``model.aux_loss * router_aux_loss_coef``). The expert loop computes a
+    gathered token batch per expert (index_select -> GEMM -> index_add);
+    single-GPU 288 GB holds large expert counts without EP — expert-parallel
+    all-to-all is the documented round-2 step (ROADMAP.md)."""
+
+    def __init__(self, args: ModelArgs):
+        super().__init__()
+        E, H, I = args.num_local_experts, args.hidden_size, args.intermediate_size
+        self.num_experts = E
+        self.top_k = max(1, int(args.num_experts_per_tok or 1))
+        self.router = FastLinear(H, E, bias=False)
+        self.w_gate_up = nn.Parameter(torch.empty(E, 2 * I, H))
+        self.w_down = nn.Parameter(torch.empty(E, H, I))
+        std = 0.02
+        nn.init.normal_(self.w_gate_up, mean=0.0, std=std)
+        nn.init.normal_(self.w_down, mean=0.0, std=std)
+        self.aux_loss: Optional[torch.Tensor] = None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        shape = x.shape
+        xf = x.reshape(-1, shape[-1])
+        n = xf.shape[0]
+        probs = torch.softmax(self.router(xf).float(), dim=-1)  # [N, E]
+        gates, idx = probs.topk(self.top_k, dim=-1)             # [N, k]
+        gates = (gates / gates.sum(-1, keepdim=True)).to(x.dtype)
+        # Switch aux loss: E * sum_e fraction_routed_e * mean_prob_e  (==1 at
+        # perfect balance); computed every forward, consumed by the trainer.
+        with torch.no_grad():
+            counts = torch.zeros(self.num_experts, device=x.device, dtype=torch.float32)
+            counts.scatter_add_(0, idx.reshape(-1),
+                                torch.ones(idx.numel(), device=x.device))
+            frac = counts / (n * self.top_k)
+        self.aux_loss = self.num_experts * (frac * probs.mean(0)).sum()
+        out = torch.zeros_like(xf)
+        for e in range(self.num_experts):
+            rows, slot = (idx == e).nonzero(as_tuple=True)
+            if rows.numel() == 0:
+                continue
+            toks = xf.index_select(0, rows)
+            h = swiglu(toks @ self.w_gate_up[e].t())
+            y = (h @ self.w_down[e].t()) * gates[rows, slot].unsqueeze(-1)
+            out.index_add_(0, rows, y.to(out.dtype))
+        return out.reshape(shape)
+
+
 class TransformerBlock(nn.Module):
     def __init__(self, args: ModelArgs, rope_table: RopeTable):
         super().__init__()
         self.attention_norm = RMSNorm(args.hidden_size, args.rms_norm_eps)
         self.attention = Attention(args, rope_table)
         self.mlp_norm = RMSNorm(args.hidden_size, args.rms_norm_eps)
-        self.mlp = MLP(args)
+        self.mlp = MoE(args) if args.num_local_experts > 0 else MLP(args)
         # fp8 consumers (wqkv / w_gate_up) reuse the norms' amax accumulators
         self.attention_norm.emit_amax = args.fp8
         self.mlp_norm.emit_amax = args.fp8
@@ -351,6 +408,10 @@ class Model(nn.Module):
                 res, delta = layer.forward_pair(
                     res, delta, cache[i] if cache is not None else None
                 )
+            # MoE: surface the summed load-balance loss for the trainer
+            aux = [layer.mlp.aux_loss for layer in self.layers
+                   if isinstance(layer.mlp, MoE) and layer.mlp.aux_loss is not None]
+            self.aux_loss = torch.stack(aux).sum() if aux else None
             if delta is None:
                 x = self.norm(res)
             else:

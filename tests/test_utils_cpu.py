@@ -174,7 +174,16 @@ def test_trainer_streams_to_stats_mesh(tmp_path):
 
     t = threading.Thread(target=run_server, daemon=True)
     t.start()
-    time.sleep(0.5)
+    # wait for the server to actually accept connections (a fixed sleep is
+    # flaky under full-suite load: the client connects once, silently)
+    import socket
+
+    for _ in range(100):
+        try:
+            with socket.create_connection(("127.0.0.1", 18766), timeout=0.2):
+                break
+        except OSError:
+            time.sleep(0.1)
 
     repo = Path(__file__).resolve().parent.parent
     cfg = Config.from_yaml(repo / "configs" / "model-config-sample.yaml")

@@ -116,6 +116,10 @@ def convert_run(run_dir: str | Path, out_path: str | Path, checkpoint: str = "fi
         tk = config.data.tokenizer
         vocab_size = int(tk.get("normal_vocab_size", 256)) + len(tk.get("special_tokens", {}))
     args = ModelArgs.from_config(config.model, vocab_size)
+    if args.num_local_experts:
+        raise NotImplementedError(
+            "convert_to_mlx_lm: MoE runs have no HF-Llama equivalent layout"
+        )
 
     model_path, _opt, _state = CheckpointManager.get_checkpoint_paths(
         str(run_dir / "checkpoints" / f"step_{checkpoint}")
