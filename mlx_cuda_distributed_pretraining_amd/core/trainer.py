@@ -171,6 +171,24 @@ class Trainer:
         if self.config.data.weight_path:
             self.model.load_weights(self.config.data.weight_path)
         broadcast_module(self.model)
+        # Tensor parallelism (parallel/tp.py): shard the broadcast model.
+        # The reference's model_parallel flags were a logged placeholder
+        # (/root/reference/core/training.py:1178-1193); here they are real.
+        self.tp_world = 1
+        if self.config.system.model_parallel and self.world_size > 1:
+            mp_size = int(self.config.system.model_parallel_size or self.world_size)
+            if mp_size != self.world_size:
+                raise ValueError(
+                    f"model_parallel_size={mp_size} must equal world size "
+                    f"{self.world_size} (no DPxTP mesh yet — ROADMAP)"
+                )
+            from ..parallel.tp import apply_tensor_parallel
+
+            apply_tensor_parallel(self.model, self.rank, self.world_size)
+            self.tp_world = self.world_size
+            if self.is_main:
+                self.logger.log(f"Tensor parallel over {self.tp_world} ranks "
+                                f"(head/intermediate sharded, 1 all-reduce per sublayer)")
         self.logger.log_model_summary(self.model)
 
     def setup_training(self) -> None:
@@ -180,9 +198,11 @@ class Trainer:
         self.grad_accum_steps = int(hp.get("gradient_accumulation_steps", 1))
         self.max_grad_norm = float(hp.get("gradient_clip", hp.get("max_grad_norm", 0.0)) or 0.0)
 
+        data_rank = 0 if self.tp_world > 1 else self.rank
+        data_world = 1 if self.tp_world > 1 else self.world_size
         self.data_manager = DataManager(
             cfg.data, self.tokenizer, self.batch_size,
-            rank=self.rank, world_size=self.world_size, seed=cfg.system.seed,
+            rank=data_rank, world_size=data_world, seed=cfg.system.seed,
         )
         if cfg.training.epochs is not None and self.data_manager.num_batches:
             self.steps_per_epoch = max(
@@ -222,8 +242,14 @@ class Trainer:
             )
         else:
             self.flat_space = FlatParamSpace(self.model)
-            self.ddp = DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb)
+            # TP ranks hold DIFFERENT shards: a data-parallel grad all-reduce
+            # would corrupt them — each rank's grads are already exact.
+            self.ddp = (None if self.tp_world > 1 else
+                        DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb))
             self.optimizer = self.opt_manager.create_optimizer(self.model)
+        if self.tp_world > 1 and int(cfg.logging.steps.get("checkpoint_interval", 0)):
+            raise ValueError("TP runs: sharded checkpointing not implemented yet "
+                             "(ROADMAP) — set logging.steps.checkpoint_interval: 0")
 
         es = cfg.training.early_stopping or {}
         self.early_stopping = (
@@ -237,6 +263,31 @@ class Trainer:
         )
 
     # ------------------------------------------------------------------
+    def _tp_clip(self) -> None:
+        """Global-norm clip under TP: norm^2 = sum of sharded-param grads over
+        ALL ranks + replicated params counted ONCE; every rank applies the
+        same scale so replicas stay bit-identical."""
+        import torch.distributed as dist
+
+        sh = torch.zeros((), dtype=torch.float32, device=self.device)
+        rep = torch.zeros((), dtype=torch.float32, device=self.device)
+        for p in self.model.parameters():
+            if p.grad is None:
+                continue
+            ss = p.grad.float().pow(2).sum()
+            if getattr(p, "_tp_sharded", False):
+                sh += ss
+            else:
+                rep += ss
+        if is_distributed():
+            dist.all_reduce(sh)
+        norm = (sh + rep).sqrt()
+        scale = self.max_grad_norm / (norm + 1e-6)
+        if float(scale) < 1.0:
+            for p in self.model.parameters():
+                if p.grad is not None:
+                    p.grad.mul_(scale)
+
     def compute_loss(self, inputs: torch.Tensor, targets: torch.Tensor):
         mpe = self.model_args.max_position_embeddings
         if mpe and inputs.shape[1] > mpe:
@@ -303,7 +354,9 @@ class Trainer:
         if isinstance(self.optimizer, FusedFlatAdamW):
             self.optimizer.step(lr=lr)
         else:
-            if self.max_grad_norm > 0 and not getattr(self.optimizer, "max_grad_norm", 0):
+            if self.max_grad_norm > 0 and self.tp_world > 1:
+                self._tp_clip()
+            elif self.max_grad_norm > 0 and not getattr(self.optimizer, "max_grad_norm", 0):
                 clip_by_global_norm(
                     [p for p in self.model.parameters() if p.requires_grad],
                     self.max_grad_norm,

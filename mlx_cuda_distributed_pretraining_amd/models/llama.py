@@ -28,6 +28,7 @@ import torch.nn as nn
 from ..ops.attention import attention_ref, flash_attention, rope_flash_attention_qkv
 from ..ops.gemv import FastLinear, linear_fast
 from ..ops.rmsnorm import RMSNorm, add_rms_norm
+from ..parallel.tp import copy_to_tp, reduce_from_tp
 from ..ops.rope import RopeTable, apply_rope
 from ..ops.swiglu import swiglu
 
@@ -149,6 +150,8 @@ class Attention(nn.Module):
 
     def forward(self, x: torch.Tensor, cache: Optional[KVCache] = None) -> torch.Tensor:
         B, S, _ = x.shape
+        if getattr(self, "_tp", False):
+            x = copy_to_tp(x)  # f: replicated activation enters column-parallel qkv
         qkv = self.wqkv(x)
 
         if getattr(cache, "static_decode", False):
@@ -157,7 +160,7 @@ class Attention(nn.Module):
             # decode attention, all shape-static
             assert S == 1, "static decode processes one token at a time"
             o = cache.attend(self, qkv)
-            return self.wo(o.reshape(B, S, -1))
+            return self.wo(o.reshape(B, S, -1))  # static decode is TP-free
 
         atype = self.args.attention_type
         if cache is None and atype != "simple":
@@ -173,7 +176,8 @@ class Attention(nn.Module):
                 prefix_len=self.args.attention_prefix_len if atype == "flex" else None,
                 alibi_slopes=self.alibi_slopes if self.args.use_alibi else None,
             )
-            return self.wo(o.reshape(B, S, -1))
+            out = self.wo(o.reshape(B, S, -1))
+            return reduce_from_tp(out) if getattr(self, "_tp", False) else out
 
         q, k, v = qkv.split(
             [
@@ -208,7 +212,8 @@ class Attention(nn.Module):
                 prefix_len=self.args.attention_prefix_len if atype == "flex" else None,
                 alibi_slopes=self.alibi_slopes if self.args.use_alibi else None,
             )
-        return self.wo(o.reshape(B, S, -1))
+        out = self.wo(o.reshape(B, S, -1))
+        return reduce_from_tp(out) if getattr(self, "_tp", False) else out
 
 
 class MLP(nn.Module):
@@ -221,8 +226,11 @@ class MLP(nn.Module):
         self.w_gate_up.fp8 = self.w_down.fp8 = args.fp8
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if getattr(self, "_tp", False):
+            x = copy_to_tp(x)
         # emit_amax: w_down's fp8 quantizer reuses the swiglu-accumulated amax
-        return self.w_down(swiglu(self.w_gate_up(x), emit_amax=self.w_down.fp8))
+        out = self.w_down(swiglu(self.w_gate_up(x), emit_amax=self.w_down.fp8))
+        return reduce_from_tp(out) if getattr(self, "_tp", False) else out
 
 
 class MoE(nn.Module):
