@@ -171,6 +171,11 @@ class GraphDecoder:
                 "static-decode path yet (token-dependent routing breaks the "
                 "fixed captured graph) — use eager generation"
             )
+        if getattr(model, "_tp_world", 1) > 1:
+            raise NotImplementedError(
+                "GraphDecoder: TP-sharded models need all-reduces inside the "
+                "captured GEMV chain (ROADMAP) — decode on a full replica"
+            )
         self.model = model
         self.batch = batch
         self.max_len = max_len
