@@ -174,21 +174,26 @@ class Trainer:
         # Tensor parallelism (parallel/tp.py): shard the broadcast model.
         # The reference's model_parallel flags were a logged placeholder
         # (/root/reference/core/training.py:1178-1193); here they are real.
-        self.tp_world = 1
+        self.tp_world, self.dp_world = 1, self.world_size
+        self.tp_rank, self.dp_rank = 0, self.rank
+        self.tp_pg = self.dp_pg = None
         if self.config.system.model_parallel and self.world_size > 1:
             mp_size = int(self.config.system.model_parallel_size or self.world_size)
-            if mp_size != self.world_size:
-                raise ValueError(
-                    f"model_parallel_size={mp_size} must equal world size "
-                    f"{self.world_size} (no DPxTP mesh yet — ROADMAP)"
-                )
-            from ..parallel.tp import apply_tensor_parallel
+            from ..parallel.tp import apply_tensor_parallel, init_tp_mesh
 
-            apply_tensor_parallel(self.model, self.rank, self.world_size)
-            self.tp_world = self.world_size
+            # DPxTP mesh: TP groups are adjacent ranks (one xGMI node per
+            # replica), DP groups stride across replicas
+            self.tp_rank, self.dp_rank, self.tp_pg, self.dp_pg = init_tp_mesh(
+                self.rank, self.world_size, mp_size
+            )
+            apply_tensor_parallel(self.model, self.tp_rank, mp_size)
+            self.tp_world = mp_size
+            self.dp_world = self.world_size // mp_size
             if self.is_main:
-                self.logger.log(f"Tensor parallel over {self.tp_world} ranks "
-                                f"(head/intermediate sharded, 1 all-reduce per sublayer)")
+                self.logger.log(
+                    f"Mesh: TP={self.tp_world} x DP={self.dp_world} "
+                    f"(head/intermediate sharded, 1 all-reduce per sublayer)"
+                )
         self.logger.log_model_summary(self.model)
 
     def setup_training(self) -> None:
@@ -198,11 +203,10 @@ class Trainer:
         self.grad_accum_steps = int(hp.get("gradient_accumulation_steps", 1))
         self.max_grad_norm = float(hp.get("gradient_clip", hp.get("max_grad_norm", 0.0)) or 0.0)
 
-        data_rank = 0 if self.tp_world > 1 else self.rank
-        data_world = 1 if self.tp_world > 1 else self.world_size
+        # TP replicas share data; DP replicas shard it (mesh coordinates)
         self.data_manager = DataManager(
             cfg.data, self.tokenizer, self.batch_size,
-            rank=data_rank, world_size=data_world, seed=cfg.system.seed,
+            rank=self.dp_rank, world_size=self.dp_world, seed=cfg.system.seed,
         )
         if cfg.training.epochs is not None and self.data_manager.num_batches:
             self.steps_per_epoch = max(
@@ -242,10 +246,15 @@ class Trainer:
             )
         else:
             self.flat_space = FlatParamSpace(self.model)
-            # TP ranks hold DIFFERENT shards: a data-parallel grad all-reduce
-            # would corrupt them — each rank's grads are already exact.
-            self.ddp = (None if self.tp_world > 1 else
-                        DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb))
+            # grads all-reduce over the DP group only: TP peers hold
+            # DIFFERENT shards and their grads are already exact
+            if self.tp_world > 1:
+                self.ddp = (DataParallelGrads(self.flat_space,
+                                              bucket_mb=cfg.system.bucket_mb,
+                                              process_group=self.dp_pg)
+                            if self.dp_world > 1 else None)
+            else:
+                self.ddp = DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb)
             self.optimizer = self.opt_manager.create_optimizer(self.model)
         if self.tp_world > 1 and int(cfg.logging.steps.get("checkpoint_interval", 0)):
             raise ValueError("TP runs: sharded checkpointing not implemented yet "
@@ -280,7 +289,7 @@ class Trainer:
             else:
                 rep += ss
         if is_distributed():
-            dist.all_reduce(sh)
+            dist.all_reduce(sh, group=self.tp_pg)
         norm = (sh + rep).sqrt()
         scale = self.max_grad_norm / (norm + 1e-6)
         if float(scale) < 1.0:

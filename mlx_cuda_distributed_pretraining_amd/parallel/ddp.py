@@ -111,5 +111,7 @@ class DataParallelGrads:
         for b in self.buckets:
             if b.work is not None:
                 b.work.wait()
-        self.space.flat_grad.div_(get_world_size())
+        self.space.flat_grad.div_(
+            dist.get_world_size(self.pg) if self.pg is not None else get_world_size()
+        )
         self._reset_pending()

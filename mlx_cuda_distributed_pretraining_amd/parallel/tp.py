@@ -25,19 +25,59 @@ Mechanics
   Sharded parameters get ``p._tp_sharded = True`` so the gradient-norm
   reduction can count replicated params once (parallel/dist.py consumers).
 
-Scope (round 1): TP degree == world size (no DP×TP mesh yet), checkpointing
-of TP runs saves rank-local shards only when explicitly enabled, MoE + TP is
-rejected. CPU-tested with 2-rank gloo (tests/test_tp_cpu.py); the collective
-pattern is backend-agnostic.
+Scope (round 1): DPxTP meshes via ``init_tp_mesh`` (TP groups = adjacent
+ranks for xGMI locality; DP groups stride across replicas; gradient
+all-reduce rides the DP group only), TP checkpointing not yet implemented,
+MoE + TP rejected. CPU-tested with 2-rank TP and a 4-rank 2x2 mesh on gloo
+(tests/test_tp_cpu.py); the collective pattern is backend-agnostic.
 """
 from __future__ import annotations
 
 import torch
 import torch.distributed as dist
 
+# the TP process group (None = the default/world group). Set by
+# init_tp_mesh() when TP degree < world size (DPxTP mesh).
+_TP_GROUP = None
+
+
+def set_tp_group(group) -> None:
+    global _TP_GROUP
+    _TP_GROUP = group
+
 
 def _world() -> int:
-    return dist.get_world_size() if dist.is_initialized() else 1
+    if not dist.is_initialized():
+        return 1
+    return dist.get_world_size(_TP_GROUP) if _TP_GROUP is not None else dist.get_world_size()
+
+
+def tp_group():
+    return _TP_GROUP
+
+
+def init_tp_mesh(rank: int, world: int, tp: int):
+    """Build the DPxTP mesh: TP groups are ADJACENT ranks (xGMI locality:
+    the tp-degree peers of one model replica sit on one node's ring), DP
+    groups stride across replicas. Returns (tp_rank, dp_rank, tp_pg, dp_pg).
+    Every rank must call this (dist.new_group is collective)."""
+    if world % tp:
+        raise ValueError(f"world {world} not divisible by TP degree {tp}")
+    dp = world // tp
+    tp_pg = dp_pg = None
+    if tp > 1 and tp < world:
+        for d in range(dp):
+            ranks = list(range(d * tp, (d + 1) * tp))
+            g = dist.new_group(ranks)
+            if rank in ranks:
+                tp_pg = g
+        for t in range(tp):
+            ranks = list(range(t, world, tp))
+            g = dist.new_group(ranks)
+            if rank in ranks:
+                dp_pg = g
+        set_tp_group(tp_pg)
+    return rank % tp, rank // tp, tp_pg, dp_pg
 
 
 class _CopyToTP(torch.autograd.Function):
@@ -51,7 +91,7 @@ class _CopyToTP(torch.autograd.Function):
     def backward(ctx, grad: torch.Tensor) -> torch.Tensor:
         if _world() > 1:
             grad = grad.contiguous()
-            dist.all_reduce(grad)
+            dist.all_reduce(grad, group=_TP_GROUP)
         return grad
 
 
@@ -62,7 +102,7 @@ class _ReduceFromTP(torch.autograd.Function):
     def forward(ctx, x: torch.Tensor) -> torch.Tensor:
         if _world() > 1:
             x = x.contiguous()
-            dist.all_reduce(x)
+            dist.all_reduce(x, group=_TP_GROUP)
         return x
 
     @staticmethod

@@ -181,3 +181,92 @@ def test_tp_trainer_replicas_stay_identical(tmp_path):
     assert res[0]["losses"] == pytest.approx(res[1]["losses"], abs=1e-6)
     for n, w in res[0]["replicated"].items():
         assert (w == res[1]["replicated"][n]).all(), f"replica drift on {n}"
+
+
+def _mesh_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.parallel.ddp import DataParallelGrads
+        from mlx_cuda_distributed_pretraining_amd.parallel.dist import broadcast_module
+        from mlx_cuda_distributed_pretraining_amd.parallel.flat import FlatParamSpace
+        from mlx_cuda_distributed_pretraining_amd.parallel.tp import (
+            apply_tensor_parallel, init_tp_mesh)
+
+        tp = 2
+        tp_rank, dp_rank, tp_pg, dp_pg = init_tp_mesh(rank, world, tp)
+        torch.manual_seed(0)
+        model = Model(_args())
+        broadcast_module(model)
+        apply_tensor_parallel(model, tp_rank, tp)
+        space = FlatParamSpace(model)
+        ddp = DataParallelGrads(space, bucket_mb=1, process_group=dp_pg)
+
+        g = torch.Generator().manual_seed(3)
+        full = torch.randint(0, 67, (4, 16), generator=g)
+        batch = full[dp_rank * 2:(dp_rank + 1) * 2]  # TP pair shares its half
+        logits = model(batch[:, :-1])
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, 67), batch[:, 1:].reshape(-1))
+        loss.backward()
+        ddp.finalize()
+        q.put({"rank": rank, "tp_rank": tp_rank,
+               "grads": {n: p.grad.numpy().copy() for n, p in model.named_parameters()},
+               "sharded": {n: bool(getattr(p, "_tp_sharded", False))
+                           for n, p in model.named_parameters()}})
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_x_tp2_mesh_grads_match_single_process():
+    """2x2 mesh: after the DP-group all-reduce-mean, every rank's grads must
+    equal the single-process full-batch grads (replicated params) / the
+    matching shard slice (sharded params)."""
+    torch.manual_seed(0)
+    model = Model(_args())
+    g = torch.Generator().manual_seed(3)
+    full = torch.randint(0, 67, (4, 16), generator=g)
+    logits = model(full[:, :-1])
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, 67), full[:, 1:].reshape(-1))
+    loss.backward()
+    ref = {n: p.grad.clone() for n, p in model.named_parameters()}
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_mesh_worker, args=(r, 4, 29515, q)) for r in range(4)]
+    for p in procs:
+        p.start()
+    res = [q.get() for _ in range(4)]
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+
+    a = _args()
+    hd, lq, lkv = a.head_dim, a.num_heads // 2, a.num_kv_heads // 2
+    li = a.intermediate_size // 2
+    for r in res:
+        rk = r["tp_rank"]
+        for n, gr in r["grads"].items():
+            gt = torch.from_numpy(gr)
+            full_g = ref[n]
+            if not r["sharded"][n]:
+                want = full_g
+            elif "wqkv" in n:
+                rows = torch.cat([
+                    torch.arange(rk * lq * hd, (rk + 1) * lq * hd),
+                    a.num_heads * hd + torch.arange(rk * lkv * hd, (rk + 1) * lkv * hd),
+                    (a.num_heads + a.num_kv_heads) * hd
+                    + torch.arange(rk * lkv * hd, (rk + 1) * lkv * hd)])
+                want = full_g[rows]
+            elif "wo" in n:
+                want = full_g[:, rk * lq * hd:(rk + 1) * lq * hd]
+            elif "w_gate_up" in n:
+                rows = torch.cat([torch.arange(rk * li, (rk + 1) * li),
+                                  a.intermediate_size + torch.arange(rk * li, (rk + 1) * li)])
+                want = full_g[rows]
+            elif "w_down" in n:
+                want = full_g[:, rk * li:(rk + 1) * li]
+            assert torch.allclose(gt, want, atol=1e-5), f"rank {r['rank']} grad {n}"
