@@ -256,9 +256,7 @@ class Trainer:
             else:
                 self.ddp = DataParallelGrads(self.flat_space, bucket_mb=cfg.system.bucket_mb)
             self.optimizer = self.opt_manager.create_optimizer(self.model)
-        if self.tp_world > 1 and int(cfg.logging.steps.get("checkpoint_interval", 0)):
-            raise ValueError("TP runs: sharded checkpointing not implemented yet "
-                             "(ROADMAP) — set logging.steps.checkpoint_interval: 0")
+
 
         es = cfg.training.early_stopping or {}
         self.early_stopping = (
@@ -428,6 +426,9 @@ class Trainer:
 
     # ------------------------------------------------------------------
     def save_checkpoint(self, tag: str, val_loss: Optional[float] = None) -> None:
+        if self.tp_world > 1:
+            self._save_checkpoint_tp(tag, val_loss)
+            return
         if not self.is_main:
             barrier()
             return
@@ -452,10 +453,53 @@ class Trainer:
             rotate_snapshots(self.checkpoint_dir, self.config.logging.max_snapshots)
         barrier()
 
+    def _save_checkpoint_tp(self, tag: str, val_loss: Optional[float]) -> None:
+        """TP runs: one dp_rank==0 rank per TP shard saves its own
+        model/optimizer triple at ``step_<tag>_tp<r>``; the global main also
+        writes a plain ``step_<tag>_state.json`` marker (holds tp_world) so
+        ``latest_checkpoint`` / --auto-resume keep working. Snapshot rotation
+        skips the shard files (markers only) — full rotation is a ROADMAP
+        item."""
+        import json as _json
+
+        base = str(self.checkpoint_dir / f"step_{tag}")
+        training_state = {
+            "step": self.current_step,
+            "total_tokens": int(self.total_tokens),
+            "validation_losses": self.validation_losses,
+            "val_ptr": getattr(self.data_manager, "val_ptr", 0),
+            "tp_world": self.tp_world,
+        }
+        if self.dp_rank == 0:
+            save_checkpoint(f"{base}_tp{self.tp_rank}", self.model,
+                            self.optimizer.state_dict(), training_state)
+        if self.is_main:
+            Path(f"{base}_state.json").write_text(_json.dumps(training_state))
+            update_metadata(
+                self.run_dir,
+                {"tag": str(tag), "step": self.current_step, "val_loss": val_loss,
+                 "path": base, "tp_world": self.tp_world, "time": time.time()},
+            )
+        barrier()
+
     def load_checkpoint(self, checkpoint_base: str, reset_optimizer: bool = False,
                         reset_training_state: bool = False) -> None:
+        if self.tp_world > 1:
+            sharded = f"{checkpoint_base}_tp{self.tp_rank}"
+            if not Path(f"{sharded}_state.json").exists():
+                raise FileNotFoundError(
+                    f"TP run (degree {self.tp_world}) needs per-shard checkpoint "
+                    f"files ({sharded}_*); re-sharding a different layout is not "
+                    f"implemented (ROADMAP)"
+                )
+            checkpoint_base = sharded
         opt_state, training_state = load_checkpoint(checkpoint_base, self.model,
                                                     map_location=str(self.device))
+        if self.tp_world > 1 and training_state.get("tp_world") not in (None, self.tp_world):
+            raise ValueError(
+                f"checkpoint was saved at tp_world={training_state.get('tp_world')}, "
+                f"this run is tp_world={self.tp_world}"
+            )
         if self.flat_space is not None:
             # re-sync the flat buffer with the freshly loaded param data
             for n, off, numel, shape in self.flat_space.segments:
@@ -476,7 +520,11 @@ class Trainer:
             self.start_step = int(training_state.get("step", 0))
             self.total_tokens = int(training_state.get("total_tokens", 0))
             self.validation_losses = list(training_state.get("validation_losses", []))
-        broadcast_module(self.model)
+        # DP replicas re-sync from rank 0; TP shards hold DIFFERENT weights
+        # by design and each already loaded its own file — a world broadcast
+        # would overwrite every shard with tp0's.
+        if self.tp_world == 1:
+            broadcast_module(self.model)
 
     # ------------------------------------------------------------------
     def run_learning_rate_finder(self) -> Dict[str, Any]:
