@@ -253,6 +253,8 @@ class MoE(nn.Module):
         E, H, I = args.num_local_experts, args.hidden_size, args.intermediate_size
         self.num_experts = E
         self.top_k = max(1, int(args.num_experts_per_tok or 1))
+        if self.top_k > E:
+            raise ValueError(f"num_experts_per_tok={self.top_k} > num_local_experts={E}")
         self.router = FastLinear(H, E, bias=False)
         self.w_gate_up = nn.Parameter(torch.empty(E, 2 * I, H))
         self.w_down = nn.Parameter(torch.empty(E, H, I))
