@@ -662,7 +662,8 @@ class Trainer:
             loss_g = all_reduce_scalar(loss.float(), "mean")
             ntok_g = all_reduce_scalar(ntok.float())
             loss_v = float(loss_g.item())
-            ntok_v = int(ntok_g.item())
+            # TP replicas process the SAME tokens: count each token once
+            ntok_v = int(ntok_g.item()) // self.tp_world
             self.total_tokens += ntok_v
 
             if stats_collector is not None:
