@@ -132,6 +132,17 @@ def _shard_cols(weight: torch.nn.Parameter, lo: int, hi: int) -> torch.nn.Parame
     return p
 
 
+def _localize_row_bias(linear: torch.nn.Linear, rank: int) -> None:
+    if linear.bias is None:
+        return
+    if rank == 0:
+        linear.bias._tp_sharded = True  # rank-local: clip reduces it over TP
+    else:
+        with torch.no_grad():
+            linear.bias.zero_()
+        linear.bias.requires_grad_(False)
+
+
 def apply_tensor_parallel(model, rank: int, world: int) -> None:
     """Shard a fully-initialized (broadcast) Llama ``Model`` in place."""
     if world <= 1:
@@ -163,13 +174,15 @@ def apply_tensor_parallel(model, rank: int, world: int) -> None:
         rows = torch.cat([q_rows, k_rows, v_rows])
         attn.wqkv.weight = _shard_rows(attn.wqkv.weight, rows)
         if attn.wqkv.bias is not None:
-            attn.wqkv.bias = _shard_rows(attn.wqkv.bias.unsqueeze(-1), rows).squeeze(-1)
+            attn.wqkv.bias = _shard_rows(attn.wqkv.bias, rows)
         attn.wo.weight = _shard_cols(attn.wo.weight, rank * lq * hd, (rank + 1) * lq * hd)
-        # wo bias is a full-output add: apply it on rank 0 only so the
-        # all-reduce sums it once
-        if attn.wo.bias is not None and rank != 0:
-            with torch.no_grad():
-                attn.wo.bias.zero_()
+        # row-parallel bias: FULL bias on tp-rank 0, zero + FROZEN elsewhere.
+        # (1/tp-scaling on every rank is wrong under optimizers: each rank
+        # applies a full-magnitude update, so the summed bias moves at tp x
+        # the learning rate — and Adam's scale-invariance defeats
+        # grad-rescaling tricks.) Rank 0's bias is flagged _tp_sharded so the
+        # clip norm reduces it across the TP group consistently.
+        _localize_row_bias(attn.wo, rank)
         attn.n_heads, attn.n_kv_heads = lq, lkv
         attn.wqkv.out_features = (lq + 2 * lkv) * hd
         attn.wo.in_features = lq * hd
@@ -183,11 +196,8 @@ def apply_tensor_parallel(model, rank: int, world: int) -> None:
         mlp.w_gate_up.weight = _shard_rows(mlp.w_gate_up.weight, torch.cat([g_rows, u_rows]))
         mlp.w_down.weight = _shard_cols(mlp.w_down.weight, rank * li, (rank + 1) * li)
         if mlp.w_gate_up.bias is not None:
-            mlp.w_gate_up.bias = _shard_rows(
-                mlp.w_gate_up.bias.unsqueeze(-1), torch.cat([g_rows, u_rows])).squeeze(-1)
-        if mlp.w_down.bias is not None and rank != 0:
-            with torch.no_grad():
-                mlp.w_down.bias.zero_()
+            mlp.w_gate_up.bias = _shard_rows(mlp.w_gate_up.bias, torch.cat([g_rows, u_rows]))
+        _localize_row_bias(mlp.w_down, rank)
         mlp.w_gate_up.out_features = 2 * li
         mlp.w_down.in_features = li
         mlp._tp = True

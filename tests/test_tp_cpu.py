@@ -340,3 +340,67 @@ def test_tp_sharded_checkpoint_roundtrip(tmp_path):
     # marker keeps auto-resume discovery working
     from mlx_cuda_distributed_pretraining_amd.core.checkpoint import latest_checkpoint
     assert latest_checkpoint(tmp_path / "save_runs" / "tp-ckpt-save").endswith("step_2")
+
+
+def _tp_bias_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.parallel.dist import broadcast_module
+        from mlx_cuda_distributed_pretraining_amd.parallel.tp import apply_tensor_parallel
+
+        args = _args()
+        args.attention_bias = True
+        args.mlp_bias = True
+        torch.manual_seed(0)
+        model = Model(args)
+        broadcast_module(model)
+        apply_tensor_parallel(model, rank, world)
+        opt = torch.optim.SGD(model.parameters(), lr=1e-2)
+        batch = _batch()
+        for _ in range(3):  # bias drift shows up over optimizer steps
+            logits = model(batch[:, :-1])
+            loss = torch.nn.functional.cross_entropy(
+                logits.reshape(-1, 67), batch[:, 1:].reshape(-1))
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+        q.put({"rank": rank, "logits": logits.detach().numpy().copy()})
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_tp_row_parallel_bias_stays_consistent():
+    """attention_bias/mlp_bias under TP: the 1/tp-scaled row-parallel biases
+    must keep TP output equal to single-process output even AFTER optimizer
+    steps (the failure mode of zeroed non-main biases)."""
+    args = _args()
+    args.attention_bias = True
+    args.mlp_bias = True
+    torch.manual_seed(0)
+    model = Model(args)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-2)
+    batch = _batch()
+    for _ in range(3):
+        logits = model(batch[:, :-1])
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, 67), batch[:, 1:].reshape(-1))
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    ref = logits.detach()
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_tp_bias_worker, args=(r, 2, 29517, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get(), q.get()]
+    for p in procs:
+        p.join(180)
+        assert p.exitcode == 0
+    for r in res:
+        assert torch.allclose(torch.from_numpy(r["logits"]), ref, atol=1e-4), \
+            (torch.from_numpy(r["logits"]) - ref).abs().max()
