@@ -157,6 +157,9 @@ class SystemConfig:
     # MI355X extensions
     distributed_backend: str = "nccl"  # "nccl" is RCCL on ROCm
     bucket_mb: int = 50
+    # with model_parallel: S-shard the inter-sublayer activation stream
+    # (Megatron sequence parallelism; parallel/tp.py)
+    sequence_parallel: bool = False
 
 
 @dataclass

@@ -177,6 +177,7 @@ class Trainer:
         self.tp_world, self.dp_world = 1, self.world_size
         self.tp_rank, self.dp_rank = 0, self.rank
         self.tp_pg = self.dp_pg = None
+        self.sp = False
         if self.config.system.model_parallel and self.world_size > 1:
             mp_size = int(self.config.system.model_parallel_size or self.world_size)
             from ..parallel.tp import apply_tensor_parallel, init_tp_mesh
@@ -186,13 +187,16 @@ class Trainer:
             self.tp_rank, self.dp_rank, self.tp_pg, self.dp_pg = init_tp_mesh(
                 self.rank, self.world_size, mp_size
             )
-            apply_tensor_parallel(self.model, self.tp_rank, mp_size)
+            self.sp = bool(self.config.system.sequence_parallel)
+            apply_tensor_parallel(self.model, self.tp_rank, mp_size,
+                                  sequence_parallel=self.sp)
             self.tp_world = mp_size
             self.dp_world = self.world_size // mp_size
             if self.is_main:
                 self.logger.log(
-                    f"Mesh: TP={self.tp_world} x DP={self.dp_world} "
-                    f"(head/intermediate sharded, 1 all-reduce per sublayer)"
+                    f"Mesh: TP={self.tp_world} x DP={self.dp_world}"
+                    + (" + sequence parallel" if self.sp else "")
+                    + " (head/intermediate sharded, 1 all-reduce per sublayer)"
                 )
         self.logger.log_model_summary(self.model)
 
@@ -306,6 +310,26 @@ class Trainer:
             inputs = inputs[:, :mpe]
             targets = targets[:, :mpe]
         logits = self.model(inputs)
+        if self.sp and self.model.training and logits.shape[1] != targets.shape[1]:
+            # sequence parallelism: logits are S-sharded; shard-local CE,
+            # then a differentiable sum over the TP group (identity backward
+            # keeps each rank's local grads unscaled)
+            from ..parallel.tp import reduce_from_tp
+
+            s_loc = logits.shape[1]
+            tgt = targets[:, self.tp_rank * s_loc:(self.tp_rank + 1) * s_loc]
+            loss_l, ntok_l = fused_cross_entropy(
+                logits.reshape(-1, logits.shape[-1]),
+                tgt.reshape(-1),
+                ignore_index=self.tokenizer.PAD_TOKEN,
+            )
+            loss_sum = reduce_from_tp((loss_l * ntok_l.float()).unsqueeze(0)).squeeze(0)
+            ntok_g = ntok_l.float().clone()
+            if is_distributed():
+                import torch.distributed as dist
+
+                dist.all_reduce(ntok_g, group=self.tp_pg)  # THIS replica's tokens
+            return loss_sum / ntok_g, ntok_g.long()
         loss, ntok = fused_cross_entropy(
             logits.reshape(-1, logits.shape[-1]),
             targets.reshape(-1),
@@ -355,6 +379,10 @@ class Trainer:
             total_tok = ntok if total_tok is None else total_tok + ntok
         if self.ddp is not None:
             self.ddp.finalize()
+        if self.sp:
+            from ..parallel.tp import sp_allreduce_replicated_grads
+
+            sp_allreduce_replicated_grads(self.model)
         if breakdown and on_gpu:
             ev[1].record()
 

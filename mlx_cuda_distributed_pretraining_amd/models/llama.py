@@ -28,7 +28,7 @@ import torch.nn as nn
 from ..ops.attention import attention_ref, flash_attention, rope_flash_attention_qkv
 from ..ops.gemv import FastLinear, linear_fast
 from ..ops.rmsnorm import RMSNorm, add_rms_norm
-from ..parallel.tp import copy_to_tp, reduce_from_tp
+from ..parallel.tp import copy_to_tp, gather_sp, reduce_from_tp, scatter_sp, tp_group
 from ..ops.rope import RopeTable, apply_rope
 from ..ops.swiglu import swiglu
 
@@ -149,9 +149,11 @@ class Attention(nn.Module):
             self.alibi_slopes = None
 
     def forward(self, x: torch.Tensor, cache: Optional[KVCache] = None) -> torch.Tensor:
-        B, S, _ = x.shape
-        if getattr(self, "_tp", False):
+        if getattr(self, "_sp", False) and self.training:
+            x = gather_sp(x)  # SP: S-sharded stream -> full sequence
+        elif getattr(self, "_tp", False):
             x = copy_to_tp(x)  # f: replicated activation enters column-parallel qkv
+        B, S, _ = x.shape
         qkv = self.wqkv(x)
 
         if getattr(cache, "static_decode", False):
@@ -177,6 +179,8 @@ class Attention(nn.Module):
                 alibi_slopes=self.alibi_slopes if self.args.use_alibi else None,
             )
             out = self.wo(o.reshape(B, S, -1))
+            if getattr(self, "_sp", False) and self.training:
+                return scatter_sp(out)
             return reduce_from_tp(out) if getattr(self, "_tp", False) else out
 
         q, k, v = qkv.split(
@@ -213,6 +217,8 @@ class Attention(nn.Module):
                 alibi_slopes=self.alibi_slopes if self.args.use_alibi else None,
             )
         out = self.wo(o.reshape(B, S, -1))
+        if getattr(self, "_sp", False) and self.training:
+            return scatter_sp(out)
         return reduce_from_tp(out) if getattr(self, "_tp", False) else out
 
 
@@ -226,10 +232,15 @@ class MLP(nn.Module):
         self.w_gate_up.fp8 = self.w_down.fp8 = args.fp8
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if getattr(self, "_tp", False):
+        sp = getattr(self, "_sp", False) and self.training
+        if sp:
+            x = gather_sp(x)
+        elif getattr(self, "_tp", False):
             x = copy_to_tp(x)
         # emit_amax: w_down's fp8 quantizer reuses the swiglu-accumulated amax
         out = self.w_down(swiglu(self.w_gate_up(x), emit_amax=self.w_down.fp8))
+        if sp:
+            return scatter_sp(out)
         return reduce_from_tp(out) if getattr(self, "_tp", False) else out
 
 
@@ -412,6 +423,20 @@ class Model(nn.Module):
             and x.shape[0] <= 8
         )
         if not static:
+            sp = getattr(self, "_sp_world", 1)
+            if sp > 1 and self.training and cache is None:
+                # SP entry: keep only this rank's S-rows; the stream stays
+                # sharded through norms/residuals (gather/scatter live inside
+                # the sublayers). Logits come back SHARDED on S — the trainer
+                # computes the shard-local CE and sum-reduces (compute_loss).
+                S = x.shape[1]
+                if S % sp:
+                    raise ValueError(f"sequence_parallel: seq len {S} not divisible by {sp}")
+                loc = S // sp
+                r = torch.distributed.get_rank(
+                    __import__("mlx_cuda_distributed_pretraining_amd.parallel.tp",
+                               fromlist=["tp_group"]).tp_group())                     if torch.distributed.is_initialized() else 0
+                x = x[:, r * loc:(r + 1) * loc]
             # residual-fused layer chain: adds live inside the RMSNorm kernels
             res, delta = x, None
             for i, layer in enumerate(self.layers):

@@ -118,6 +118,97 @@ def reduce_from_tp(x: torch.Tensor) -> torch.Tensor:
     return _ReduceFromTP.apply(x)
 
 
+# ---- sequence parallelism (SP) on top of TP ----------------------------
+# Between sublayers the activation stream is sharded on the SEQUENCE dim
+# (norms + residual adds run on S/tp rows per rank); entering a
+# column-parallel projection it is all-gathered (bwd: reduce-scatter of the
+# per-rank partial grads), and the row-parallel output is reduce-scattered
+# (bwd: all-gather). Same total comm volume as the f/g all-reduces, but the
+# replicated norm/residual work and their activation memory shrink by tp.
+# reduce-scatter is composed as all-reduce + local slice so the SAME code
+# runs on gloo (CPU tests) and RCCL; a fused reduce_scatter_tensor fast
+# path is a ROADMAP item.
+
+
+def _sp_slice(x: torch.Tensor, rank: int, world: int) -> torch.Tensor:
+    S = x.shape[1]
+    loc = S // world
+    return x[:, rank * loc:(rank + 1) * loc].contiguous()
+
+
+def _sp_rank_world():
+    if not dist.is_initialized():
+        return 0, 1
+    if _TP_GROUP is not None:
+        return dist.get_rank(_TP_GROUP), dist.get_world_size(_TP_GROUP)
+    return dist.get_rank(), dist.get_world_size()
+
+
+def _all_gather_seq(x: torch.Tensor) -> torch.Tensor:
+    _, world = _sp_rank_world()
+    parts = [torch.empty_like(x) for _ in range(world)]
+    dist.all_gather(parts, x.contiguous(), group=_TP_GROUP)
+    return torch.cat(parts, dim=1)
+
+
+def _reduce_scatter_seq(x: torch.Tensor) -> torch.Tensor:
+    rank, world = _sp_rank_world()
+    x = x.contiguous()
+    dist.all_reduce(x, group=_TP_GROUP)
+    return _sp_slice(x, rank, world)
+
+
+class _GatherSP(torch.autograd.Function):
+    """S-sharded -> full: all-gather fwd, reduce-scatter bwd."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor) -> torch.Tensor:
+        if _world() == 1:
+            return x
+        return _all_gather_seq(x)
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor) -> torch.Tensor:
+        if _world() == 1:
+            return grad
+        return _reduce_scatter_seq(grad)
+
+
+class _ScatterSP(torch.autograd.Function):
+    """partial-sum full -> S-sharded reduced: reduce-scatter fwd, all-gather bwd."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor) -> torch.Tensor:
+        if _world() == 1:
+            return x
+        return _reduce_scatter_seq(x)
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor) -> torch.Tensor:
+        if _world() == 1:
+            return grad
+        return _all_gather_seq(grad)
+
+
+def gather_sp(x: torch.Tensor) -> torch.Tensor:
+    return _GatherSP.apply(x)
+
+
+def scatter_sp(x: torch.Tensor) -> torch.Tensor:
+    return _ScatterSP.apply(x)
+
+
+def sp_allreduce_replicated_grads(model) -> None:
+    """Under SP the replicated params (embeddings, norms, lm head) see only
+    their local S-rows in backward: sum their grads over the TP group before
+    the optimizer so the replicas move identically."""
+    if _world() == 1:
+        return
+    for p in model.parameters():
+        if p.grad is not None and not getattr(p, "_tp_sharded", False):
+            dist.all_reduce(p.grad, group=_TP_GROUP)
+
+
 def _shard_rows(weight: torch.nn.Parameter, row_idx: torch.Tensor) -> torch.nn.Parameter:
     p = torch.nn.Parameter(weight.detach()[row_idx].contiguous(),
                            requires_grad=weight.requires_grad)
@@ -143,8 +234,11 @@ def _localize_row_bias(linear: torch.nn.Linear, rank: int) -> None:
         linear.bias.requires_grad_(False)
 
 
-def apply_tensor_parallel(model, rank: int, world: int) -> None:
-    """Shard a fully-initialized (broadcast) Llama ``Model`` in place."""
+def apply_tensor_parallel(model, rank: int, world: int,
+                          sequence_parallel: bool = False) -> None:
+    """Shard a fully-initialized (broadcast) Llama ``Model`` in place.
+    sequence_parallel: activations between sublayers are S-sharded during
+    TRAINING forward (gather_sp/scatter_sp replace the f/g all-reduces)."""
     if world <= 1:
         return
     args = model.args
@@ -201,5 +295,7 @@ def apply_tensor_parallel(model, rank: int, world: int) -> None:
         mlp.w_gate_up.out_features = 2 * li
         mlp.w_down.in_features = li
         mlp._tp = True
+        attn._sp = mlp._sp = sequence_parallel
 
     model._tp_world = world
+    model._sp_world = world if sequence_parallel else 1

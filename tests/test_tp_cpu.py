@@ -404,3 +404,89 @@ def test_tp_row_parallel_bias_stays_consistent():
     for r in res:
         assert torch.allclose(torch.from_numpy(r["logits"]), ref, atol=1e-4), \
             (torch.from_numpy(r["logits"]) - ref).abs().max()
+
+
+def _sp_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.core.config import Config
+        from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer
+
+        cfg = Config.from_dict({
+            "name": "sp-test",
+            "overwrite": True,
+            "data": {"synthetic": True, "synthetic_vocab_size": 64,
+                     "preprocessing": {"max_context_size": 33}},  # 32 after shift
+            "model": {"dimensions": {"hidden_size": 32, "intermediate_size": 64,
+                                     "num_layers": 2},
+                      "attention": {"num_heads": 4, "num_kv_heads": 2,
+                                    "max_position_embeddings": 64}},
+            "training": {"hyperparameters": {"iters": 3, "batch_size": 2,
+                                             "learning_rate": 1e-3}},
+            "logging": {"steps": {"logging_interval": 0, "checkpoint_interval": 0,
+                                  "validation_interval": 0}},
+            "system": {"device": "cpu", "distributed": True,
+                       "distributed_backend": "gloo",
+                       "model_parallel": True, "model_parallel_size": 2,
+                       "sequence_parallel": True},
+        })
+        t = Trainer(cfg, runs_root=f"/tmp/sp_runs_{rank}")
+        losses = [float(t.train_step(i)[0]) for i in range(3)]
+        reps = {n: p.detach().numpy().copy() for n, p in t.model.named_parameters()
+                if not getattr(p, "_tp_sharded", False)}
+        q.put({"rank": rank, "losses": losses, "replicated": reps})
+    finally:
+        dist.destroy_process_group()
+
+
+def _sp_reference_losses():
+    """Same config WITHOUT any parallelism, single process."""
+    import shutil
+    from mlx_cuda_distributed_pretraining_amd.core.config import Config
+    from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer
+
+    cfg = Config.from_dict({
+        "name": "sp-ref",
+        "overwrite": True,
+        "data": {"synthetic": True, "synthetic_vocab_size": 64,
+                 "preprocessing": {"max_context_size": 33}},
+        "model": {"dimensions": {"hidden_size": 32, "intermediate_size": 64,
+                                 "num_layers": 2},
+                  "attention": {"num_heads": 4, "num_kv_heads": 2,
+                                "max_position_embeddings": 64}},
+        "training": {"hyperparameters": {"iters": 3, "batch_size": 2,
+                                         "learning_rate": 1e-3}},
+        "logging": {"steps": {"logging_interval": 0, "checkpoint_interval": 0,
+                              "validation_interval": 0}},
+        "system": {"device": "cpu"},
+    })
+    t = Trainer(cfg, runs_root="/tmp/sp_ref_runs")
+    out = [float(t.train_step(i)[0]) for i in range(3)]
+    shutil.rmtree("/tmp/sp_ref_runs", ignore_errors=True)
+    return out
+
+
+@pytest.mark.timeout(300)
+def test_sequence_parallel_matches_single_process():
+    """TP=2 + sequence parallelism: per-step training losses must match the
+    single-process run (same data, same seed), ranks agree, and replicated
+    params stay identical (the post-backward TP-group grad sum works)."""
+    ref = _sp_reference_losses()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_sp_worker, args=(r, 2, 29518, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get(), q.get()]
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    res.sort(key=lambda r: r["rank"])
+    assert res[0]["losses"] == pytest.approx(res[1]["losses"], abs=1e-6)
+    assert res[0]["losses"] == pytest.approx(ref, abs=2e-5)
+    for n, w in res[0]["replicated"].items():
+        assert (w == res[1]["replicated"][n]).all(), f"replica drift on {n}"
