@@ -490,3 +490,51 @@ def test_sequence_parallel_matches_single_process():
     assert res[0]["losses"] == pytest.approx(ref, abs=2e-5)
     for n, w in res[0]["replicated"].items():
         assert (w == res[1]["replicated"][n]).all(), f"replica drift on {n}"
+
+
+def test_merge_tp_checkpoint_roundtrip(tmp_path):
+    """apply_tensor_parallel sharding is exactly inverted by the merge tool
+    (no process group needed: sharding itself is pure slicing)."""
+    import copy
+    import json
+    import yaml
+    from safetensors.torch import save_file
+    from mlx_cuda_distributed_pretraining_amd.parallel.tp import apply_tensor_parallel
+    from tools.merge_tp_checkpoint import merge_checkpoint
+
+    args = _args()
+    args.attention_bias = True
+    args.mlp_bias = True
+    torch.manual_seed(0)
+    full = Model(args)
+    shards = []
+    for r in range(2):
+        m = copy.deepcopy(full)
+        apply_tensor_parallel(m, r, 2)
+        shards.append({k: v.detach().contiguous() for k, v in m.state_dict().items()})
+
+    run = tmp_path / "run"
+    ck = run / "checkpoints"
+    ck.mkdir(parents=True)
+    base = str(ck / "step_5")
+    for r, sd in enumerate(shards):
+        save_file(sd, f"{base}_tp{r}_model.safetensors")
+    (run / "config.yaml").write_text(yaml.safe_dump({
+        "name": "merge-test",
+        "model": {"dimensions": {"hidden_size": args.hidden_size,
+                                 "intermediate_size": args.intermediate_size,
+                                 "num_layers": args.num_layers},
+                  "attention": {"num_heads": args.num_heads,
+                                "num_kv_heads": args.num_kv_heads,
+                                "max_position_embeddings": 64},
+                  "misc": {"attention_bias": True, "mlp_bias": True}},
+    }))
+    (ck / "step_5_state.json").write_text(json.dumps({"step": 5, "tp_world": 2}))
+
+    merge_checkpoint(base)
+    from safetensors.torch import load_file
+    merged = load_file(f"{base}_model.safetensors")
+    want = full.state_dict()
+    assert set(merged.keys()) == set(want.keys())
+    for k in want:
+        assert torch.equal(merged[k], want[k]), k
