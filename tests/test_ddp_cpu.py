@@ -11,6 +11,14 @@ import torch.multiprocessing as mp
 from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
 
 
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _tiny_args():
     return ModelArgs(hidden_size=32, intermediate_size=64, num_layers=2,
                      num_heads=2, num_kv_heads=2, vocab_size=67)
@@ -71,7 +79,7 @@ def test_ddp_grads_match_single_process():
     expected, _params = _single_process_grads()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29511
+    port = _free_port()
     procs = [ctx.Process(target=_worker, args=(r, 2, port, q)) for r in range(2)]
     for p in procs:
         p.start()
@@ -122,7 +130,8 @@ def test_zero1_step_runs_and_syncs():
     all ranks with finite, synchronized params."""
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_worker_zero1, args=(r, 2, 29513, q)) for r in range(2)]
+    port = _free_port()
+    procs = [ctx.Process(target=_worker_zero1, args=(r, 2, port, q)) for r in range(2)]
     for p in procs:
         p.start()
     flat = torch.from_numpy(q.get())

@@ -16,6 +16,14 @@ import torch.multiprocessing as mp
 from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
 
 
+def _free_port():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _args():
     return ModelArgs(hidden_size=32, intermediate_size=64, num_layers=2,
                      num_heads=4, num_kv_heads=2, vocab_size=67,
@@ -75,7 +83,8 @@ def test_tp2_matches_single_process():
     ref_logits, ref_loss, ref_grads = _reference()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_tp_worker, args=(r, 2, 29513, q)) for r in range(2)]
+    port = _free_port()
+    procs = [ctx.Process(target=_tp_worker, args=(r, 2, port, q)) for r in range(2)]
     for p in procs:
         p.start()
     results = [q.get(), q.get()]
@@ -168,8 +177,9 @@ def test_tp_trainer_replicas_stay_identical(tmp_path):
     (clip scale and grads agree)."""
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
+    port = _free_port()
     procs = [ctx.Process(target=_tp_trainer_worker,
-                         args=(r, 2, 29514, q, str(tmp_path / f"runs{r}")))
+                         args=(r, 2, port, q, str(tmp_path / f"runs{r}")))
              for r in range(2)]
     for p in procs:
         p.start()
@@ -236,7 +246,8 @@ def test_dp2_x_tp2_mesh_grads_match_single_process():
 
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_mesh_worker, args=(r, 4, 29515, q)) for r in range(4)]
+    port = _free_port()
+    procs = [ctx.Process(target=_mesh_worker, args=(r, 4, port, q)) for r in range(4)]
     for p in procs:
         p.start()
     res = [q.get() for _ in range(4)]
@@ -323,8 +334,9 @@ def test_tp_sharded_checkpoint_roundtrip(tmp_path):
     ctx = mp.get_context("spawn")
     for phase, runs in (("save", tmp_path / "save_runs"), ("load", tmp_path / "load_runs")):
         q = ctx.SimpleQueue()
+        port = _free_port()
         procs = [ctx.Process(target=_tp_ckpt_worker,
-                             args=(r, 2, 29516, q, str(runs), phase)) for r in range(2)]
+                             args=(r, 2, port, q, str(runs), phase)) for r in range(2)]
         for p in procs:
             p.start()
         res = [q.get(), q.get()]
@@ -394,7 +406,8 @@ def test_tp_row_parallel_bias_stays_consistent():
 
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_tp_bias_worker, args=(r, 2, 29517, q)) for r in range(2)]
+    port = _free_port()
+    procs = [ctx.Process(target=_tp_bias_worker, args=(r, 2, port, q)) for r in range(2)]
     for p in procs:
         p.start()
     res = [q.get(), q.get()]
@@ -478,7 +491,8 @@ def test_sequence_parallel_matches_single_process():
     ref = _sp_reference_losses()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_sp_worker, args=(r, 2, 29518, q)) for r in range(2)]
+    port = _free_port()
+    procs = [ctx.Process(target=_sp_worker, args=(r, 2, port, q)) for r in range(2)]
     for p in procs:
         p.start()
     res = [q.get(), q.get()]
