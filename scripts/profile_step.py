@@ -40,8 +40,8 @@ def main():
     groups = {}
     for name, calls, dur, avg in rows:
         groups[group_of(name)] = groups.get(group_of(name), 0) + dur
-    print(f"Total GPU kernel time: {tot/1e6:.3f} s ({tot/a.steps/1e3:.1f} ms/step)\n")
-    print("| group | ms/step | share |\n|---|---|---|")
+    print(f"Total GPU kernel time: {tot/1e6:.3f} s over {a.steps} step(s) ({tot/a.steps/1e3:.1f} ms/step GPU-busy)\n")
+    print("| group | ms (whole capture) | share |\n|---|---|---|")
     for g, d in sorted(groups.items(), key=lambda x: -x[1]):
         print(f"| {g} | {d/a.steps/1e3:.1f} | {100*d/tot:.1f}% |")
     if a.top:
