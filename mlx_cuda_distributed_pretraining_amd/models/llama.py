@@ -114,6 +114,12 @@ class KVCache:
     def offset(self) -> int:
         return 0 if self.k is None else self.k.shape[1]
 
+    def trim(self, n: int) -> None:
+        """Drop the last n cached positions (speculative-decoding rewind)."""
+        if self.k is not None and n > 0:
+            self.k = self.k[:, :-n].contiguous()
+            self.v = self.v[:, :-n].contiguous()
+
     def update(self, k: torch.Tensor, v: torch.Tensor):
         if self.k is None:
             self.k, self.v = k, v

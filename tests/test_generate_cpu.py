@@ -62,3 +62,31 @@ def test_repetition_penalty():
     out = procs[0]([1, 2, 3], logits)
     assert out[1] == 0.5 and out[2] == 0.5 and out[3] == 0.5
     assert out[0] == 1.0
+
+
+def test_speculative_decoding_matches_greedy():
+    """Draft-assisted decoding must be token-for-token identical to
+    target-only greedy generation (for ANY draft: the draft only changes
+    acceptance rate, never the output). Exercises partial acceptance
+    (random draft), full acceptance (draft == target), k=1, and a 1-token
+    prompt."""
+    import torch
+    from mlx_cuda_distributed_pretraining_amd.inference.generate import generate_step
+    from mlx_cuda_distributed_pretraining_amd.inference.speculative import (
+        speculative_generate_tokens)
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+
+    args = ModelArgs(hidden_size=32, intermediate_size=64, num_layers=2,
+                     num_heads=2, num_kv_heads=2, vocab_size=61,
+                     max_position_embeddings=128)
+    torch.manual_seed(0)
+    target = Model(args).eval()
+    torch.manual_seed(7)
+    draft = Model(args).eval()
+
+    for prompt in ([5, 17, 3, 9], [11]):
+        want = list(generate_step(target, prompt, max_tokens=24))
+        for d, kk in ((draft, 4), (draft, 1), (draft, 7), (target, 4)):
+            got = speculative_generate_tokens(target, d, prompt,
+                                              max_tokens=24, k=kk)
+            assert got == want, (kk, d is target, got, want)
