@@ -144,11 +144,18 @@ class StatsClient:
                                             "info": self.info})
                         self.connected.set()
                         last_hb = time.time()
-                        while not self._stop.is_set():
+                        while True:
+                            # drain the buffer before honoring stop(): the
+                            # trainer enqueues its final step metrics moments
+                            # before stopping, and dropping them loses the
+                            # tail of every short run
                             try:
                                 msg = self._out.get_nowait()
                                 await ws.send_json(msg)
+                                continue
                             except queue.Empty:
+                                if self._stop.is_set():
+                                    break
                                 await asyncio.sleep(0.05)
                             if time.time() - last_hb > HEARTBEAT_INTERVAL_S:
                                 await ws.send_json({"type": "heartbeat",
