@@ -153,15 +153,26 @@ class Trainer:
         if self.config.model.architecture == "llama_standard":
             args.attention_type = "simple"
         self.model_args = args
-        # Build directly ON the training device: CPU weight init is
-        # single-threaded torch.normal_ and takes minutes at 7B scale, while
-        # the same init on the GPU is sub-second (288 GB HBM holds the whole
-        # model anyway).
-        try:
-            with torch.device(self.device):
-                self.model = Model(args)
-        except Exception:
-            self.model = Model(args)  # CPU init fallback (slow but safe)
+        # Multi-billion-parameter models build directly ON the training
+        # device: CPU weight init is single-threaded torch.normal_ and takes
+        # minutes at 7B scale, while the same init on the GPU is sub-second
+        # (288 GB HBM holds the whole model anyway). Smaller models keep the
+        # plain CPU-init path.
+        est_params = (
+            args.vocab_size * args.hidden_size
+            + args.num_layers
+            * (args.hidden_size * (args.num_heads + 2 * args.num_kv_heads) * args.head_dim
+               + args.hidden_size * args.num_heads * args.head_dim
+               + 3 * args.hidden_size * args.intermediate_size)
+        )
+        if est_params > 3_000_000_000 and self.device.type == "cuda":
+            try:
+                with torch.device(self.device):
+                    self.model = Model(args)
+            except Exception:
+                self.model = Model(args)  # CPU init fallback (slow but safe)
+        else:
+            self.model = Model(args)
         self.model = self.model.to(device=self.device, dtype=self.param_dtype)
         if self.config.system.gradient_checkpointing:
             ratio = self.config.system.gradient_checkpointing_ratio
