@@ -172,6 +172,17 @@ class _FlashAttnFn(torch.autograd.Function):
         )
 
 
+
+def _norm_alibi(alibi_slopes, num_heads: int, device) -> "Optional[torch.Tensor]":
+    """Accept either a per-head slope tensor or an int head count (builds the
+    standard 2^(-8(i+1)/H) slopes) — matches the reference's loose API."""
+    if alibi_slopes is None or isinstance(alibi_slopes, torch.Tensor):
+        return alibi_slopes
+    h = int(num_heads)
+    return torch.tensor([2 ** (-8.0 * (i + 1) / h) for i in range(h)],
+                        dtype=torch.float32, device=device)
+
+
 def flash_attention(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -183,6 +194,7 @@ def flash_attention(
     alibi_slopes: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     """Tiled attention, [B,Hq,S,D] x [B,Hkv,S,D] -> [B,Hq,S,D]."""
+    alibi_slopes = _norm_alibi(alibi_slopes, q.shape[2], q.device)
     return _FlashAttnFn.apply(q, k, v, causal, scale, window, prefix_len, alibi_slopes)
 
 
