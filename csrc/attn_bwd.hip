@@ -11,12 +11,17 @@
 // (merged, the two 32x128 fp32 accumulators alone are 128 VGPRs).
 //
 // Same gfx950 structure as the forward (attn_fwd.hip): 8 waves/block sharing
-// every staged tile, XOR-swizzled LDS images (guide §6 Guideline 4 — padding
-// does not fix the D=128 column-read conflict), DOUBLE-BUFFERED tiles with
-// one barrier per iteration, T14 register-prefetch of the next tile under
-// the MFMA clusters, base-2 exponentials with log2(e) folded into the scale,
-// mask-free fast path on interior causal tiles, s_setprio around MFMAs,
-// MFMA layout + acc_to_afrag transform from attn_common.h.
+// every staged tile, DOUBLE-BUFFERED tiles with one barrier per iteration,
+// T14 register-prefetch of the next tile under the MFMA clusters, base-2
+// exponentials with log2(e) folded into the scale, mask-free fast path on
+// interior causal tiles, s_setprio around MFMAs, MFMA layout +
+// acc_to_afrag transform from attn_common.h.
+//
+// The bwd-specific layout trick: ALL images are row-major with the s_tr
+// swizzle (attn_common.h), conflict-free for BOTH read shapes — b128 row
+// reads AND the k-strided B-fragment gathers, which use gfx950's
+// ds_read_b64_tr_b16 hardware transpose-read (tr16_frag). No transposed
+// LDS copies are staged anywhere in the backward.
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include "attn_common.h"
