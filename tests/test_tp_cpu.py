@@ -552,3 +552,83 @@ def test_merge_tp_checkpoint_roundtrip(tmp_path):
     assert set(merged.keys()) == set(want.keys())
     for k in want:
         assert torch.equal(merged[k], want[k]), k
+
+
+def _sp_accum_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.core.config import Config
+        from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer
+
+        cfg = Config.from_dict({
+            "name": "sp-accum",
+            "overwrite": True,
+            "data": {"synthetic": True, "synthetic_vocab_size": 64,
+                     "preprocessing": {"max_context_size": 33}},
+            "model": {"dimensions": {"hidden_size": 32, "intermediate_size": 64,
+                                     "num_layers": 2},
+                      "attention": {"num_heads": 4, "num_kv_heads": 2,
+                                    "max_position_embeddings": 64}},
+            "training": {"hyperparameters": {"iters": 2, "batch_size": 2,
+                                             "gradient_accumulation_steps": 2,
+                                             "learning_rate": 1e-3}},
+            "logging": {"steps": {"logging_interval": 0, "checkpoint_interval": 0,
+                                  "validation_interval": 0}},
+            "system": {"device": "cpu", "distributed": True,
+                       "distributed_backend": "gloo",
+                       "model_parallel": True, "model_parallel_size": 2,
+                       "sequence_parallel": True},
+        })
+        t = Trainer(cfg, runs_root=f"/tmp/sp_accum_runs_{rank}")
+        losses = [float(t.train_step(i)[0]) for i in range(2)]
+        q.put({"rank": rank, "losses": losses})
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_sequence_parallel_with_grad_accum_matches_single_process():
+    """SP + gradient accumulation: the once-per-step replicated-grad sum is
+    linear over micro-batches, so losses must still match the plain
+    single-process accumulation run."""
+    import shutil
+    from mlx_cuda_distributed_pretraining_amd.core.config import Config
+    from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer
+
+    cfg = Config.from_dict({
+        "name": "sp-accum-ref",
+        "overwrite": True,
+        "data": {"synthetic": True, "synthetic_vocab_size": 64,
+                 "preprocessing": {"max_context_size": 33}},
+        "model": {"dimensions": {"hidden_size": 32, "intermediate_size": 64,
+                                 "num_layers": 2},
+                  "attention": {"num_heads": 4, "num_kv_heads": 2,
+                                "max_position_embeddings": 64}},
+        "training": {"hyperparameters": {"iters": 2, "batch_size": 2,
+                                         "gradient_accumulation_steps": 2,
+                                         "learning_rate": 1e-3}},
+        "logging": {"steps": {"logging_interval": 0, "checkpoint_interval": 0,
+                              "validation_interval": 0}},
+        "system": {"device": "cpu"},
+    })
+    t = Trainer(cfg, runs_root="/tmp/sp_accum_ref")
+    ref = [float(t.train_step(i)[0]) for i in range(2)]
+    shutil.rmtree("/tmp/sp_accum_ref", ignore_errors=True)
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_sp_accum_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get(), q.get()]
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    res.sort(key=lambda r: r["rank"])
+    assert res[0]["losses"] == pytest.approx(res[1]["losses"], abs=1e-6)
+    assert res[0]["losses"] == pytest.approx(ref, abs=5e-5)
