@@ -103,7 +103,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv_ex", &gemv_ex,
         "fused decode GEMV (rmsnorm/swiglu staging, residual epilogue)");
   m.def("fp8_quantize", &fp8_quantize, "fused bf16 -> e4m3 quantize (codes, scale)");
-  m.def("fp8_quantize_pre", &fp8_quantize_pre, "e4m3 quantize with producer-supplied amax bits");
+  m.def("fp8_quantize_pre", &fp8_quantize_pre, "e4m3 quantize with producer-supplied per-block amax partials");
   m.def("mfma_tile_test", &mfma_tile_test, "debug: one 32x32x16 MFMA tile");
   m.def("afrag_transform_test", &afrag_transform_test, "debug: acc->A-frag transform");
   m.def("tr16_frag_test", &tr16_frag_test, "debug: ds_read_b64_tr_b16 B-fragment gather");
