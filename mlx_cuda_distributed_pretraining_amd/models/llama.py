@@ -275,7 +275,9 @@ class MoE(nn.Module):
         self.router = FastLinear(H, E, bias=False)
         self.w_gate_up = nn.Parameter(torch.empty(E, 2 * I, H))
         self.w_down = nn.Parameter(torch.empty(E, H, I))
-        std = 0.02
+        # same depth-scaled init as the dense projections (_init_weights
+        # only touches nn.Linear/nn.Embedding, not raw expert Parameters)
+        std = 0.02 / math.sqrt(2 * args.num_layers)
         nn.init.normal_(self.w_gate_up, mean=0.0, std=std)
         nn.init.normal_(self.w_down, mean=0.0, std=std)
         self.aux_loss: Optional[torch.Tensor] = None
