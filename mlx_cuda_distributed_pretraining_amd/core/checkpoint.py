@@ -171,21 +171,27 @@ def update_metadata(run_dir: Path, entry: Dict[str, Any]) -> None:
 
 
 def latest_checkpoint(run_dir: str | Path) -> str | None:
-    """Base path (``.../checkpoints/step_<N>``) of the run's newest step
-    checkpoint, or None. Used by ``--auto-resume`` (elastic restarts):
-    numeric steps are preferred over 'final' (a finished run has nothing to
-    resume)."""
+    """Base path (``.../checkpoints/step_<N>`` or ``step_emergency_<N>``) of
+    the run's newest checkpoint, or None. Used by ``--auto-resume`` (elastic
+    restarts). Emergency checkpoints (written on training-loop failure)
+    count too — after a crash they ARE the newest state; at equal step the
+    regular snapshot wins. 'final' is ignored (a finished run has nothing
+    to resume)."""
     ckdir = Path(run_dir) / "checkpoints"
     if not ckdir.is_dir():
         return None
-    best = -1
+    best = (-1, 0)  # (step, regular-beats-emergency)
+    best_name = None
     for pth in ckdir.glob("step_*_state.json"):
-        m = re.match(r"step_(\d+)_state\.json", pth.name)
+        m = re.match(r"step_(emergency_)?(\d+)_state\.json", pth.name)
         if m:
-            best = max(best, int(m.group(1)))
-    if best < 0:
+            key = (int(m.group(2)), 0 if m.group(1) else 1)
+            if key > best:
+                best = key
+                best_name = pth.name[: -len("_state.json")]
+    if best_name is None:
         return None
-    return str(ckdir / f"step_{best}")
+    return str(ckdir / best_name)
 
 
 def rotate_snapshots(checkpoint_dir: Path, max_snapshots: int) -> None:

@@ -197,3 +197,19 @@ def test_auto_resume_end_to_end(tmp_path):
     assert latest_checkpoint(runs / "autoresume-test").endswith("step_6")
     text = (runs / "autoresume-test" / "log.txt").read_text()
     assert "Resumed from" in text
+
+
+def test_latest_checkpoint_prefers_emergency_when_newer(tmp_path):
+    from mlx_cuda_distributed_pretraining_amd.core.checkpoint import latest_checkpoint
+
+    ck = tmp_path / "run" / "checkpoints"
+    ck.mkdir(parents=True)
+    for name in ("step_2_state.json", "step_emergency_5_state.json",
+                 "step_final_state.json"):
+        (ck / name).write_text("{}")
+    assert latest_checkpoint(tmp_path / "run").endswith("step_emergency_5")
+    # regular snapshot at the same (or higher) step wins
+    (ck / "step_5_state.json").write_text("{}")
+    assert latest_checkpoint(tmp_path / "run").endswith("step_5")
+    (ck / "step_emergency_7_state.json").write_text("{}")
+    assert latest_checkpoint(tmp_path / "run").endswith("step_emergency_7")
