@@ -117,6 +117,13 @@ class FusedFlatAdamW:
         }
 
     def load_state_dict(self, sd: dict) -> None:
+        if sd["master"].numel() != self.master.numel():
+            raise ValueError(
+                f"optimizer state shard size mismatch: checkpoint has "
+                f"{sd['master'].numel()} master elements, this run expects "
+                f"{self.master.numel()} — ZeRO-1 checkpoints resume at the "
+                f"same world size (re-sharding is a ROADMAP item)"
+            )
         self.step_count = int(sd["step"])
         self.master.copy_(sd["master"].to(self.master.device))
         self.exp_avg.copy_(sd["exp_avg"].to(self.exp_avg.device))
