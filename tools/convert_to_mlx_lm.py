@@ -89,6 +89,9 @@ def hf_config_from(config: Config, args: ModelArgs) -> Dict:
         "max_position_embeddings": int(args.max_position_embeddings or 2048),
         "rms_norm_eps": args.rms_norm_eps,
         "rope_theta": args.rope_theta,
+        # linear rope scaling survives the export (HF "rope_scaling" dict)
+        **({"rope_scaling": {"type": "linear", "factor": float(args.rope_scaling)}}
+           if args.rope_scaling else {}),
         "tie_word_embeddings": args.tie_word_embeddings,
         "hidden_act": "silu",
         "attention_bias": args.attention_bias,
