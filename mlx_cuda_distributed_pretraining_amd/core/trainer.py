@@ -732,9 +732,10 @@ class Trainer:
                     epochs=cfg.training.epochs,
                     steps_per_epoch=self.steps_per_epoch,
                     grad_accum_steps=self.grad_accum_steps,
+                    # TP replicas share data: only DP multiplies the batch
                     effective_batch_size=self.batch_size
                     * self.grad_accum_steps
-                    * self.world_size,
+                    * self.dp_world,
                 )
                 if cfg.logging.log_step_breakdown:
                     line += self.format_breakdown()
