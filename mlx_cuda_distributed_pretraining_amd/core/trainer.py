@@ -394,6 +394,12 @@ class Trainer:
             from ..parallel.tp import sp_allreduce_replicated_grads
 
             sp_allreduce_replicated_grads(self.model)
+        elif self.tp_world > 1:
+            # EP MoE routers have partial grads (each rank backprops only its
+            # local experts' gate terms) — sum them over the group
+            from ..parallel.tp import ep_allreduce_router_grads
+
+            ep_allreduce_router_grads(self.model)
         if breakdown and on_gpu:
             ev[1].record()
 

@@ -283,10 +283,17 @@ class MoE(nn.Module):
         self.aux_loss: Optional[torch.Tensor] = None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        # expert parallelism (parallel/tp.py _shard_experts): activations
+        # replicated, experts sharded — this rank computes its local experts'
+        # contributions for every token and the g all-reduce sums the
+        # partials (no all-to-all needed in this form)
+        ep = getattr(self, "_tp", False)
+        if ep:
+            x = copy_to_tp(x)
         shape = x.shape
         xf = x.reshape(-1, shape[-1])
         n = xf.shape[0]
-        probs = torch.softmax(self.router(xf).float(), dim=-1)  # [N, E]
+        probs = torch.softmax(self.router(xf).float(), dim=-1)  # [N, E] (full E)
         gates, idx = probs.topk(self.top_k, dim=-1)             # [N, k]
         gates = (gates / gates.sum(-1, keepdim=True)).to(x.dtype)
         # Switch aux loss: E * sum_e fraction_routed_e * mean_prob_e  (==1 at
@@ -297,15 +304,25 @@ class MoE(nn.Module):
                                 torch.ones(idx.numel(), device=x.device))
             frac = counts / (n * self.top_k)
         self.aux_loss = self.num_experts * (frac * probs.mean(0)).sum()
+        if ep:
+            # computed identically on every EP rank, and the trainer SUMS the
+            # router grads over the group -> pre-divide so the aux term
+            # arrives unscaled (parallel/tp.py ep_allreduce_router_grads)
+            self.aux_loss = self.aux_loss / self._ep_world
+        n_local = self.w_gate_up.shape[0]
+        e0 = self._ep_rank * n_local if ep else 0
         out = torch.zeros_like(xf)
-        for e in range(self.num_experts):
+        for el in range(n_local):
+            e = e0 + el
             rows, slot = (idx == e).nonzero(as_tuple=True)
             if rows.numel() == 0:
                 continue
             toks = xf.index_select(0, rows)
-            h = swiglu(toks @ self.w_gate_up[e].t())
-            y = (h @ self.w_down[e].t()) * gates[rows, slot].unsqueeze(-1)
+            h = swiglu(toks @ self.w_gate_up[el].t())
+            y = (h @ self.w_down[el].t()) * gates[rows, slot].unsqueeze(-1)
             out.index_add_(0, rows, y.to(out.dtype))
+        if ep:
+            return reduce_from_tp(out).reshape(shape)
         return out.reshape(shape)
 
 

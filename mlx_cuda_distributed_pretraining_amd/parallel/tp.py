@@ -234,6 +234,46 @@ def _localize_row_bias(linear: torch.nn.Linear, rank: int) -> None:
         linear.bias.requires_grad_(False)
 
 
+def _shard_experts(moe, rank: int, world: int, sequence_parallel: bool) -> None:
+    """Expert parallelism for MoE layers inside the model-parallel group:
+    experts are sharded across ranks while activations stay replicated, so
+    routing needs NO all-to-all — each rank computes its local experts'
+    contributions for every token and the row-parallel g all-reduce sums
+    them (capacity-bound token dispatch with a real all-to-all is the
+    bandwidth-optimal follow-up, ROADMAP).
+
+    The ROUTER stays replicated but its gradient is PARTIAL per rank (each
+    rank only backprops its local experts' gate terms) -> the trainer sums
+    router grads over the group (ep_allreduce_router_grads). The aux
+    load-balance loss is computed identically on every rank, so MoE.forward
+    pre-divides it by the EP degree to survive that sum unscaled."""
+    if sequence_parallel:
+        raise NotImplementedError("SP + MoE expert parallelism: ROADMAP")
+    E = moe.num_experts
+    if E % world:
+        raise ValueError(f"EP degree {world} must divide num_local_experts={E}")
+    le = E // world
+    moe.w_gate_up = torch.nn.Parameter(
+        moe.w_gate_up.detach()[rank * le:(rank + 1) * le].contiguous())
+    moe.w_down = torch.nn.Parameter(
+        moe.w_down.detach()[rank * le:(rank + 1) * le].contiguous())
+    moe.w_gate_up._tp_sharded = True
+    moe.w_down._tp_sharded = True
+    moe.router.weight._ep_router = True  # grads summed over the group
+    moe._ep_rank, moe._ep_world = rank, world
+    moe._tp = True
+
+
+def ep_allreduce_router_grads(model) -> None:
+    """Sum the (partial) router gradients of EP MoE layers over the TP
+    group after backward — see _shard_experts."""
+    if _world() == 1:
+        return
+    for p in model.parameters():
+        if p.grad is not None and getattr(p, "_ep_router", False):
+            dist.all_reduce(p.grad, group=_TP_GROUP)
+
+
 def apply_tensor_parallel(model, rank: int, world: int,
                           sequence_parallel: bool = False) -> None:
     """Shard a fully-initialized (broadcast) Llama ``Model`` in place.
@@ -242,8 +282,6 @@ def apply_tensor_parallel(model, rank: int, world: int,
     if world <= 1:
         return
     args = model.args
-    if getattr(args, "num_local_experts", 0):
-        raise NotImplementedError("TP + MoE needs expert parallelism (ROADMAP)")
     if args.num_heads % world or args.num_kv_heads % world:
         raise ValueError(
             f"TP degree {world} must divide num_heads={args.num_heads} and "
@@ -285,6 +323,9 @@ def apply_tensor_parallel(model, rank: int, world: int,
         attn._tp = True
 
         mlp = layer.mlp
+        if hasattr(mlp, "num_experts"):  # MoE: expert parallelism
+            _shard_experts(mlp, rank, world, sequence_parallel)
+            continue
         g_rows = torch.arange(rank * li, (rank + 1) * li, device=dev)
         u_rows = inter + g_rows
         mlp.w_gate_up.weight = _shard_rows(mlp.w_gate_up.weight, torch.cat([g_rows, u_rows]))
