@@ -720,3 +720,62 @@ def test_expert_parallel_matches_single_process():
                 continue  # attention shards covered by the TP test
             else:
                 assert torch.allclose(g, full, atol=1e-5), f"replicated grad {n}"
+
+
+def _ep_trainer_worker(rank, world, port, q, runs_root):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.core.config import Config
+        from mlx_cuda_distributed_pretraining_amd.core.trainer import Trainer
+
+        cfg = Config.from_dict({
+            "name": "ep-trainer",
+            "overwrite": True,
+            "data": {"synthetic": True, "synthetic_vocab_size": 64,
+                     "preprocessing": {"max_context_size": 32}},
+            "model": {"dimensions": {"hidden_size": 32, "intermediate_size": 48,
+                                     "num_layers": 2, "num_local_experts": 4,
+                                     "num_experts_per_tok": 2},
+                      "attention": {"num_heads": 4, "num_kv_heads": 2,
+                                    "max_position_embeddings": 64}},
+            "training": {"hyperparameters": {"iters": 3, "batch_size": 2,
+                                             "learning_rate": 1e-3,
+                                             "gradient_clip": 1.0}},
+            "logging": {"steps": {"logging_interval": 0, "checkpoint_interval": 0,
+                                  "validation_interval": 0}},
+            "system": {"device": "cpu", "distributed": True,
+                       "distributed_backend": "gloo",
+                       "model_parallel": True, "model_parallel_size": 2},
+        })
+        t = Trainer(cfg, runs_root=runs_root)
+        losses = [float(t.train_step(i)[0]) for i in range(3)]
+        router = t.model.layers[0].mlp.router.weight.detach().numpy().copy()
+        q.put({"rank": rank, "losses": losses, "router": router})
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ep_trainer_router_replicas_stay_identical(tmp_path):
+    """MoE + EP through the full Trainer (router-grad group-sum hook runs in
+    train_step): losses agree across ranks each step and the REPLICATED
+    router weights stay bit-identical after clipped optimizer updates."""
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_ep_trainer_worker,
+                         args=(r, 2, port, q, str(tmp_path / f"runs{r}")))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get(), q.get()]
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    res.sort(key=lambda r: r["rank"])
+    assert res[0]["losses"] == pytest.approx(res[1]["losses"], abs=1e-6)
+    assert (res[0]["router"] == res[1]["router"]).all(), "router replica drift"
