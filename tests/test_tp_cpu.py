@@ -779,3 +779,49 @@ def test_ep_trainer_router_replicas_stay_identical(tmp_path):
     res.sort(key=lambda r: r["rank"])
     assert res[0]["losses"] == pytest.approx(res[1]["losses"], abs=1e-6)
     assert (res[0]["router"] == res[1]["router"]).all(), "router replica drift"
+
+
+def test_merge_ep_checkpoint_roundtrip(tmp_path):
+    """EP expert shards merge back to the full stacked expert tensors."""
+    import copy
+    import json
+    import yaml
+    from safetensors.torch import load_file, save_file
+    from mlx_cuda_distributed_pretraining_amd.parallel.tp import apply_tensor_parallel
+    from tools.merge_tp_checkpoint import merge_checkpoint
+
+    args = _args()
+    args.num_local_experts = 4
+    args.num_experts_per_tok = 2
+    torch.manual_seed(0)
+    full = Model(args)
+    shards = []
+    for r in range(2):
+        m = copy.deepcopy(full)
+        apply_tensor_parallel(m, r, 2)
+        shards.append({k: v.detach().contiguous() for k, v in m.state_dict().items()})
+
+    run = tmp_path / "run"
+    ck = run / "checkpoints"
+    ck.mkdir(parents=True)
+    base = str(ck / "step_3")
+    for r, sd in enumerate(shards):
+        save_file(sd, f"{base}_tp{r}_model.safetensors")
+    (run / "config.yaml").write_text(yaml.safe_dump({
+        "name": "merge-ep",
+        "model": {"dimensions": {"hidden_size": args.hidden_size,
+                                 "intermediate_size": args.intermediate_size,
+                                 "num_layers": args.num_layers,
+                                 "num_local_experts": 4,
+                                 "num_experts_per_tok": 2},
+                  "attention": {"num_heads": args.num_heads,
+                                "num_kv_heads": args.num_kv_heads,
+                                "max_position_embeddings": 64}},
+    }))
+    (ck / "step_3_state.json").write_text(json.dumps({"step": 3, "tp_world": 2}))
+    merge_checkpoint(base)
+    merged = load_file(f"{base}_model.safetensors")
+    want = full.state_dict()
+    assert set(merged.keys()) == set(want.keys())
+    for k in want:
+        assert torch.equal(merged[k], want[k]), k

@@ -40,7 +40,11 @@ def merge_tp_state_dicts(shards, num_heads: int, num_kv_heads: int, head_dim: in
     out: Dict[str, torch.Tensor] = {}
     for name, t0 in shards[0].items():
         parts = [sd[name] for sd in shards]
-        if "wqkv" in name:
+        if t0.dim() == 3:
+            # stacked MoE expert weights: expert-parallel shards concat on
+            # the expert dim (parallel/tp.py _shard_experts)
+            out[name] = torch.cat(parts, dim=0)
+        elif "wqkv" in name:
             q = torch.cat([p[:lq] for p in parts], dim=0)
             k = torch.cat([p[lq:lq + lkv] for p in parts], dim=0)
             v = torch.cat([p[lq + lkv:] for p in parts], dim=0)
