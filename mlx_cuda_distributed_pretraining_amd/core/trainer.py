@@ -321,6 +321,21 @@ class Trainer:
             inputs = inputs[:, :mpe]
             targets = targets[:, :mpe]
         logits = self.model(inputs)
+        vp0 = getattr(self.model, "_vp_vocab0", -1)
+        if vp0 >= 0 and self.model.training:
+            # vocab-parallel CE over the sharded lm head (parallel/tp.py):
+            # the full [N, V] logits replica never materializes
+            from ..parallel.tp import vocab_parallel_cross_entropy
+
+            loss, ntok = vocab_parallel_cross_entropy(
+                logits.reshape(-1, logits.shape[-1]),
+                targets.reshape(-1), vp0,
+                ignore_index=self.tokenizer.PAD_TOKEN,
+            )
+            aux = getattr(self.model, "aux_loss", None)
+            if aux is not None:
+                loss = loss + self.model_args.router_aux_loss_coef * aux
+            return loss, ntok
         if self.sp and self.model.training and logits.shape[1] != targets.shape[1]:
             # sequence parallelism: logits are S-sharded; shard-local CE,
             # then a differentiable sum over the TP group (identity backward

@@ -476,7 +476,20 @@ class Model(nn.Module):
                 x = self.norm(res)
             else:
                 _, x = add_rms_norm(res, delta, self.norm.weight, self.norm.eps)
-            if self.args.tie_word_embeddings:
+            vp0 = getattr(self, "_vp_vocab0", -1)
+            if vp0 >= 0:
+                # vocab-parallel lm head (parallel/tp.py): SHARDED logits in
+                # training (the trainer runs vocab_parallel_cross_entropy);
+                # gathered for eval/generation API compatibility
+                x = copy_to_tp(x)
+                logits = self.output(x)
+                if not self.training:
+                    import torch.distributed as dist_
+
+                    parts = [torch.empty_like(logits) for _ in range(self._tp_world)]
+                    dist_.all_gather(parts, logits.contiguous(), group=tp_group())
+                    logits = torch.cat(parts, dim=-1)
+            elif self.args.tie_word_embeddings:
                 logits = linear_fast(x, self.tok_embeddings.weight)
             else:
                 logits = self.output(x)

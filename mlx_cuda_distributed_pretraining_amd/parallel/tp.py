@@ -234,6 +234,42 @@ def _localize_row_bias(linear: torch.nn.Linear, rank: int) -> None:
         linear.bias.requires_grad_(False)
 
 
+def vocab_parallel_cross_entropy(logits_local: torch.Tensor, targets: torch.Tensor,
+                                 v0: int, ignore_index: int = -100):
+    """Cross entropy over VOCAB-SHARDED logits ([N, V/tp] on each rank,
+    this rank owning vocab ids [v0, v0 + V/tp)). Returns (mean_loss, ntok)
+    — numerically the full-vocab CE, but the [N, V] logits replica never
+    exists. Composition keeps autograd exact: the logsumexp shift uses a
+    DETACHED group max (lse is shift-invariant), and the cross-rank sums go
+    through reduce_from_tp (sum forward / identity backward), so each
+    rank's backward produces exactly its shard of the softmax-minus-onehot
+    gradient."""
+    lf = logits_local.float()
+    n, v_loc = lf.shape
+    mask = targets != ignore_index
+    ntok = mask.sum()
+
+    with torch.no_grad():
+        m = lf.max(dim=-1).values
+        if _world() > 1:
+            m = m.contiguous()
+            dist.all_reduce(m, op=dist.ReduceOp.MAX, group=_TP_GROUP)
+    se_local = torch.exp(lf - m.unsqueeze(-1)).sum(-1, keepdim=False)
+    se = reduce_from_tp(se_local.unsqueeze(0)).squeeze(0)
+    lse = m + torch.log(se)
+
+    t_local = targets.clamp(min=0) - v0
+    in_shard = mask & (t_local >= 0) & (t_local < v_loc)
+    safe = t_local.clamp(0, v_loc - 1)
+    picked = lf.gather(-1, safe.unsqueeze(-1)).squeeze(-1)
+    tgt_local = torch.where(in_shard, picked, torch.zeros_like(picked))
+    tgt = reduce_from_tp(tgt_local.unsqueeze(0)).squeeze(0)
+
+    loss_rows = torch.where(mask, lse - tgt, torch.zeros_like(lse))
+    loss = loss_rows.sum() / ntok.clamp(min=1).float()
+    return loss, ntok
+
+
 def _shard_experts(moe, rank: int, world: int, sequence_parallel: bool) -> None:
     """Expert parallelism for MoE layers inside the model-parallel group:
     experts are sharded across ranks while activations stay replicated, so
@@ -338,5 +374,19 @@ def apply_tensor_parallel(model, rank: int, world: int,
         mlp._tp = True
         attn._sp = mlp._sp = sequence_parallel
 
+    # vocab-parallel lm head (untied models, V % tp == 0, non-SP): the
+    # [B,S,V] logits replica never materializes — the trainer computes
+    # vocab_parallel_cross_entropy on the shard (Megatron-style)
+    model._vp_vocab0 = -1
+    if (hasattr(model, "output") and not sequence_parallel
+            and args.vocab_size % world == 0):
+        V = args.vocab_size
+        lv = V // world
+        model.output.weight = _shard_rows(
+            model.output.weight,
+            torch.arange(rank * lv, (rank + 1) * lv,
+                         device=model.output.weight.device))
+        model.output.out_features = lv
+        model._vp_vocab0 = rank * lv
     model._tp_world = world
     model._sp_world = world if sequence_parallel else 1

@@ -825,3 +825,96 @@ def test_merge_ep_checkpoint_roundtrip(tmp_path):
     assert set(merged.keys()) == set(want.keys())
     for k in want:
         assert torch.equal(merged[k], want[k]), k
+
+
+def _vp_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.parallel.dist import broadcast_module
+        from mlx_cuda_distributed_pretraining_amd.parallel.tp import (
+            apply_tensor_parallel, vocab_parallel_cross_entropy)
+
+        args = _args()
+        args.vocab_size = 64  # % 2 == 0 -> vocab-parallel head activates
+        args.tie_word_embeddings = False
+        torch.manual_seed(0)
+        model = Model(args)
+        broadcast_module(model)
+        apply_tensor_parallel(model, rank, world)
+        assert model._vp_vocab0 == rank * 32
+
+        g = torch.Generator().manual_seed(3)
+        batch = torch.randint(0, 64, (2, 16), generator=g)
+        model.train()
+        logits = model(batch[:, :-1])
+        assert logits.shape[-1] == 32  # sharded
+        loss, ntok = vocab_parallel_cross_entropy(
+            logits.reshape(-1, 32), batch[:, 1:].reshape(-1), model._vp_vocab0)
+        loss.backward()
+        loss = loss.detach()
+        # eval path gathers full logits
+        model.eval()
+        with torch.no_grad():
+            full = model(batch[:, :-1])
+        q.put({
+            "rank": rank,
+            "loss": float(loss),
+            "ntok": int(ntok),
+            "eval_logits": full.numpy().copy(),
+            "grads": {n: p.grad.numpy().copy() for n, p in model.named_parameters()
+                      if p.grad is not None},
+            "sharded": {n: bool(getattr(p, "_tp_sharded", False))
+                        for n, p in model.named_parameters()},
+        })
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_vocab_parallel_head_and_ce_match_single_process():
+    """Vocab-parallel lm head + CE: loss, eval logits, replicated-param
+    grads, and the lm-head shard grads all match the single-process
+    full-vocab run exactly."""
+    args = _args()
+    args.vocab_size = 64
+    args.tie_word_embeddings = False
+    torch.manual_seed(0)
+    model = Model(args)
+    g = torch.Generator().manual_seed(3)
+    batch = torch.randint(0, 64, (2, 16), generator=g)
+    model.train()
+    logits = model(batch[:, :-1])
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, 64), batch[:, 1:].reshape(-1))
+    loss.backward()
+    ref_grads = {n: p.grad.clone() for n, p in model.named_parameters()
+                 if p.grad is not None}
+    model.eval()
+    with torch.no_grad():
+        ref_eval = model(batch[:, :-1])
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_vp_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get(), q.get()]
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    res.sort(key=lambda r: r["rank"])
+
+    for r in res:
+        assert r["loss"] == pytest.approx(float(loss), abs=1e-5)
+        assert torch.allclose(torch.from_numpy(r["eval_logits"]), ref_eval, atol=1e-5)
+        rk = r["rank"]
+        for n, gr in r["grads"].items():
+            gt = torch.from_numpy(gr)
+            full = ref_grads[n]
+            if n == "output.weight":
+                assert torch.allclose(gt, full[rk * 32:(rk + 1) * 32], atol=1e-5), n
+            elif not r["sharded"][n]:
+                assert torch.allclose(gt, full, atol=1e-5), f"replicated grad {n}"

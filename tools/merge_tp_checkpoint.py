@@ -51,6 +51,12 @@ def merge_tp_state_dicts(shards, num_heads: int, num_kv_heads: int, head_dim: in
             out[name] = torch.cat([q, k, v], dim=0)
         elif "wo.weight" in name or "w_down.weight" in name:
             out[name] = torch.cat(parts, dim=1)
+        elif (name == "output.weight"
+              and "tok_embeddings.weight" in shards[0]
+              and parts[0].shape[0] != shards[0]["tok_embeddings.weight"].shape[0]):
+            # vocab-parallel lm head (rows != full vocab -> sharded):
+            # row-concat back to the full vocab
+            out[name] = torch.cat(parts, dim=0)
         elif "w_gate_up" in name:
             g = torch.cat([p[:li] for p in parts], dim=0)
             u = torch.cat([p[li:] for p in parts], dim=0)
