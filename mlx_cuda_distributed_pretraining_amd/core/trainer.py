@@ -529,6 +529,7 @@ class Trainer:
             "validation_losses": self.validation_losses,
             "val_ptr": getattr(self.data_manager, "val_ptr", 0),
             "tp_world": self.tp_world,
+            "vocab_parallel": getattr(self.model, "_vp_vocab0", -1) >= 0,
         }
         if self.dp_rank == 0:
             save_checkpoint(f"{base}_tp{self.tp_rank}", self.model,

@@ -438,7 +438,14 @@ class Model(nn.Module):
     def forward(
         self, tokens: torch.Tensor, cache: Optional[List[KVCache]] = None
     ) -> torch.Tensor:
-        x = self.tok_embeddings(tokens)
+        vp0 = getattr(self, "_vp_vocab0", -1)
+        vp_tied = vp0 >= 0 and self.args.tie_word_embeddings
+        if vp_tied:
+            from ..parallel.tp import vp_embedding
+
+            x = vp_embedding(tokens, self.tok_embeddings.weight, vp0)
+        else:
+            x = self.tok_embeddings(tokens)
         static = (
             cache is not None
             and getattr(cache[0], "static_decode", False)
@@ -476,13 +483,15 @@ class Model(nn.Module):
                 x = self.norm(res)
             else:
                 _, x = add_rms_norm(res, delta, self.norm.weight, self.norm.eps)
-            vp0 = getattr(self, "_vp_vocab0", -1)
             if vp0 >= 0:
                 # vocab-parallel lm head (parallel/tp.py): SHARDED logits in
                 # training (the trainer runs vocab_parallel_cross_entropy);
                 # gathered for eval/generation API compatibility
                 x = copy_to_tp(x)
-                logits = self.output(x)
+                if self.args.tie_word_embeddings:
+                    logits = x @ self.tok_embeddings.weight.t()
+                else:
+                    logits = self.output(x)
                 if not self.training:
                     import torch.distributed as dist_
 

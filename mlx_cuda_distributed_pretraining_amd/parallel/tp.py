@@ -234,6 +234,21 @@ def _localize_row_bias(linear: torch.nn.Linear, rank: int) -> None:
         linear.bias.requires_grad_(False)
 
 
+def vp_embedding(tokens: torch.Tensor, weight: torch.Tensor, v0: int) -> torch.Tensor:
+    """Embedding lookup over a VOCAB-SHARDED weight ([V/tp, H], this rank
+    owning ids [v0, v0 + V/tp)): masked local lookup, zeros elsewhere, then
+    the group sum assembles the full embedding on every rank. Backward is
+    exact: each rank's weight shard receives grads only for its own ids
+    (reduce_from_tp is identity in backward)."""
+    lv = weight.shape[0]
+    local = tokens - v0
+    in_shard = (local >= 0) & (local < lv)
+    safe = local.clamp(0, lv - 1)
+    emb = torch.nn.functional.embedding(safe, weight)
+    emb = emb * in_shard.unsqueeze(-1).to(emb.dtype)
+    return reduce_from_tp(emb)
+
+
 def vocab_parallel_cross_entropy(logits_local: torch.Tensor, targets: torch.Tensor,
                                  v0: int, ignore_index: int = -100):
     """Cross entropy over VOCAB-SHARDED logits ([N, V/tp] on each rank,
@@ -374,19 +389,27 @@ def apply_tensor_parallel(model, rank: int, world: int,
         mlp._tp = True
         attn._sp = mlp._sp = sequence_parallel
 
-    # vocab-parallel lm head (untied models, V % tp == 0, non-SP): the
-    # [B,S,V] logits replica never materializes — the trainer computes
-    # vocab_parallel_cross_entropy on the shard (Megatron-style)
+    # vocab-parallel lm head (V % tp == 0, non-SP): the [B,S,V] logits
+    # replica never materializes — the trainer computes
+    # vocab_parallel_cross_entropy on the shard (Megatron-style).
+    # Tied models shard the EMBEDDING weight (shared with the head); the
+    # lookup becomes masked-local + group sum (vp_embedding).
     model._vp_vocab0 = -1
-    if (hasattr(model, "output") and not sequence_parallel
-            and args.vocab_size % world == 0):
+    if not sequence_parallel and args.vocab_size % world == 0:
         V = args.vocab_size
         lv = V // world
-        model.output.weight = _shard_rows(
-            model.output.weight,
-            torch.arange(rank * lv, (rank + 1) * lv,
-                         device=model.output.weight.device))
-        model.output.out_features = lv
-        model._vp_vocab0 = rank * lv
+        rows = None  # torch.arange built per tensor for device correctness
+        if hasattr(model, "output"):
+            w = model.output.weight
+            model.output.weight = _shard_rows(
+                w, torch.arange(rank * lv, (rank + 1) * lv, device=w.device))
+            model.output.out_features = lv
+            model._vp_vocab0 = rank * lv
+        elif args.tie_word_embeddings:
+            w = model.tok_embeddings.weight
+            model.tok_embeddings.weight = _shard_rows(
+                w, torch.arange(rank * lv, (rank + 1) * lv, device=w.device))
+            model.tok_embeddings.num_embeddings = lv
+            model._vp_vocab0 = rank * lv
     model._tp_world = world
     model._sp_world = world if sequence_parallel else 1

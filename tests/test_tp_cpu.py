@@ -918,3 +918,88 @@ def test_vocab_parallel_head_and_ce_match_single_process():
                 assert torch.allclose(gt, full[rk * 32:(rk + 1) * 32], atol=1e-5), n
             elif not r["sharded"][n]:
                 assert torch.allclose(gt, full, atol=1e-5), f"replicated grad {n}"
+
+
+def _vp_tied_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.parallel.dist import broadcast_module
+        from mlx_cuda_distributed_pretraining_amd.parallel.tp import (
+            apply_tensor_parallel, vocab_parallel_cross_entropy)
+
+        args = _args()
+        args.vocab_size = 64
+        args.tie_word_embeddings = True  # embedding IS the head -> sharded
+        torch.manual_seed(0)
+        model = Model(args)
+        broadcast_module(model)
+        apply_tensor_parallel(model, rank, world)
+        assert model._vp_vocab0 == rank * 32
+        assert model.tok_embeddings.weight.shape[0] == 32
+
+        g = torch.Generator().manual_seed(3)
+        batch = torch.randint(0, 64, (2, 16), generator=g)
+        model.train()
+        logits = model(batch[:, :-1])
+        assert logits.shape[-1] == 32
+        loss, _ = vocab_parallel_cross_entropy(
+            logits.reshape(-1, 32), batch[:, 1:].reshape(-1), model._vp_vocab0)
+        loss.backward()
+        model.eval()
+        with torch.no_grad():
+            full = model(batch[:, :-1])
+        q.put({
+            "rank": rank,
+            "loss": float(loss.detach()),
+            "eval_logits": full.numpy().copy(),
+            "emb_grad": model.tok_embeddings.weight.grad.numpy().copy(),
+            "norm_grad": model.norm.weight.grad.numpy().copy(),
+        })
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_vocab_parallel_tied_embedding_matches_single_process():
+    """Tied-embedding VP (the DEFAULT model family): masked vp_embedding
+    lookup + sharded tied head reproduce the single-process loss, eval
+    logits, embedding-shard grads (lookup + head paths summed) and
+    replicated norm grads exactly."""
+    args = _args()
+    args.vocab_size = 64
+    args.tie_word_embeddings = True
+    torch.manual_seed(0)
+    model = Model(args)
+    g = torch.Generator().manual_seed(3)
+    batch = torch.randint(0, 64, (2, 16), generator=g)
+    model.train()
+    logits = model(batch[:, :-1])
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, 64), batch[:, 1:].reshape(-1))
+    loss.backward()
+    emb_grad = model.tok_embeddings.weight.grad.clone()
+    norm_grad = model.norm.weight.grad.clone()
+    model.eval()
+    with torch.no_grad():
+        ref_eval = model(batch[:, :-1])
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_vp_tied_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get(), q.get()]
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    res.sort(key=lambda r: r["rank"])
+    for r in res:
+        rk = r["rank"]
+        assert r["loss"] == pytest.approx(float(loss), abs=1e-5)
+        assert torch.allclose(torch.from_numpy(r["eval_logits"]), ref_eval, atol=1e-5)
+        assert torch.allclose(torch.from_numpy(r["emb_grad"]),
+                              emb_grad[rk * 32:(rk + 1) * 32], atol=1e-5)
+        assert torch.allclose(torch.from_numpy(r["norm_grad"]), norm_grad, atol=1e-5)

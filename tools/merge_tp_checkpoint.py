@@ -32,8 +32,10 @@ sys.path.insert(0, str(REPO))
 
 
 def merge_tp_state_dicts(shards, num_heads: int, num_kv_heads: int, head_dim: int,
-                         intermediate_size: int) -> Dict[str, torch.Tensor]:
-    """shards: list of state dicts ordered by tp_rank."""
+                         intermediate_size: int,
+                         vocab_parallel: bool = False) -> Dict[str, torch.Tensor]:
+    """shards: list of state dicts ordered by tp_rank. vocab_parallel: the
+    lm head (and, for tied models, the embedding) rows are vocab shards."""
     world = len(shards)
     lq, lkv = num_heads // world * head_dim, num_kv_heads // world * head_dim
     li = intermediate_size // world
@@ -51,12 +53,14 @@ def merge_tp_state_dicts(shards, num_heads: int, num_kv_heads: int, head_dim: in
             out[name] = torch.cat([q, k, v], dim=0)
         elif "wo.weight" in name or "w_down.weight" in name:
             out[name] = torch.cat(parts, dim=1)
-        elif (name == "output.weight"
-              and "tok_embeddings.weight" in shards[0]
-              and parts[0].shape[0] != shards[0]["tok_embeddings.weight"].shape[0]):
-            # vocab-parallel lm head (rows != full vocab -> sharded):
-            # row-concat back to the full vocab
-            out[name] = torch.cat(parts, dim=0)
+        elif vocab_parallel and name in ("output.weight", "tok_embeddings.weight"):
+            # vocab-parallel head / tied embedding: row-concat back to the
+            # full vocab. Untied replicated embeddings are caught below: when
+            # output.weight exists the embedding was NOT sharded.
+            if name == "tok_embeddings.weight" and "output.weight" in shards[0]:
+                out[name] = t0  # untied: embedding stayed replicated
+            else:
+                out[name] = torch.cat(parts, dim=0)
         elif "w_gate_up" in name:
             g = torch.cat([p[:li] for p in parts], dim=0)
             u = torch.cat([p[li:] for p in parts], dim=0)
@@ -73,6 +77,7 @@ def merge_checkpoint(base: str) -> str:
 
     state = json.loads(Path(f"{base}_state.json").read_text())
     world = int(state.get("tp_world", 0))
+    vp = bool(state.get("vocab_parallel", False))
     if world < 2:
         raise ValueError(f"{base}_state.json has no tp_world — not a TP checkpoint")
     shards = [load_file(f"{base}_tp{r}_model.safetensors") for r in range(world)]
@@ -83,10 +88,13 @@ def merge_checkpoint(base: str) -> str:
     run_dir = Path(base).parent.parent
     cfg = Config.from_yaml(run_dir / "config.yaml")
     vocab = shards[0]["tok_embeddings.weight"].shape[0]
+    if vp and "output.weight" not in shards[0]:
+        vocab *= world  # tied sharded embedding: shard rows = V / tp
     args = ModelArgs.from_config(cfg.model, vocab)
 
     merged = merge_tp_state_dicts(shards, args.num_heads, args.num_kv_heads,
-                                  args.head_dim, args.intermediate_size)
+                                  args.head_dim, args.intermediate_size,
+                                  vocab_parallel=vp)
     save_file(merged, f"{base}_model.safetensors", metadata={"format": "pt"})
     state.pop("tp_world", None)
     Path(f"{base}_state.json").write_text(json.dumps(state))
