@@ -908,7 +908,7 @@ def test_vocab_parallel_head_and_ce_match_single_process():
     res.sort(key=lambda r: r["rank"])
 
     for r in res:
-        assert r["loss"] == pytest.approx(float(loss), abs=1e-5)
+        assert r["loss"] == pytest.approx(float(loss.detach()), abs=1e-5)
         assert torch.allclose(torch.from_numpy(r["eval_logits"]), ref_eval, atol=1e-5)
         rk = r["rank"]
         for n, gr in r["grads"].items():
@@ -998,7 +998,7 @@ def test_vocab_parallel_tied_embedding_matches_single_process():
     res.sort(key=lambda r: r["rank"])
     for r in res:
         rk = r["rank"]
-        assert r["loss"] == pytest.approx(float(loss), abs=1e-5)
+        assert r["loss"] == pytest.approx(float(loss.detach()), abs=1e-5)
         assert torch.allclose(torch.from_numpy(r["eval_logits"]), ref_eval, atol=1e-5)
         assert torch.allclose(torch.from_numpy(r["emb_grad"]),
                               emb_grad[rk * 32:(rk + 1) * 32], atol=1e-5)
