@@ -1003,3 +1003,51 @@ def test_vocab_parallel_tied_embedding_matches_single_process():
         assert torch.allclose(torch.from_numpy(r["emb_grad"]),
                               emb_grad[rk * 32:(rk + 1) * 32], atol=1e-5)
         assert torch.allclose(torch.from_numpy(r["norm_grad"]), norm_grad, atol=1e-5)
+
+
+def _tp_gen_worker(rank, world, port, q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from mlx_cuda_distributed_pretraining_amd.inference.generate import generate_step
+        from mlx_cuda_distributed_pretraining_amd.parallel.dist import broadcast_module
+        from mlx_cuda_distributed_pretraining_amd.parallel.tp import apply_tensor_parallel
+
+        args = _args()
+        args.vocab_size = 64
+        torch.manual_seed(0)
+        model = Model(args).eval()
+        broadcast_module(model)
+        apply_tensor_parallel(model, rank, world)
+        toks = list(generate_step(model, [5, 11, 3], max_tokens=16))
+        q.put({"rank": rank, "tokens": toks})
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp_generation_matches_single_process():
+    """Greedy generation through the KV-cache path on a TP=2 model (with the
+    vocab-parallel head's eval gather) is token-identical to the
+    single-process model."""
+    args = _args()
+    args.vocab_size = 64
+    torch.manual_seed(0)
+    from mlx_cuda_distributed_pretraining_amd.inference.generate import generate_step
+
+    model = Model(args).eval()
+    want = list(generate_step(model, [5, 11, 3], max_tokens=16))
+
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_tp_gen_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    res = [q.get(), q.get()]
+    for p in procs:
+        p.join(240)
+        assert p.exitcode == 0
+    for r in res:
+        assert r["tokens"] == want, (r["rank"], r["tokens"], want)
