@@ -25,11 +25,19 @@ Mechanics
   Sharded parameters get ``p._tp_sharded = True`` so the gradient-norm
   reduction can count replicated params once (parallel/dist.py consumers).
 
-Scope (round 1): DPxTP meshes via ``init_tp_mesh`` (TP groups = adjacent
-ranks for xGMI locality; DP groups stride across replicas; gradient
-all-reduce rides the DP group only), TP checkpointing not yet implemented,
-MoE + TP rejected. CPU-tested with 2-rank TP and a 4-rank 2x2 mesh on gloo
-(tests/test_tp_cpu.py); the collective pattern is backend-agnostic.
+Scope (round 1, all CPU-verified EXACT against single-process runs —
+tests/test_tp_cpu.py; the collective pattern is backend-agnostic):
+  * DPxTP meshes via ``init_tp_mesh`` (TP groups = adjacent ranks for xGMI
+    locality; DP groups stride across replicas; gradient all-reduce rides
+    the DP group only),
+  * sequence parallelism (``gather_sp``/``scatter_sp``: S-sharded
+    inter-sublayer activations),
+  * expert parallelism for MoE (``_shard_experts``: replicated activations,
+    sharded experts, router grads group-summed),
+  * vocab-parallel lm head + CE (``vocab_parallel_cross_entropy`` /
+    ``vp_embedding`` for tied models — no [B,S,V] logits replica),
+  * per-shard TP checkpointing (trainer) + merge tool
+    (tools/merge_tp_checkpoint.py).
 """
 from __future__ import annotations
 

@@ -1051,3 +1051,21 @@ def test_tp_generation_matches_single_process():
         assert p.exitcode == 0
     for r in res:
         assert r["tokens"] == want, (r["rank"], r["tokens"], want)
+
+
+def test_tp_validation_errors():
+    """Divisibility violations fail fast with clear errors."""
+    args = _args()          # 4 heads, 2 kv heads
+    torch.manual_seed(0)
+    from mlx_cuda_distributed_pretraining_amd.parallel.tp import apply_tensor_parallel
+
+    m = Model(args)
+    with pytest.raises(ValueError, match="must divide num_heads"):
+        apply_tensor_parallel(m, 0, 3)
+
+    args2 = _args()
+    args2.num_local_experts = 3
+    args2.num_experts_per_tok = 1
+    m2 = Model(args2)
+    with pytest.raises(ValueError, match="num_local_experts"):
+        apply_tensor_parallel(m2, 0, 2)
