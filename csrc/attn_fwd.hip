@@ -48,7 +48,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   constexpr int VROW = 64;                    // V^T image row (elements; swzt())
   constexpr int TILE = KVB * D + D * VROW;    // elements per buffer
 
-  __shared__ __hip_bfloat16 smem[2 * TILE];
+  // 3-slot ring: with PV lagging QK^T by one tile (cross-tile pipeline),
+  // iteration j reads K(j) from slot j%3, V(j-1) from slot (j-1)%3 and
+  // writes tile j+1 into slot (j+1)%3 — all distinct, still ONE barrier
+  // per tile (slot j%3's readers both finish before the end-of-(j+1) barrier
+  // that precedes its reuse at j+3).
+  __shared__ __hip_bfloat16 smem[3 * TILE];
 
   const int b = blockIdx.z;
   const int hq = blockIdx.y;
@@ -167,18 +172,32 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[dc][r] = 0.f;
 
-  // prologue: stage tile 0 into buf 0
+  // prologue: stage tile 0 into slot 0
   stage_load(kv_lo);
   stage_write(0);
   __syncthreads();
 
-  int buf = 0;
+  // Cross-tile software pipeline: PV lags QK^T by one tile, so tile j's
+  // exp2/psum/pack VALU is register-independent of tile j-1's 16 PV MFMAs
+  // and the scheduler interleaves them (today's serial order exposes the
+  // whole softmax between the two MFMA clusters). T13 ordering is kept:
+  // the o/l rescale for tile j's max applies AFTER PV(j-1) completes, so
+  // P_{j-1} enters o at its own scale exactly once.
+  bf16x8 pa[KT][2];  // A-fragments of the PREVIOUS tile's P (tile -1: zeros)
+#pragma unroll
+  for (int kt = 0; kt < KT; ++kt) {
+    pa[kt][0] = bf16x8{};
+    pa[kt][1] = bf16x8{};
+  }
+  // pv_slot starts at slot 0 (= tile 0's finite data): pa==0 keeps the
+  // dummy PV contribution exactly 0 without reading uninitialized LDS.
+  int buf = 0, pv_slot = 0, nslot = 1;
   for (int kv0 = kv_lo; kv0 < kv_hi; kv0 += KVB) {
     const bool has_next = kv0 + KVB < kv_hi;
     if (has_next) stage_load(kv0 + KVB);  // loads in flight under the MFMAs
 
     const __hip_bfloat16* k_lds = smem + buf * TILE;
-    const __hip_bfloat16* vt_lds = k_lds + KVB * D;
+    const __hip_bfloat16* vt_prev = smem + pv_slot * TILE + KVB * D;
 
     // ---- S^T = mfma(K, Q) per 32-row k sub-tile ----
     f32x16 st[KT];
@@ -198,7 +217,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- scale + mask + online softmax (base-2 domain) ----
+    // ---- tile max (tree reduce, raw units on the full fast path) ----
     // interior tiles (wave-uniform): every (q,k) pair of this wave is kept
     bool full = kv0 + KVB <= Skv;
     if constexpr (MOD == MOD_CAUSAL) {
@@ -212,17 +231,22 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
       full = false;  // slope term needs per-element positions anyway
     }
 
-    float p[KT][16];
-    float tmax = -INFINITY;
+    float tmax;
     if (full) {
+      // exp2+fma fold: scale2 is folded into the exp argument below, so the
+      // 32-element scale pass disappears; the max runs on raw scores
+      // (scale2 > 0 keeps it monotone) as a depth-5 tree, not a 31-op chain.
+      float tm[8];
 #pragma unroll
-      for (int kt = 0; kt < KT; ++kt)
+      for (int i = 0; i < 8; ++i)
+        tm[i] = fmaxf(fmaxf(st[0][i], st[0][i + 8]),
+                      KT > 1 ? fmaxf(st[1][i], st[1][i + 8]) : -INFINITY);
 #pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          p[kt][reg] = st[kt][reg] * scale2;
-          tmax = fmaxf(tmax, p[kt][reg]);
-        }
+      for (int i = 0; i < 4; ++i) tm[i] = fmaxf(tm[i], tm[i + 4]);
+      tmax = fmaxf(fmaxf(tm[0], tm[1]), fmaxf(tm[2], tm[3])) * scale2;
     } else {
+      // masked path: materialize base-2 scores in place of st
+      tmax = -INFINITY;
 #pragma unroll
       for (int kt = 0; kt < KT; ++kt)
 #pragma unroll
@@ -231,8 +255,8 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
           const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
           float s = st[kt][reg] * scale2;
           if constexpr (MOD == MOD_ALIBI) s += slope2 * (k_pos - q_pos);
-          p[kt][reg] = keep ? s : -INFINITY;
-          tmax = fmaxf(tmax, p[kt][reg]);
+          st[kt][reg] = keep ? s : -INFINITY;
+          tmax = fmaxf(tmax, st[kt][reg]);
         }
     }
     tmax = cross32_max(tmax);
@@ -240,37 +264,15 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     // max by more than THR (base-2), keep the old max and SKIP the O/l
     // rescale entirely — P is then bounded by 2^THR, which the fp32
     // accumulators tolerate (~3x max-abs error vs THR=0). The decision is
-    // wave-uniform and taken before any of this tile's P*V.
+    // wave-uniform; the o_acc rescale it gates runs after PV(j-1) below.
     const float m_old = m;
     const bool defer = __all(tmax - m <= 8.0f);
     if (!defer) m = fmaxf(m, tmax);
     const float mc = fmaxf(m, -1e30f);
-    float psum = 0.f;
-#pragma unroll
-    for (int kt = 0; kt < KT; ++kt)
-#pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        p[kt][reg] = __builtin_amdgcn_exp2f(p[kt][reg] - mc);
-        psum += p[kt][reg];
-      }
-    psum = cross32_sum(psum);
-    if (defer) {
-      l += psum;
-    } else {
-      const float alpha = __builtin_amdgcn_exp2f(m_old - mc);
-      l = l * alpha + psum;
-#pragma unroll
-      for (int reg = 0; reg < 16; ++reg) {
-        const float ar = __shfl(alpha, acc_row(reg, hi));
-#pragma unroll
-        for (int dc = 0; dc < DCOL; ++dc) o_acc[dc][reg] *= ar;
-      }
-    }
 
-    // ---- P fragments + PV ----
-    bf16x8 pa[KT][2];
-#pragma unroll
-    for (int kt = 0; kt < KT; ++kt) acc_to_afrag(p[kt], pa[kt][0], pa[kt][1]);
+    // ---- PV(j-1): accumulates P_{j-1}*V at scale m_{j-1} (o_acc still at
+    // that scale). The exp/psum/pack block below is independent of o_acc
+    // and interleaves into these MFMAs' issue gaps. ----
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dc = 0; dc < DCOL; ++dc) {
@@ -284,7 +286,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
           const int drow = dc * 32 + lq;
           Bf16x8U vfr;
           *reinterpret_cast<uint4*>(vfr.s) = *reinterpret_cast<const uint4*>(
-              vt_lds + drow * VROW + ((kt * 32 + ks * 16 + hi * 8) ^ swzt(drow)));
+              vt_prev + drow * VROW + ((kt * 32 + ks * 16 + hi * 8) ^ swzt(drow)));
           acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kt][ks], vfr.v, acc, 0, 0, 0);
         }
 #pragma unroll
@@ -292,14 +294,76 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // write tile t+1 into the other buffer LAST: the vmcnt wait on
-    // stage_load's global loads has the whole iteration's MFMA/softmax work
-    // to hide under (T14/async-STAGE). buf^1 was last read in iteration
-    // t-1, before the barrier every wave has already passed.
-    if (has_next) stage_write(buf ^ 1);
+    // ---- exp (base-2) + row sum + pack for the NEXT iteration's PV ----
+    float p[KT][16];
+    float psum = 0.f;
+    if (full) {
+#pragma unroll
+      for (int kt = 0; kt < KT; ++kt)
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          p[kt][reg] = __builtin_amdgcn_exp2f(fmaf(st[kt][reg], scale2, -mc));
+          psum += p[kt][reg];
+        }
+    } else {
+#pragma unroll
+      for (int kt = 0; kt < KT; ++kt)
+#pragma unroll
+        for (int reg = 0; reg < 16; ++reg) {
+          p[kt][reg] = __builtin_amdgcn_exp2f(st[kt][reg] - mc);
+          psum += p[kt][reg];
+        }
+    }
+    psum = cross32_sum(psum);
+    if (defer) {
+      l += psum;
+    } else {
+      const float alpha = __builtin_amdgcn_exp2f(m_old - mc);
+      l = l * alpha + psum;
+      // o_acc rescale: strictly after PV(j-1), before PV(j) (T13 hazard)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const float ar = __shfl(alpha, acc_row(reg, hi));
+#pragma unroll
+        for (int dc = 0; dc < DCOL; ++dc) o_acc[dc][reg] *= ar;
+      }
+    }
+#pragma unroll
+    for (int kt = 0; kt < KT; ++kt) acc_to_afrag(p[kt], pa[kt][0], pa[kt][1]);
 
-    __syncthreads();  // buf consumed by all waves; buf^1 fully written
-    buf ^= 1;
+    // write tile t+1 LAST: the vmcnt wait on stage_load's global loads has
+    // the whole iteration's MFMA/softmax work to hide under (T14).
+    if (has_next) stage_write(nslot);
+
+    __syncthreads();
+    pv_slot = buf;
+    buf = nslot;
+    nslot = (nslot == 2) ? 0 : nslot + 1;
+  }
+
+  // ---- drain: PV of the last tile ----
+  {
+    const __hip_bfloat16* vt_prev = smem + pv_slot * TILE + KVB * D;
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dc = 0; dc < DCOL; ++dc) {
+      f32x16 acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[r] = o_acc[dc][r];
+#pragma unroll
+      for (int kt = 0; kt < KT; ++kt)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          const int drow = dc * 32 + lq;
+          Bf16x8U vfr;
+          *reinterpret_cast<uint4*>(vfr.s) = *reinterpret_cast<const uint4*>(
+              vt_prev + drow * VROW + ((kt * 32 + ks * 16 + hi * 8) ^ swzt(drow)));
+          acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[kt][ks], vfr.v, acc, 0, 0, 0);
+        }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) o_acc[dc][r] = acc[r];
+    }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- epilogue ----

@@ -224,8 +224,10 @@ class Trainer:
             rank=self.dp_rank, world_size=self.dp_world, seed=cfg.system.seed,
         )
         if cfg.training.epochs is not None and self.data_manager.num_batches:
+            # Data is sharded over the DP group only (TP replicas share
+            # batches), so epochs divide by dp_world, not world_size.
             self.steps_per_epoch = max(
-                self.data_manager.num_batches // self.world_size, 1
+                self.data_manager.num_batches // self.dp_world, 1
             )
             self.total_steps = self.steps_per_epoch * cfg.training.epochs
         else:
@@ -684,6 +686,14 @@ class Trainer:
                 reset_training_state=cfg.resume.reset_training_state,
             )
             self.logger.log(f"Resumed from {cfg.resume.checkpoint} at step {self.start_step}")
+            if self.start_step > 0:
+                skipped = self.data_manager.fast_forward(
+                    self.start_step * self.grad_accum_steps
+                )
+                if skipped:
+                    self.logger.log(
+                        f"Streaming source fast-forwarded past {skipped} batches"
+                    )
 
         if (cfg.training.lr_finder or {}).get("enabled"):
             self.run_learning_rate_finder()

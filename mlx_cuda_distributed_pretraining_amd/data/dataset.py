@@ -137,6 +137,24 @@ class DataManager:
             batch[j, : len(r)] = r
         return torch.from_numpy(batch)
 
+    def fast_forward(self, num_batches: int) -> int:
+        """Advance a streaming source past ``num_batches`` already-consumed
+        batches (auto-resume: the stream otherwise replays from shard 0 and
+        re-trains on seen data; tokens_emitted is re-advanced as a side
+        effect, so the max_tokens budget resumes correctly too). No-op for
+        the in-memory path, whose generate_batch is indexed by step.
+        Returns the number of batches actually skipped."""
+        if getattr(self, "stream", None) is None or num_batches <= 0:
+            return 0
+        skipped = 0
+        for _ in range(num_batches):
+            try:
+                next(self.stream)
+                skipped += 1
+            except StopIteration:
+                break
+        return skipped
+
     def generate_batch(self, step: int) -> torch.Tensor:
         if self.synthetic:
             g = torch.Generator().manual_seed(self.seed * 1_000_003 + step * self.world_size + self.rank)

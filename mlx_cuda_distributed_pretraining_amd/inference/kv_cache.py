@@ -34,6 +34,12 @@ class ChunkedKVCache(KVCache):
     def offset(self) -> int:
         return self._len
 
+    def trim(self, n: int) -> None:
+        # Base-class trim slices capacity; here only the logical length moves
+        # (later updates overwrite the trimmed rows in the preallocated slab).
+        if n > 0:
+            self._len = max(0, self._len - n)
+
     def update(self, k: torch.Tensor, v: torch.Tensor):
         B, S, H, D = k.shape
         need = self._len + S
@@ -92,6 +98,18 @@ class QuantizedKVCache(KVCache):
     @property
     def offset(self) -> int:
         return self._len
+
+    def trim(self, n: int) -> None:
+        # Base-class trim is a no-op here (self.k is None); slice the
+        # quantized storage instead so speculative-decoding rewind works.
+        if n <= 0 or self._len == 0:
+            return
+        keep = max(0, self._len - n)
+        for name in ("_ck", "_cv", "_sk", "_sv", "_zk", "_zv"):
+            t = getattr(self, name)
+            if t is not None:
+                setattr(self, name, t[:, :keep])
+        self._len = keep
 
     def _append(self, codes, scale, zero, which: str):
         cn, sn, zn = f"_c{which}", f"_s{which}", f"_z{which}"

@@ -193,7 +193,8 @@ def flash_attention(
     prefix_len: Optional[int] = None,
     alibi_slopes: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
-    """Tiled attention, [B,Hq,S,D] x [B,Hkv,S,D] -> [B,Hq,S,D]."""
+    """Tiled attention, BSHD layout: [B,S,Hq,D] x [B,S,Hkv,D] -> [B,S,Hq,D]
+    (q.shape[2] is the HEAD count — _norm_alibi relies on that)."""
     alibi_slopes = _norm_alibi(alibi_slopes, q.shape[2], q.device)
     return _FlashAttnFn.apply(q, k, v, causal, scale, window, prefix_len, alibi_slopes)
 

@@ -101,3 +101,38 @@ def test_generate_step_with_quantized_kv():
     out2 = list(generate_step(model, [1, 5, 9], max_tokens=8))
     # greedy int8-KV decode should match the fp path on a tiny model
     assert out == out2
+
+
+def test_chunked_trim_rewind_matches_simple():
+    """inherited trim sliced capacity without moving _len — later
+    updates then wrote at stale positions (speculative decoding rewind)."""
+    torch.manual_seed(1)
+    simple, chunked = KVCache(), ChunkedKVCache(chunk=8)
+    k0, v0 = torch.randn(1, 6, 2, 16), torch.randn(1, 6, 2, 16)
+    simple.update(k0, v0)
+    chunked.update(k0, v0)
+    simple.trim(2)
+    chunked.trim(2)
+    assert chunked.offset == simple.offset == 4
+    k1, v1 = torch.randn(1, 3, 2, 16), torch.randn(1, 3, 2, 16)
+    ks, vs = simple.update(k1, v1)
+    kc, vc = chunked.update(k1, v1)
+    assert torch.equal(ks, kc) and torch.equal(vs, vc)
+    assert chunked.offset == simple.offset == 7
+
+
+def test_quantized_trim_rewind():
+    """inherited trim was a silent no-op (self.k is None)."""
+    torch.manual_seed(2)
+    q = QuantizedKVCache(bits=8, group=64)
+    k = torch.randn(1, 6, 2, 128)
+    v = torch.randn(1, 6, 2, 128)
+    q.update(k, v)
+    q.trim(2)
+    assert q.offset == 4
+    k1, v1 = torch.randn(1, 1, 2, 128), torch.randn(1, 1, 2, 128)
+    kd, vd = q.update(k1, v1)
+    assert kd.shape[1] == 5 and vd.shape[1] == 5
+    # rewound rows really were dropped: position 4 is the NEW row
+    assert torch.allclose(kd[:, 4].float(), k1[:, 0].float(), atol=0.02)
+    assert torch.allclose(kd[:, :4].float(), k[:, :4].float(), atol=0.02)

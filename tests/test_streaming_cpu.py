@@ -128,3 +128,29 @@ def test_streaming_with_disk_manager(tmp_path):
     blocks = list(ds.iter_token_blocks())
     assert len(blocks) == 3
     assert mgr.used_bytes() > 0  # shards were cached
+
+
+def test_data_manager_fast_forward_resumes_stream(tmp_path):
+    """auto-resume replayed the stream from shard 0; fast_forward
+    skips the already-consumed batches so resumed runs see fresh data."""
+    from mlx_cuda_distributed_pretraining_amd.data.dataset import DataManager
+
+    d = _make_shards(tmp_path, n_shards=4, docs_per_shard=8)
+
+    def make_dm():
+        cfg = type("D", (), {})()
+        cfg.preprocessing = {"max_context_size": 16, "chunk_overlap": 0}
+        cfg.input_file = None
+        cfg.validation_file = None
+        cfg.tokenizer = {}
+        cfg.synthetic = None
+        cfg.streaming = {"source": str(d)}
+        return DataManager(cfg, ByteTok(), batch_size=2)
+
+    full = make_dm()
+    consumed = [full.generate_batch(i) for i in range(4)]
+
+    resumed = make_dm()
+    assert resumed.fast_forward(2) == 2
+    b2 = resumed.generate_batch(0)
+    assert torch.equal(b2, consumed[2]), "fast_forward did not skip consumed batches"
