@@ -109,7 +109,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
 
   __shared__ __align__(16) __hip_bfloat16 smem[2 * TILE];
 
-  const int b = blockIdx.z, hq = blockIdx.y, qtile = blockIdx.x;
+  const int b = blockIdx.z, hq = blockIdx.y, qtile = balance_x();
   const int hkv = hq / (Hq / Hkv);
   const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
   const int lq = lane & 31, hi = lane >> 5;
@@ -327,7 +327,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   __shared__ __align__(16) __hip_bfloat16 smem[2 * TILE];
   __shared__ float stats_lds[2][2][64];  // [buf][L|D][s*32 + qrow]
 
-  const int b = blockIdx.z, hkv = blockIdx.y, kvtile = blockIdx.x;
+  const int b = blockIdx.z, hkv = blockIdx.y, kvtile = balance_x();
   const int group = Hq / Hkv;
   const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
   const int lk = lane & 31, hi = lane >> 5;

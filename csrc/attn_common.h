@@ -162,6 +162,25 @@ __device__ __forceinline__ int swzt(int drow) {
   return (((drow & 7) ^ ((drow >> 3) & 7)) << 3);
 }
 
+// Causal block-skip makes per-block work a linear function of the x-tile
+// index (q-tile: 1..Sq/KVB live kv tiles; kv-tile in the dKV kernel:
+// mirrored). These kernels run ~1 block/CU, so the dispatcher refills CUs
+// at stride gridDim.x*gridDim.y*gridDim.z / 256-ish = exactly 256 apart in
+// linear id; with gridDim.x | 256 every refill hands a CU the SAME
+// blockIdx.x — at S=2048 the all-late-tile CUs do 8x the mean work and set
+// the wall clock. Rotating x by the (y,z)-group index (constant within a
+// group -> bijective per group) makes each CU's successive blocks cycle
+// through the x values instead.
+__device__ __forceinline__ int balance_x() {
+  const int gx = gridDim.x;
+  int x = blockIdx.x;
+  if ((256 % gx) == 0) {
+    const int gid = blockIdx.y + gridDim.y * blockIdx.z;
+    x = (x + gid / (256 / gx)) % gx;
+  }
+  return x;
+}
+
 // keep/mask decision for one score element. q_pos/k_pos are ABSOLUTE
 // positions (q_pos = q_row + Skv - Sq handles KV-cache decode).
 template <int MOD>

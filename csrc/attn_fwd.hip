@@ -57,7 +57,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
 
   const int b = blockIdx.z;
   const int hq = blockIdx.y;
-  const int qtile = blockIdx.x;
+  const int qtile = balance_x();
   const int hkv = hq / (Hq / Hkv);
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;

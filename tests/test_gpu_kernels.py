@@ -204,6 +204,31 @@ def test_attn_fwd_gpu(ext, case):
     assert err < 3e-2, f"attn fwd max err {err} for case {case}"
 
 
+def test_attn_fwd_rescale_branch_forced(ext):
+    """Guide rule 26: the defer-max rescale branch is rare on random data —
+    force it (spike K rows at a late tile so the running max jumps >> THR=8
+    base-2 after earlier tiles accumulated) and check against the fp32
+    oracle. Exercises the pipelined o/l rescale ordering (T13 hazard)."""
+    from mlx_cuda_distributed_pretraining_amd.ops import attention_ref, flash_attention
+
+    B, S, Hq, Hkv, D = 1, 512, 2, 2, 128
+    torch.manual_seed(3)
+    q = torch.randn(B, S, Hq, D, device=dev(), dtype=torch.bfloat16) * 0.3
+    k = torch.randn(B, S, Hkv, D, device=dev(), dtype=torch.bfloat16) * 0.3
+    v = torch.randn(B, S, Hkv, D, device=dev(), dtype=torch.bfloat16)
+    # every q row gets a huge score against k rows 400..403 (tile 6 of 8):
+    # k_spike = 40 * mean(q) makes q.k ~ 40*|q|^2/S >> previous maxima
+    u = q.float().mean(dim=1, keepdim=True)  # [B,1,Hq,D]
+    u = (40.0 * u / u.norm(dim=-1, keepdim=True)).to(torch.bfloat16)
+    for r in range(400, 404):
+        k[:, r] = u[:, 0]
+    o = flash_attention(q, k, v, causal=True)
+    ref = attention_ref(q.float(), k.float(), v.float(), causal=True,
+                        scale=1.0 / math.sqrt(D))
+    err = (o.float() - ref).abs().max().item()
+    assert err < 3e-2, f"forced-rescale max err {err}"
+
+
 def test_attn_fwd_lse_matches(ext):
     from mlx_cuda_distributed_pretraining_amd.ops import attention_ref
     from mlx_cuda_distributed_pretraining_amd.ops._ext import get_ext

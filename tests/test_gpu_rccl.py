@@ -33,7 +33,11 @@ def test_rccl_backend_world1_ddp_zero1_step():
     os.environ.setdefault("LOCAL_RANK", "0")
     created = not dist.is_initialized()
     if created:
-        init_distributed(backend="nccl")
+        # init_distributed intentionally skips group creation at world 1;
+        # here the point IS to bring up the RCCL backend, so init directly.
+        torch.cuda.set_device(0)
+        dist.init_process_group(backend="nccl", rank=0, world_size=1)
+    assert init_distributed(backend="nccl") == (0, 1, 0)  # idempotent path
     try:
         assert dist.get_backend() in ("nccl", "cclx")  # RCCL on ROCm
         torch.manual_seed(0)
