@@ -30,6 +30,11 @@ void lion_step(at::Tensor param, at::Tensor master, at::Tensor grad, at::Tensor 
 void sgd_step(at::Tensor param, at::Tensor master, at::Tensor grad, at::Tensor buf,
               at::Tensor sumsq_t, double lr, double mom, double wd, long decay_boundary,
               bool nesterov, double max_norm);
+// muon.hip (K8 Newton-Schulz GEMM chain)
+void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
+                  double beta, at::Tensor E);
+void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a);
+
 // attn_fwd.hip / attn_bwd.hip
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale,
                                  long mod, long modarg, at::Tensor slopes);
@@ -85,6 +90,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step", &adamw_step, "fused AdamW over flat buffers");
   m.def("lion_step", &lion_step, "fused Lion over flat buffers");
   m.def("sgd_step", &sgd_step, "fused SGD over flat buffers");
+  m.def("muon_gemm_nt", &muon_gemm_nt, "C = alpha*X@Y^T + beta*E (bf16 MFMA)");
+  m.def("muon_gemm_nn_ax", &muon_gemm_nn_ax, "C = Bm@X + a*X (bf16 MFMA)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (o, lse)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward (dq, dk, dv)");
   m.def("attn_bwd_out", &attn_bwd_out,

@@ -1,0 +1,314 @@
+// K8: Muon Newton-Schulz orthogonalization — hand-written gfx950 MFMA GEMM
+// chain (reference algorithm
+// /root/reference/optimizers/muon.py:54-83).
+//
+// The NS-5 iteration per step is   A = X Xᵀ;  B = b·A + c·A·Aᵀ;  X = a·X + B·X
+// (A symmetric, so A² = A·Aᵀ).  Two kernels cover the chain:
+//   muon_gemm_nt   : C[M,N] = alpha·X[M,K]·Y[N,K]ᵀ + beta·E[M,N]   (NT form —
+//                    both operands row-major with contiguous K, staged by
+//                    global_load_lds with a source-side XOR swizzle; the
+//                    quintic combine b·A + c·A² runs as the fused beta·E
+//                    epilogue instead of two extra elementwise kernels)
+//   muon_gemm_nn_ax: C[M,N] = Bm[M,K]·X[K,N] + a·X[M,N]             (NN form —
+//                    X's K index is its row, so the X tile is staged as a
+//                    transposed LDS image exactly like attention's V tile,
+//                    and a·X is the fused epilogue)
+//
+// Design notes (measured rationale, guide §5):
+//  - A triangle-only syrk would halve the off-diagonal FLOPs, but at the NS
+//    shapes (m ≤ 2048 → T = m/128 ≤ 16 → T(T+1)/2 ≤ 136 blocks) the upper
+//    triangle under-fills the 256-CU chip by ~2x; the full (ti,tj) grid does
+//    2x the FLOPs at 2x the occupancy — same wall clock, no mirror pass, and
+//    the operands of X·Xᵀ are the SAME tensor so L2/L3 serve the re-reads.
+//  - 128x128 tile, BK=64, 4 waves (2x2), 64x64 per wave, 16 MFMAs
+//    (32x32x16 bf16) per wave per K-step, double-buffered LDS, the guide's
+//    minimum-2-phase schedule (stage t+1 issued before the ds_read+MFMA of
+//    tile t, one vmcnt(0)+barrier per K-step).
+//  - fp32 accumulation, bf16 storage; shapes padded to 128/64 multiples by
+//    the caller (zero padding is exact for the whole NS chain: padded rows
+//    and columns stay zero through XXᵀ, the combine and B·X).
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "attn_common.h"
+
+namespace {
+
+constexpr int BMN = 128;  // square output tile
+constexpr int BK = 64;    // K step
+constexpr int TPB = 256;  // 4 waves
+
+// byte-level XOR swizzle for the row-major [128][BK] bf16 images: 16-B chunk
+// index (3 bits) XOR (row & 7) — ds_read_b128 lane groups read 16 different
+// rows at one chunk column, which a linear image makes an 8..16-way bank
+// conflict (guide §6 G4).
+__device__ __forceinline__ int swz_chunk(int chunk, int row) { return chunk ^ (row & 7); }
+
+// stage a [128][BK] row-major bf16 tile into a swizzled lane-linear LDS image
+// via global_load_lds (16 B per lane; the swizzle is applied to the SOURCE
+// address — guide §5.4 rule 21). src points at element [0][0] of the tile,
+// ld = row stride in elements. lds_base is the wave-uniform image base.
+__device__ __forceinline__ void glds_tile(const __hip_bfloat16* src, long ld,
+                                          __hip_bfloat16* lds_base) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int w = tid >> 6;
+  // each wave stages 32 rows: 4 glds of 8 rows (64 lanes x 16 B = 1 KiB)
+#pragma unroll
+  for (int g = 0; g < 4; ++g) {
+    const int row = w * 32 + g * 8 + (lane >> 3);
+    const int chunk = swz_chunk(lane & 7, row);
+    const __hip_bfloat16* gp = src + (long)row * ld + chunk * 8;
+    __hip_bfloat16* lp = lds_base + (w * 32 + g * 8) * BK;  // wave-uniform
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gp,
+        (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+  }
+}
+
+// read one 32x16 A/B-operand fragment: rows r0+lane&31, k-chunk kc*2+hi of
+// the swizzled [128][BK] image (b128, conflict <= 2-way)
+__device__ __forceinline__ bf16x8 img_frag(const __hip_bfloat16* img, int r0,
+                                           int kc, int lane) {
+  const int row = r0 + (lane & 31);
+  const int chunk = swz_chunk(kc * 2 + (lane >> 5), row);
+  Bf16x8U u;
+  *reinterpret_cast<uint4*>(u.s) =
+      *reinterpret_cast<const uint4*>(img + row * BK + chunk * 8);
+  return u.v;
+}
+
+// ---------------------------------------------------------------------------
+// C[M,N] = alpha * X[M,K] @ Y[N,K]^T + beta * E[M,N]
+template <bool HAS_E>
+__global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
+    const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ Y,
+    const __hip_bfloat16* __restrict__ E, __hip_bfloat16* __restrict__ C,
+    int M, int N, int K, float alpha, float beta) {
+  constexpr int TILE = BMN * BK;          // elements per image
+  __shared__ __hip_bfloat16 smem[2 * 2 * TILE];
+
+  const int tm = blockIdx.x, tn = blockIdx.y;
+  const int m0 = tm * BMN, n0 = tn * BMN;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = (wave >> 1) * 64, wc = (wave & 1) * 64;
+  const int lq = lane & 31, hi = lane >> 5;
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x16{};
+
+  const int ksteps = K / BK;
+  // prologue: stage tile 0 into buf 0
+  glds_tile(X + (long)m0 * K, K, smem);
+  glds_tile(Y + (long)n0 * K, K, smem + TILE);
+  __syncthreads();
+
+  for (int kt = 0; kt < ksteps; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < ksteps) {  // issue next-tile staging FIRST (T3 min-2-phase)
+      glds_tile(X + (long)m0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE);
+      glds_tile(Y + (long)n0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE + TILE);
+    }
+    const __hip_bfloat16* ax = smem + buf * 2 * TILE;
+    const __hip_bfloat16* by = ax + TILE;
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kc = 0; kc < BK / 16; ++kc) {
+      bf16x8 af[2], bf[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) af[i] = img_frag(ax, wr + i * 32, kc, lane);
+#pragma unroll
+      for (int j = 0; j < 2; ++j) bf[j] = img_frag(by, wc + j * 32, kc, lane);
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();  // drains the in-flight glds (vmcnt(0)) + buffer swap
+  }
+
+  // epilogue: C = alpha*acc + beta*E, element (m,n) at lane n=lq, reg row
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int m = m0 + wr + i * 32 + acc_row(reg, hi);
+        const int n = n0 + wc + j * 32 + lq;
+        float v = alpha * acc[i][j][reg];
+        if constexpr (HAS_E) v += beta * __bfloat162float(E[(long)m * N + n]);
+        C[(long)m * N + n] = __float2bfloat16(v);
+      }
+}
+
+// ---------------------------------------------------------------------------
+// C[M,N] = Bm[M,K] @ X[K,N] + a * X2[M,N]   (X2 = the same X when M == K)
+__global__ __launch_bounds__(TPB) void muon_gemm_nn_ax_kernel(
+    const __hip_bfloat16* __restrict__ Bm, const __hip_bfloat16* __restrict__ X,
+    const __hip_bfloat16* __restrict__ X2, __hip_bfloat16* __restrict__ C,
+    int M, int N, int K, float a) {
+  constexpr int TILE = BMN * BK;   // A image [128 m][64 k]
+  constexpr int VROW = 64;         // transposed X image rows [128 n][64 k]
+  __shared__ __hip_bfloat16 smem[2 * (TILE + BMN * VROW)];
+
+  const int tm = blockIdx.x, tn = blockIdx.y;
+  const int m0 = tm * BMN, n0 = tn * BMN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64, wc = (wave & 1) * 64;
+  const int lq = lane & 31, hi = lane >> 5;
+
+  // register staging for the transposed X image (glds cannot transpose):
+  // chunks of (2 k-rows x 8 n-cols), pair-packed b32 writes (attention's V
+  // staging pattern; conflict-free reads via swzt)
+  constexpr int CH = (BK / 2) * (BMN / 8) / TPB;  // = 4 chunks per thread
+  uint4 xreg[CH][2];
+
+  auto xstage_load = [&](int k0) {
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int u = tid + c * TPB;
+      const int kr = (u / (BMN / 8)) * 2;
+      const int n = (u % (BMN / 8)) * 8;
+      const __hip_bfloat16* base = X + (long)(k0 + kr) * N + n0 + n;
+      xreg[c][0] = *reinterpret_cast<const uint4*>(base);
+      xreg[c][1] = *reinterpret_cast<const uint4*>(base + N);
+    }
+  };
+  auto xstage_write = [&](__hip_bfloat16* xt) {
+#pragma unroll
+    for (int c = 0; c < CH; ++c) {
+      const int u = tid + c * TPB;
+      const int kr = (u / (BMN / 8)) * 2;
+      const int n = (u % (BMN / 8)) * 8;
+      Bf16x8U v0, v1;
+      *reinterpret_cast<uint4*>(v0.s) = xreg[c][0];
+      *reinterpret_cast<uint4*>(v1.s) = xreg[c][1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int nrow = n + j;
+        const uint pair = (uint)v0.s[j] | ((uint)v1.s[j] << 16);
+        *reinterpret_cast<uint*>(xt + nrow * VROW + (kr ^ swzt(nrow))) = pair;
+      }
+    }
+  };
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x16{};
+
+  const int ksteps = K / BK;
+  glds_tile(Bm + (long)m0 * K, K, smem);
+  xstage_load(0);
+  xstage_write(smem + TILE);
+  __syncthreads();
+
+  for (int kt = 0; kt < ksteps; ++kt) {
+    const int buf = kt & 1;
+    __hip_bfloat16* cur = smem + buf * (TILE + BMN * VROW);
+    __hip_bfloat16* nxt = smem + (buf ^ 1) * (TILE + BMN * VROW);
+    const bool has_next = kt + 1 < ksteps;
+    if (has_next) {
+      glds_tile(Bm + (long)m0 * K + (kt + 1) * BK, K, nxt);
+      xstage_load((kt + 1) * BK);  // loads in flight under the MFMAs (T14)
+    }
+    const __hip_bfloat16* aimg = cur;
+    const __hip_bfloat16* xt = cur + TILE;
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int kc = 0; kc < BK / 16; ++kc) {
+      bf16x8 af[2], bf[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i) af[i] = img_frag(aimg, wr + i * 32, kc, lane);
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        const int nrow = wc + j * 32 + lq;
+        Bf16x8U u;
+        *reinterpret_cast<uint4*>(u.s) = *reinterpret_cast<const uint4*>(
+            xt + nrow * VROW + ((kc * 16 + hi * 8) ^ swzt(nrow)));
+        bf[j] = u.v;
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (has_next) xstage_write(nxt + TILE);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int m = m0 + wr + i * 32 + acc_row(reg, hi);
+        const int n = n0 + wc + j * 32 + lq;
+        const float v = acc[i][j][reg] + a * __bfloat162float(X2[(long)m * N + n]);
+        C[(long)m * N + n] = __float2bfloat16(v);
+      }
+}
+
+void check_2d_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16 && t.dim() == 2 &&
+                  t.is_contiguous(),
+              "muon: ", name, " must be a contiguous 2-D bf16 CUDA tensor");
+}
+
+}  // namespace
+
+void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
+                  double beta, at::Tensor E) {
+  check_2d_bf16(X, "X");
+  check_2d_bf16(Y, "Y");
+  check_2d_bf16(C, "C");
+  const int M = X.size(0), K = X.size(1), N = Y.size(0);
+  TORCH_CHECK(Y.size(1) == K && C.size(0) == M && C.size(1) == N,
+              "muon_gemm_nt: shape mismatch");
+  TORCH_CHECK(M % BMN == 0 && N % BMN == 0 && K % BK == 0,
+              "muon_gemm_nt: pad to 128/128/64 multiples");
+  const bool has_e = E.numel() > 0;
+  if (has_e) check_2d_bf16(E, "E");
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid(M / BMN, N / BMN), block(TPB);
+  auto* xp = reinterpret_cast<const __hip_bfloat16*>(X.data_ptr());
+  auto* yp = reinterpret_cast<const __hip_bfloat16*>(Y.data_ptr());
+  auto* ep = has_e ? reinterpret_cast<const __hip_bfloat16*>(E.data_ptr()) : nullptr;
+  auto* cp = reinterpret_cast<__hip_bfloat16*>(C.data_ptr());
+  if (has_e)
+    muon_gemm_nt_kernel<true><<<grid, block, 0, stream>>>(xp, yp, ep, cp, M, N, K,
+                                                          (float)alpha, (float)beta);
+  else
+    muon_gemm_nt_kernel<false><<<grid, block, 0, stream>>>(xp, yp, ep, cp, M, N, K,
+                                                           (float)alpha, (float)beta);
+}
+
+void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a) {
+  check_2d_bf16(Bm, "Bm");
+  check_2d_bf16(X, "X");
+  check_2d_bf16(C, "C");
+  const int M = Bm.size(0), K = Bm.size(1), N = X.size(1);
+  TORCH_CHECK(X.size(0) == K && C.size(0) == M && C.size(1) == N,
+              "muon_gemm_nn_ax: shape mismatch");
+  TORCH_CHECK(M % BMN == 0 && N % BMN == 0 && K % BK == 0,
+              "muon_gemm_nn_ax: pad to 128/128/64 multiples");
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid(M / BMN, N / BMN), block(TPB);
+  muon_gemm_nn_ax_kernel<<<grid, block, 0, stream>>>(
+      reinterpret_cast<const __hip_bfloat16*>(Bm.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(X.data_ptr()),
+      reinterpret_cast<const __hip_bfloat16*>(X.data_ptr()),
+      reinterpret_cast<__hip_bfloat16*>(C.data_ptr()), M, N, K, (float)a);
+}
