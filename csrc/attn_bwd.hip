@@ -40,7 +40,9 @@ constexpr float LOG2E = 1.4426950408889634f;
 inline int bwd_nw() {
   static int nw = []() {
     const char* e = getenv("MCDP_ATTN_BWD_NW");
-    return e ? atoi(e) : 8;
+    // default 4: two co-resident 4-wave blocks per CU hide each other's tile
+    // barriers (dkv WAIT_ANY 52-59% at NW=8; bwd 329 -> 372 TF measured)
+    return e ? atoi(e) : 4;
   }();
   return nw;
 }
@@ -374,33 +376,45 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   // (2 rows x 8 d), threads 256..511 own dO chunks; each thread carries one
   // chunk pair per sub-tile.
   constexpr int CH_TOT = (32 / 2) * (D / 8);  // 256 chunks per tensor-tile
-  const bool is_q_half = tid < CH_TOT;
-  uint4 sreg[2];  // ONE sub-tile's chunk pair at a time (split-half staging:
-                  // load s -> compute s-1 -> write s; keeps peak VGPR down)
+  // chunk units per thread: 1 at NW=8 (Q half / dO half split across the
+  // 512 threads), 2 at NW=4 — each unit decides Q-vs-dO by its global index
+  constexpr int NU = (2 * CH_TOT + TPB - 1) / TPB;
+  uint4 sreg[NU][2];  // ONE sub-tile's chunk pairs at a time (split-half
+                      // staging: load s -> compute s-1 -> write s)
 
   auto stage_load = [&](int hq_, int q0) {
-    const int u = is_q_half ? tid : tid - CH_TOT;
-    if (u >= CH_TOT) return;
-    const int row = (u / (D / 8)) * 2;
-    const int d0 = (u % (D / 8)) * 8;
-    const __hip_bfloat16* src = is_q_half ? q : dout;
-    const long rs_ = is_q_half ? q_rs : do_rs;
-    const bool ok0 = q0 + row < Sq;
-    const bool ok1 = q0 + row + 1 < Sq;
-    const long base = ((long)b * Sq + q0 + row) * rs_ + (long)hq_ * D + d0;
-    sreg[0] = ok0 ? *reinterpret_cast<const uint4*>(src + base) : uint4{0, 0, 0, 0};
-    sreg[1] = ok1 ? *reinterpret_cast<const uint4*>(src + base + rs_) : uint4{0, 0, 0, 0};
+#pragma unroll
+    for (int cu = 0; cu < NU; ++cu) {
+      const int u0 = tid + cu * TPB;
+      if (u0 >= 2 * CH_TOT) continue;
+      const bool isq = u0 < CH_TOT;
+      const int u = isq ? u0 : u0 - CH_TOT;
+      const int row = (u / (D / 8)) * 2;
+      const int d0 = (u % (D / 8)) * 8;
+      const __hip_bfloat16* src = isq ? q : dout;
+      const long rs_ = isq ? q_rs : do_rs;
+      const bool ok0 = q0 + row < Sq;
+      const bool ok1 = q0 + row + 1 < Sq;
+      const long base = ((long)b * Sq + q0 + row) * rs_ + (long)hq_ * D + d0;
+      sreg[cu][0] = ok0 ? *reinterpret_cast<const uint4*>(src + base) : uint4{0, 0, 0, 0};
+      sreg[cu][1] = ok1 ? *reinterpret_cast<const uint4*>(src + base + rs_) : uint4{0, 0, 0, 0};
+    }
   };
   auto stage_write = [&](int bufsel, int s) {
     __hip_bfloat16* q_lds = smem + bufsel * TILE;
     __hip_bfloat16* do_lds = q_lds + 64 * D;
-    const int u = is_q_half ? tid : tid - CH_TOT;
-    if (u >= CH_TOT) return;
-    const int row = (u / (D / 8)) * 2;
-    const int d0 = (u % (D / 8)) * 8;
-    __hip_bfloat16* dst = (is_q_half ? q_lds : do_lds) + s * 32 * D;
-    *reinterpret_cast<uint4*>(dst + rm_swz<D>(row, d0)) = sreg[0];
-    *reinterpret_cast<uint4*>(dst + rm_swz<D>(row + 1, d0)) = sreg[1];
+#pragma unroll
+    for (int cu = 0; cu < NU; ++cu) {
+      const int u0 = tid + cu * TPB;
+      if (u0 >= 2 * CH_TOT) continue;
+      const bool isq = u0 < CH_TOT;
+      const int u = isq ? u0 : u0 - CH_TOT;
+      const int row = (u / (D / 8)) * 2;
+      const int d0 = (u % (D / 8)) * 8;
+      __hip_bfloat16* dst = (isq ? q_lds : do_lds) + s * 32 * D;
+      *reinterpret_cast<uint4*>(dst + rm_swz<D>(row, d0)) = sreg[cu][0];
+      *reinterpret_cast<uint4*>(dst + rm_swz<D>(row + 1, d0)) = sreg[cu][1];
+    }
   };
 
   // iterate (GQA head, q tile) pairs with a flat prefetch pipeline

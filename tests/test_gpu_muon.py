@@ -33,8 +33,10 @@ def test_muon_gemm_nt_numerics(M, N, K):
     E = torch.randn(M, N, device=dev).to(torch.bfloat16)
     ext.muon_gemm_nt(X, Y, C, 2.0315, -4.775, E)
     ref2 = 2.0315 * ref - 4.775 * E.float()
+    # C is stored bf16: tolerance is relative to the output magnitude
+    tol = 1e-2 * ref2.abs().max().item() + 2e-2
     err2 = (C.float() - ref2).abs().max().item()
-    assert err2 < 6e-2, f"NT+E epilogue max err {err2}"
+    assert err2 < tol, f"NT+E epilogue max err {err2} (tol {tol})"
 
 
 @pytest.mark.parametrize("M,N", [(128, 128), (256, 640)])
@@ -46,8 +48,9 @@ def test_muon_gemm_nn_ax_numerics(M, N):
     C = torch.empty(M, N, dtype=torch.bfloat16, device=dev)
     ext.muon_gemm_nn_ax(B, X, C, 3.4445)
     ref = B.float() @ X.float() + 3.4445 * X.float()
+    tol = 1e-2 * ref.abs().max().item() + 2e-2
     err = (C.float() - ref).abs().max().item()
-    assert err < 4e-2, f"NN+aX gemm max err {err}"
+    assert err < tol, f"NN+aX gemm max err {err} (tol {tol})"
 
 
 @pytest.mark.parametrize("m,n", [(256, 512), (200, 300), (512, 384), (1024, 2816)])
