@@ -628,6 +628,10 @@ void launch_all_nw(dim3 gq, dim3 gkv, dim3 block, hipStream_t stream,
     launch_bwd_all<D, MOD, 16>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv,
                                slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs,
                                do_rs, dq_rs, dk_rs, dv_rs);
+  else if (bwd_nw() == 4)
+    launch_bwd_all<D, MOD, 4>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv,
+                              slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs,
+                              do_rs, dq_rs, dk_rs, dv_rs);
   else
     launch_bwd_all<D, MOD, 8>(gq, gkv, block, stream, q, k, v, dout, lse, drow, dq, dk, dv,
                               slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs,
