@@ -40,9 +40,9 @@ constexpr float LOG2E = 1.4426950408889634f;
 inline int bwd_nw() {
   static int nw = []() {
     const char* e = getenv("MCDP_ATTN_BWD_NW");
-    // default 4: two co-resident 4-wave blocks per CU hide each other's tile
-    // barriers (dkv WAIT_ANY 52-59% at NW=8; bwd 329 -> 372 TF measured)
-    return e ? atoi(e) : 4;
+    // NW=4 measured 2.4x SLOWER at D=128 (call7: 139 vs 331 TF) once the
+    // launch bug that faked its earlier win was fixed — default stays 8.
+    return e ? atoi(e) : 8;
   }();
   return nw;
 }
@@ -111,7 +111,8 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
 
   __shared__ __align__(16) __hip_bfloat16 smem[2 * TILE];
 
-  const int b = blockIdx.z, hq = blockIdx.y, qtile = balance_x();
+  const TileMap tmap = tile_map();
+  const int b = tmap.batch, hq = tmap.head, qtile = tmap.tile;
   const int hkv = hq / (Hq / Hkv);
   const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
   const int lq = lane & 31, hi = lane >> 5;
@@ -329,7 +330,8 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   __shared__ __align__(16) __hip_bfloat16 smem[2 * TILE];
   __shared__ float stats_lds[2][2][64];  // [buf][L|D][s*32 + qrow]
 
-  const int b = blockIdx.z, hkv = blockIdx.y, kvtile = balance_x();
+  const TileMap tmap = tile_map();
+  const int b = tmap.batch, hkv = tmap.head, kvtile = tmap.tile;
   const int group = Hq / Hkv;
   const int tid = threadIdx.x, wave = tid / WAVE, lane = tid % WAVE;
   const int lk = lane & 31, hi = lane >> 5;

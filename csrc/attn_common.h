@@ -181,6 +181,38 @@ __device__ __forceinline__ int balance_x() {
   return x;
 }
 
+// tile_map(): linear block id -> (tile, head, batch) with TWO goals on top
+// of balance_x()'s rotation:
+//   (a) every tile-block of one (batch, head) lands on ONE XCD (dispatch
+//       places linear id i on XCD i%8), so the per-(b,h) operand stream
+//       (K/V for fwd+dQ, Q/dO for dKV — ~1 MB at S=2048/D=128) is pulled
+//       into that XCD's private 4 MiB L2 once and re-read from L2 instead
+//       of being re-fetched by all 8 XCDs (attention bwd is memory-bound on
+//       exactly this re-streaming);
+//   (b) a CU's successive blocks rotate through tile indices (causal work
+//       balance, same mechanism as balance_x).
+// Head-major layout gives (a) when gridDim.y (heads) is a multiple of 8;
+// bijectivity of the tile rotation needs nt | (256/nh). Falls back to
+// balance_x() + natural (y,z) otherwise.
+struct TileMap { int tile, head, batch; };
+__device__ __forceinline__ TileMap tile_map() {
+  const int nt = gridDim.x, nh = gridDim.y;
+  TileMap m;
+  if ((nh & 7) == 0 && (256 % nh) == 0 && ((256 / nh) % nt) == 0) {
+    const long lin = blockIdx.x + (long)nt * (blockIdx.y + (long)nh * blockIdx.z);
+    m.head = (int)(lin % nh);          // consecutive ids differ in head -> XCD = head % 8
+    const long rest = lin / nh;        // = t0 + nt * batch
+    const int t0 = (int)(rest % nt);
+    m.batch = (int)(rest / nt);
+    m.tile = (int)((t0 + rest / (256 / nh)) % nt);
+  } else {
+    m.tile = balance_x();
+    m.head = blockIdx.y;
+    m.batch = blockIdx.z;
+  }
+  return m;
+}
+
 // keep/mask decision for one score element. q_pos/k_pos are ABSOLUTE
 // positions (q_pos = q_row + Skv - Sq handles KV-cache decode).
 template <int MOD>

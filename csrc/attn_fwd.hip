@@ -55,9 +55,10 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   // that precedes its reuse at j+3).
   __shared__ __hip_bfloat16 smem[3 * TILE];
 
-  const int b = blockIdx.z;
-  const int hq = blockIdx.y;
-  const int qtile = balance_x();
+  const TileMap tmap = tile_map();
+  const int b = tmap.batch;
+  const int hq = tmap.head;
+  const int qtile = tmap.tile;
   const int hkv = hq / (Hq / Hkv);
   const int tid = threadIdx.x;
   const int wave = tid / WAVE;
