@@ -173,6 +173,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[dc][r] = 0.f;
 
+  // T5 static form: the second-dispatched half (waves 4-7) loses VALU
+  // arbitration on every segment; one static priority raise for it (and no
+  // per-cluster flips: s_setprio is a scheduling fence that was keeping the
+  // exp/pack VALU OUT of the PV MFMA issue gaps -- seen in the .s).
+  if (NW == 8 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+
   // prologue: stage tile 0 into slot 0
   stage_load(kv_lo);
   stage_write(0);
@@ -204,7 +210,6 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     f32x16 st[KT];
 #pragma unroll
     for (int kt = 0; kt < KT; ++kt) st[kt] = f32x16{};
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int kt = 0; kt < KT; ++kt) {
       const int krow = kt * 32 + lq;
@@ -216,7 +221,6 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
         st[kt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf.v, qf[dblk], st[kt], 0, 0, 0);
       }
     }
-    __builtin_amdgcn_s_setprio(0);
 
     // ---- tile max (tree reduce, raw units on the full fast path) ----
     // interior tiles (wave-uniform): every (q,k) pair of this wave is kept
@@ -270,11 +274,16 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     const bool defer = __all(tmax - m <= 8.0f);
     if (!defer) m = fmaxf(m, tmax);
     const float mc = fmaxf(m, -1e30f);
+    // branchless forms so [PV MFMAs + exp + psum + pack] is ONE basic block
+    // (a branch splits the scheduling region and exiles the VALU from the
+    // MFMA issue gaps): masked-path st is pre-scaled, so sc2 folds to 1.
+    const float sc2 = full ? scale2 : 1.0f;
+    // alpha == 1 exactly when defer (mc == m_old); exp2(-inf)=0 at tile 0.
+    const float alpha = __builtin_amdgcn_exp2f(m_old - mc);
 
     // ---- PV(j-1): accumulates P_{j-1}*V at scale m_{j-1} (o_acc still at
     // that scale). The exp/psum/pack block below is independent of o_acc
     // and interleaves into these MFMAs' issue gaps. ----
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dc = 0; dc < DCOL; ++dc) {
       f32x16 acc;
@@ -293,35 +302,38 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) o_acc[dc][r] = acc[r];
     }
-    __builtin_amdgcn_s_setprio(0);
 
     // ---- exp (base-2) + row sum + pack for the NEXT iteration's PV ----
+    // branchless (sc2 / alpha above) — same basic block as the PV MFMAs
     float p[KT][16];
     float psum = 0.f;
-    if (full) {
 #pragma unroll
-      for (int kt = 0; kt < KT; ++kt)
+    for (int kt = 0; kt < KT; ++kt)
 #pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          p[kt][reg] = __builtin_amdgcn_exp2f(fmaf(st[kt][reg], scale2, -mc));
-          psum += p[kt][reg];
-        }
-    } else {
-#pragma unroll
-      for (int kt = 0; kt < KT; ++kt)
-#pragma unroll
-        for (int reg = 0; reg < 16; ++reg) {
-          p[kt][reg] = __builtin_amdgcn_exp2f(st[kt][reg] - mc);
-          psum += p[kt][reg];
-        }
-    }
+      for (int reg = 0; reg < 16; ++reg) {
+        p[kt][reg] = __builtin_amdgcn_exp2f(fmaf(st[kt][reg], sc2, -mc));
+        psum += p[kt][reg];
+      }
     psum = cross32_sum(psum);
-    if (defer) {
-      l += psum;
-    } else {
-      const float alpha = __builtin_amdgcn_exp2f(m_old - mc);
-      l = l * alpha + psum;
-      // o_acc rescale: strictly after PV(j-1), before PV(j) (T13 hazard)
+    l = l * alpha + psum;  // alpha == 1 on the defer path
+#pragma unroll
+    for (int kt = 0; kt < KT; ++kt) acc_to_afrag(p[kt], pa[kt][0], pa[kt][1]);
+
+    // sm-split (guide T19 + ladder rung 2b): the scheduler otherwise emits
+    // the whole exp/psum/pack VALU block AFTER the PV MFMAs (checked in the
+    // .s); these directives interleave ~24 VALU per 2 MFMAs so the VALU
+    // issues inside the matrix-pipe gaps.
+#pragma unroll
+    for (int g = 0; g < 8; ++g) {
+      __builtin_amdgcn_sched_group_barrier(0x100, 4, 0);  // DS_READ (V frags)
+      __builtin_amdgcn_sched_group_barrier(0x008, 2, 0);  // MFMA
+      __builtin_amdgcn_sched_group_barrier(0x002, 24, 0); // VALU slice
+    }
+
+    // o_acc rescale: strictly after PV(j-1), before PV(j) (T13 hazard);
+    // skipped on the defer path (alpha == 1 there, the l-update above is
+    // already folded in)
+    if (!defer) {
 #pragma unroll
       for (int reg = 0; reg < 16; ++reg) {
         const float ar = __shfl(alpha, acc_row(reg, hi));
@@ -329,8 +341,6 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
         for (int dc = 0; dc < DCOL; ++dc) o_acc[dc][reg] *= ar;
       }
     }
-#pragma unroll
-    for (int kt = 0; kt < KT; ++kt) acc_to_afrag(p[kt], pa[kt][0], pa[kt][1]);
 
     // write tile t+1 LAST: the vmcnt wait on stage_load's global loads has
     // the whole iteration's MFMA/softmax work to hide under (T14).
@@ -345,7 +355,6 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   // ---- drain: PV of the last tile ----
   {
     const __hip_bfloat16* vt_prev = smem + pv_slot * TILE + KVB * D;
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dc = 0; dc < DCOL; ++dc) {
       f32x16 acc;
@@ -364,7 +373,6 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) o_acc[dc][r] = acc[r];
     }
-    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- epilogue ----
