@@ -26,7 +26,11 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
   unsigned packed;
-  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(packed) : "v"(lo), "v"(hi));
+  // s_nop 1 = the 2 wait states of the gfx950 "VALU write -> v_permlane*
+  // read" hazard (guide T21): the consumer is often permlane32_swap and
+  // hipcc cannot pad an inline-asm write (it bit under sched_group_barrier
+  // interleaving: every fwd numerics test failed until padded here).
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(packed) : "v"(lo), "v"(hi));
   return packed;
 }
 
