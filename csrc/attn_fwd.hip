@@ -178,6 +178,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   // per-cluster flips: s_setprio is a scheduling fence that was keeping the
   // exp/pack VALU OUT of the PV MFMA issue gaps -- seen in the .s).
   if (NW == 8 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
 
   // prologue: stage tile 0 into slot 0
   stage_load(kv_lo);
