@@ -177,8 +177,10 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   // arbitration on every segment; one static priority raise for it (and no
   // per-cluster flips: s_setprio is a scheduling fence that was keeping the
   // exp/pack VALU OUT of the PV MFMA issue gaps -- seen in the .s).
+#if !defined(MCDP_NO_STATPRIO)
   if (NW == 8 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);
+#endif
 
   // prologue: stage tile 0 into slot 0
   stage_load(kv_lo);
@@ -324,12 +326,14 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     // the whole exp/psum/pack VALU block AFTER the PV MFMAs (checked in the
     // .s); these directives interleave ~24 VALU per 2 MFMAs so the VALU
     // issues inside the matrix-pipe gaps.
+#if !defined(MCDP_NO_SGB)
 #pragma unroll
     for (int g = 0; g < 8; ++g) {
       __builtin_amdgcn_sched_group_barrier(0x100, 4, 0);  // DS_READ (V frags)
       __builtin_amdgcn_sched_group_barrier(0x008, 2, 0);  // MFMA
       __builtin_amdgcn_sched_group_barrier(0x002, 24, 0); // VALU slice
     }
+#endif
 
     // o_acc rescale: strictly after PV(j-1), before PV(j) (T13 hazard);
     // skipped on the defer path (alpha == 1 there, the l-update above is
