@@ -31,7 +31,7 @@ constexpr int QPW = 32;  // q rows per wave
 constexpr float LOG2E = 1.4426950408889634f;
 constexpr float LN2 = 0.6931471805599453f;
 
-template <int D, int MOD, int NW, int KVB>
+template <int D, int MOD, int NW, int KVB, int VAR = 3>
 __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ q, const __hip_bfloat16* __restrict__ k,
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
@@ -177,10 +177,8 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   // arbitration on every segment; one static priority raise for it (and no
   // per-cluster flips: s_setprio is a scheduling fence that was keeping the
   // exp/pack VALU OUT of the PV MFMA issue gaps -- seen in the .s).
-#if !defined(MCDP_NO_STATPRIO)
-  if (NW == 8 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+  if ((VAR & 1) && NW == 8 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);
-#endif
 
   // prologue: stage tile 0 into slot 0
   stage_load(kv_lo);
@@ -326,14 +324,14 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     // the whole exp/psum/pack VALU block AFTER the PV MFMAs (checked in the
     // .s); these directives interleave ~24 VALU per 2 MFMAs so the VALU
     // issues inside the matrix-pipe gaps.
-#if !defined(MCDP_NO_SGB)
+    if constexpr ((VAR & 2) != 0) {
 #pragma unroll
-    for (int g = 0; g < 8; ++g) {
-      __builtin_amdgcn_sched_group_barrier(0x100, 4, 0);  // DS_READ (V frags)
-      __builtin_amdgcn_sched_group_barrier(0x008, 2, 0);  // MFMA
-      __builtin_amdgcn_sched_group_barrier(0x002, 24, 0); // VALU slice
+      for (int g = 0; g < 8; ++g) {
+        __builtin_amdgcn_sched_group_barrier(0x100, 4, 0);  // DS_READ (V frags)
+        __builtin_amdgcn_sched_group_barrier(0x008, 2, 0);  // MFMA
+        __builtin_amdgcn_sched_group_barrier(0x002, 24, 0); // VALU slice
+      }
     }
-#endif
 
     // o_acc rescale: strictly after PV(j-1), before PV(j) (T13 hazard);
     // skipped on the defer path (alpha == 1 there, the l-update above is
@@ -415,9 +413,25 @@ void launch_fwd_cfg(dim3 grid, dim3 block, hipStream_t stream,
   const char* e = getenv("MCDP_ATTN_KVB");
   const int kvb = e ? atoi(e) : 64;
   const int nw = fwd_qpb() / QPW;
+  static const int var = []() {
+    const char* e = getenv("MCDP_ATTN_FWD_VAR");
+    return e ? atoi(e) : 3;  // bit0 static-prio, bit1 sched_group_barrier
+  }();
 #define LAUNCH(NW_, KVB_)                                                              \
-  attn_fwd_kernel<D, MOD, NW_, KVB_><<<grid, block, 0, stream>>>(                      \
-      q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs)
+  do {                                                                                 \
+    if (var == 0)                                                                      \
+      attn_fwd_kernel<D, MOD, NW_, KVB_, 0><<<grid, block, 0, stream>>>(               \
+          q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs); \
+    else if (var == 1)                                                                 \
+      attn_fwd_kernel<D, MOD, NW_, KVB_, 1><<<grid, block, 0, stream>>>(               \
+          q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs); \
+    else if (var == 2)                                                                 \
+      attn_fwd_kernel<D, MOD, NW_, KVB_, 2><<<grid, block, 0, stream>>>(               \
+          q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs); \
+    else                                                                               \
+      attn_fwd_kernel<D, MOD, NW_, KVB_, 3><<<grid, block, 0, stream>>>(               \
+          q, k, v, o, lse, slopes, B, Sq, Skv, Hq, Hkv, scale, modarg, q_rs, k_rs, v_rs); \
+  } while (0)
   if (nw == 8 && kvb == 64) LAUNCH(8, 64);
   else if (nw == 8 && kvb == 32) LAUNCH(8, 32);
   else if (nw == 4 && kvb == 64) LAUNCH(4, 64);
