@@ -415,7 +415,11 @@ void launch_fwd_cfg(dim3 grid, dim3 block, hipStream_t stream,
   const int nw = fwd_qpb() / QPW;
   static const int var = []() {
     const char* e = getenv("MCDP_ATTN_FWD_VAR");
-    return e ? atoi(e) : 3;  // bit0 static-prio, bit1 sched_group_barrier
+    // default 1 = static-prio only. bit1 (sched_group_barrier interleave)
+    // CORRUPTS numerics (NaN via a poisoned cross-lane max; bisect
+    // gpurun_out/r2_call14.log: VAR 0/1 exact, VAR 2/3 NaN on every shape)
+    // — kept compiled for future debugging, not for use.
+    return e ? atoi(e) : 1;
   }();
 #define LAUNCH(NW_, KVB_)                                                              \
   do {                                                                                 \
