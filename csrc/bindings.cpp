@@ -34,6 +34,7 @@ void sgd_step(at::Tensor param, at::Tensor master, at::Tensor grad, at::Tensor b
 void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
                   double beta, at::Tensor E);
 void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a);
+void shampoo_stats_update(at::Tensor G, at::Tensor S, double beta);
 
 // attn_fwd.hip / attn_bwd.hip
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale,
@@ -92,6 +93,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step, "fused SGD over flat buffers");
   m.def("muon_gemm_nt", &muon_gemm_nt, "C = alpha*X@Y^T + beta*E (bf16 MFMA)");
   m.def("muon_gemm_nn_ax", &muon_gemm_nn_ax, "C = Bm@X + a*X (bf16 MFMA)");
+  m.def("shampoo_stats_update", &shampoo_stats_update,
+        "S = beta*S + (1-beta)*G@G^T (fp32 state, bf16 MFMA)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (o, lse)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward (dq, dk, dv)");
   m.def("attn_bwd_out", &attn_bwd_out,

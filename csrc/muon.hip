@@ -79,10 +79,10 @@ __device__ __forceinline__ bf16x8 img_frag(const __hip_bfloat16* img, int r0,
 
 // ---------------------------------------------------------------------------
 // C[M,N] = alpha * X[M,K] @ Y[N,K]^T + beta * E[M,N]
-template <bool HAS_E>
+template <bool HAS_E, typename TO = __hip_bfloat16>
 __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
     const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ Y,
-    const __hip_bfloat16* __restrict__ E, __hip_bfloat16* __restrict__ C,
+    const TO* __restrict__ E, TO* __restrict__ C,
     int M, int N, int K, float alpha, float beta) {
   constexpr int TILE = BMN * BK;          // elements per image
   __shared__ __hip_bfloat16 smem[2 * 2 * TILE];
@@ -142,8 +142,8 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
         const int m = m0 + wr + i * 32 + acc_row(reg, hi);
         const int n = n0 + wc + j * 32 + lq;
         float v = alpha * acc[i][j][reg];
-        if constexpr (HAS_E) v += beta * __bfloat162float(E[(long)m * N + n]);
-        C[(long)m * N + n] = __float2bfloat16(v);
+        if constexpr (HAS_E) v += beta * to_f32(E[(long)m * N + n]);
+        from_f32(&C[(long)m * N + n], v);
       }
 }
 
@@ -293,6 +293,25 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
   else
     muon_gemm_nt_kernel<false><<<grid, block, 0, stream>>>(xp, yp, ep, cp, M, N, K,
                                                            (float)alpha, (float)beta);
+}
+
+// K9 Shampoo statistics EMA on the same MFMA NT kernel, fp32 state:
+// S = beta*S + (1-beta) * G @ G^T   (left stats; pass G^T for right stats).
+// In-place E==C is safe: each thread reads its E element once before the
+// write. Parity: /root/reference/optimizers/shampoo.py:229-255.
+void shampoo_stats_update(at::Tensor G, at::Tensor S, double beta) {
+  check_2d_bf16(G, "G");
+  TORCH_CHECK(S.is_cuda() && S.scalar_type() == at::kFloat && S.dim() == 2 &&
+                  S.is_contiguous(), "shampoo: S must be contiguous fp32 2-D");
+  const int M = G.size(0), K = G.size(1);
+  TORCH_CHECK(S.size(0) == M && S.size(1) == M, "shampoo: S must be [M,M]");
+  TORCH_CHECK(M % BMN == 0 && K % BK == 0, "shampoo: pad to 128/64 multiples");
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid(M / BMN, M / BMN), block(TPB);
+  auto* gp = reinterpret_cast<const __hip_bfloat16*>(G.data_ptr());
+  auto* sp = S.data_ptr<float>();
+  muon_gemm_nt_kernel<true, float><<<grid, block, 0, stream>>>(
+      gp, gp, sp, sp, M, M, K, (float)(1.0 - beta), (float)beta);
 }
 
 void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a) {

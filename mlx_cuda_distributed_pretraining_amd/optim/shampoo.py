@@ -60,6 +60,33 @@ def matrix_inverse_pth_root(
     return X
 
 
+def _stats_update_hip(S: torch.Tensor, G: torch.Tensor, beta2: float, left: bool) -> None:
+    """K9: S = beta2*S + (1-beta2) * X @ X^T on the hand-written gfx950 MFMA
+    NT kernel (csrc/muon.hip, fp32-state instantiation), X = G or G^T.
+
+    Inputs are bf16-rounded for the matrix cores (fp32 accumulation, fp32
+    EMA state — the ~0.4% input rounding is far below the stats EMA noise);
+    shapes are zero-padded to tile multiples (exact: padded rows stay zero).
+    Parity: /root/reference/optimizers/shampoo.py:229-255.
+    """
+    from ..ops._ext import require_ext
+
+    ext = require_ext()
+    X = G if left else G.t()
+    d, k = X.shape
+    dp = (d + 127) // 128 * 128
+    kp = (k + 63) // 64 * 64
+    Gp = torch.zeros(dp, kp, dtype=torch.bfloat16, device=G.device)
+    Gp[:d, :k] = X
+    if dp == d:
+        ext.shampoo_stats_update(Gp, S, beta2)
+    else:
+        Sp = torch.zeros(dp, dp, dtype=torch.float32, device=G.device)
+        Sp[:d, :d] = S
+        ext.shampoo_stats_update(Gp, Sp, beta2)
+        S.copy_(Sp[:d, :d])
+
+
 class Shampoo(Optimizer):
     def __init__(
         self,
@@ -117,9 +144,13 @@ class Shampoo(Optimizer):
 
                 use_prec = "stat_l" in state
                 if use_prec:
-                    # statistics EMA
-                    state["stat_l"].mul_(hp.beta2).add_(g @ g.t(), alpha=1 - hp.beta2)
-                    state["stat_r"].mul_(hp.beta2).add_(g.t() @ g, alpha=1 - hp.beta2)
+                    # statistics EMA (K9 MFMA syrk kernel on GPU)
+                    if g.is_cuda:
+                        _stats_update_hip(state["stat_l"], g, hp.beta2, left=True)
+                        _stats_update_hip(state["stat_r"], g, hp.beta2, left=False)
+                    else:
+                        state["stat_l"].mul_(hp.beta2).add_(g @ g.t(), alpha=1 - hp.beta2)
+                        state["stat_r"].mul_(hp.beta2).add_(g.t() @ g, alpha=1 - hp.beta2)
                     if t >= hp.start_preconditioning_step and (
                         t % hp.update_period == 0 or t == hp.start_preconditioning_step
                     ):
