@@ -48,6 +48,18 @@ class DataManager:
 
         if self.synthetic:
             self.synthetic_vocab = int(getattr(data_cfg, "synthetic_vocab_size", 32000))
+            # synthetic ids feed the embedding directly: ids >= the model
+            # vocab are an out-of-bounds gather = a GPU memory fault with no
+            # python traceback (cost a day at 7B) — cap to the tokenizer
+            # vocab and warn instead.
+            tv = getattr(tokenizer, "vocab_size", None)
+            if tv is not None and self.synthetic_vocab > tv:
+                import warnings
+
+                warnings.warn(
+                    f"synthetic_vocab_size {self.synthetic_vocab} > tokenizer vocab {tv}; "
+                    f"capping to {tv} (set data.tokenizer.normal_vocab_size to match)")
+                self.synthetic_vocab = tv
             return
 
         self.stream = None
