@@ -61,6 +61,7 @@ class ModelArgs:
     # MoE layer is implemented there; kept for config parity.
     num_local_experts: int = 0
     num_experts_per_tok: int = 0
+    moe_capacity_factor: float = 0.0  # 0 = dropless per-expert loop
 
     def __post_init__(self):
         if self.num_kv_heads is None:
@@ -100,6 +101,7 @@ class ModelArgs:
             use_alibi=bool(attn.get("alibi", False)),
             num_local_experts=int(dims.get("num_local_experts", 0) or 0),
             num_experts_per_tok=int(dims.get("num_experts_per_tok", 0) or 0),
+            moe_capacity_factor=float(dims.get("moe_capacity_factor", 0.0) or 0.0),
         )
 
 
@@ -281,6 +283,7 @@ class MoE(nn.Module):
         nn.init.normal_(self.w_gate_up, mean=0.0, std=std)
         nn.init.normal_(self.w_down, mean=0.0, std=std)
         self.aux_loss: Optional[torch.Tensor] = None
+        self.capacity_factor = float(getattr(args, "moe_capacity_factor", 0.0) or 0.0)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         # expert parallelism (parallel/tp.py _shard_experts): activations
@@ -311,19 +314,61 @@ class MoE(nn.Module):
             self.aux_loss = self.aux_loss / self._ep_world
         n_local = self.w_gate_up.shape[0]
         e0 = self._ep_rank * n_local if ep else 0
-        out = torch.zeros_like(xf)
-        for el in range(n_local):
-            e = e0 + el
-            rows, slot = (idx == e).nonzero(as_tuple=True)
-            if rows.numel() == 0:
-                continue
-            toks = xf.index_select(0, rows)
-            h = swiglu(toks @ self.w_gate_up[el].t())
-            y = (h @ self.w_down[el].t()) * gates[rows, slot].unsqueeze(-1)
-            out.index_add_(0, rows, y.to(out.dtype))
+        if self.capacity_factor > 0:
+            out = self._forward_capacity(xf, gates, idx, e0, n_local)
+        else:
+            out = torch.zeros_like(xf)
+            for el in range(n_local):
+                e = e0 + el
+                rows, slot = (idx == e).nonzero(as_tuple=True)
+                if rows.numel() == 0:
+                    continue
+                toks = xf.index_select(0, rows)
+                h = swiglu(toks @ self.w_gate_up[el].t())
+                y = (h @ self.w_down[el].t()) * gates[rows, slot].unsqueeze(-1)
+                out.index_add_(0, rows, y.to(out.dtype))
         if ep:
             return reduce_from_tp(out).reshape(shape)
         return out.reshape(shape)
+
+    def _forward_capacity(self, xf: torch.Tensor, gates: torch.Tensor,
+                          idx: torch.Tensor, e0: int, n_local: int) -> torch.Tensor:
+        """Capacity-bound dispatch (Switch-style): tokens are permuted into
+        per-expert slots of fixed capacity C and the expert FFNs run as ONE
+        grouped GEMM pair (torch.bmm over [n_local, C, H] on hipBLASLt's
+        batched path) instead of a per-expert python loop; tokens beyond an
+        expert's capacity are dropped (they keep their other top-k routes).
+        Each EP rank dispatches only to its local experts; the surrounding
+        reduce_from_tp sums the per-rank partials, so EP stays exact.
+        Beyond reference parity (config stubs only at
+        /root/reference/models/llama.py:40-41)."""
+        n = xf.shape[0]
+        C = max(1, int(math.ceil(self.capacity_factor * n * self.top_k / self.num_experts)))
+        dev = xf.device
+        # flat routing pairs (token t, slot s) -> expert idx[t, s]
+        ef = idx.reshape(-1)  # [n*k]
+        # position of each pair within its expert's queue (stable arrival order)
+        order = torch.argsort(ef, stable=True)
+        ranks = torch.empty_like(order)
+        ranks[order] = torch.arange(ef.numel(), device=dev) - torch.searchsorted(
+            ef[order], torch.arange(self.num_experts, device=dev), side="left"
+        ).index_select(0, ef[order])
+        local = (ef >= e0) & (ef < e0 + n_local) & (ranks < C)
+        pair = local.nonzero(as_tuple=True)[0]           # kept (t,s) pairs
+        tok = pair // self.top_k
+        slot = pair % self.top_k
+        el = ef.index_select(0, pair) - e0               # local expert id
+        dst = el * C + ranks.index_select(0, pair)       # slot in the buffer
+        H = xf.shape[-1]
+        buf = torch.zeros(n_local * C, H, dtype=xf.dtype, device=dev)
+        buf.index_copy_(0, dst, xf.index_select(0, tok))
+        h = swiglu(torch.bmm(buf.reshape(n_local, C, H),
+                             self.w_gate_up.transpose(1, 2)))
+        y = torch.bmm(h, self.w_down.transpose(1, 2)).reshape(n_local * C, H)
+        g = gates[tok, slot].unsqueeze(-1)
+        out = torch.zeros_like(xf)
+        out.index_add_(0, tok, (y.index_select(0, dst) * g).to(out.dtype))
+        return out
 
 
 class TransformerBlock(nn.Module):

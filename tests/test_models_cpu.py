@@ -141,3 +141,52 @@ def test_moe_trainer_end_to_end(tmp_path):
     w1 = t.model.layers[0].mlp.w_gate_up.detach()
     w2 = t2.model.layers[0].mlp.w_gate_up.detach()
     assert torch.equal(w1, w2)
+
+
+def test_moe_capacity_dispatch_matches_dropless_at_high_capacity():
+    """capacity_factor large enough that nothing drops == the dropless
+    per-expert-loop path exactly (same weights, same routing)."""
+    import torch
+    from mlx_cuda_distributed_pretraining_amd.models.llama import MoE, ModelArgs
+
+    torch.manual_seed(1)
+    args = ModelArgs(hidden_size=32, intermediate_size=48, num_layers=1,
+                     num_heads=2, num_kv_heads=2, vocab_size=50,
+                     max_position_embeddings=32,
+                     num_local_experts=4, num_experts_per_tok=2)
+    moe = MoE(args)
+    x = torch.randn(3, 8, 32)
+    y_loop = moe(x)
+    moe.capacity_factor = 16.0  # C >= every expert's queue -> no drops
+    y_cap = moe(x)
+    assert torch.allclose(y_loop, y_cap, atol=1e-5), \
+        (y_loop - y_cap).abs().max().item()
+    # grads flow through the grouped-bmm path
+    y_cap.pow(2).mean().backward()
+    assert moe.w_gate_up.grad is not None and moe.router.weight.grad is not None
+
+
+def test_moe_capacity_dispatch_drops_overflow():
+    """With a tight capacity factor, overloaded experts drop their overflow
+    tokens (Switch semantics); output stays finite and differs from the
+    dropless path only on dropped routes."""
+    import torch
+    from mlx_cuda_distributed_pretraining_amd.models.llama import MoE, ModelArgs
+
+    torch.manual_seed(2)
+    args = ModelArgs(hidden_size=32, intermediate_size=48, num_layers=1,
+                     num_heads=2, num_kv_heads=2, vocab_size=50,
+                     max_position_embeddings=64,
+                     num_local_experts=4, num_experts_per_tok=2,
+                     moe_capacity_factor=0.5)
+    moe = MoE(args)
+    x = torch.randn(2, 32, 32)
+    y = moe(x)
+    assert torch.isfinite(y).all()
+    # capacity C = ceil(0.5 * 64*2 / 4) = 16 slots/expert; with 64 tokens x2
+    # routes and any imbalance, some routes MUST drop -> less total output
+    # mass than the dropless path
+    moe2_out = None
+    moe.capacity_factor = 0.0
+    moe2_out = moe(x)
+    assert not torch.allclose(y, moe2_out), "tight capacity dropped nothing"

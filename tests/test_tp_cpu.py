@@ -646,6 +646,7 @@ def _ep_worker(rank, world, port, q):
         args = _args()
         args.num_local_experts = 4
         args.num_experts_per_tok = 2
+        args.moe_capacity_factor = float(os.environ.get("TEST_MOE_CAP", "0"))
         torch.manual_seed(0)
         model = Model(args)
         broadcast_module(model)
@@ -1069,3 +1070,42 @@ def test_tp_validation_errors():
     m2 = Model(args2)
     with pytest.raises(ValueError, match="num_local_experts"):
         apply_tensor_parallel(m2, 0, 2)
+
+
+@pytest.mark.timeout(300)
+def test_expert_parallel_capacity_dispatch_matches_single_process():
+    """Capacity-bound dispatch under EP: per-expert queue positions are
+    computed from the replicated router output, so each rank's kept set is
+    the exact local slice of the single-process kept set — forward and
+    grads must match to numerical noise."""
+    args = _args()
+    args.num_local_experts = 4
+    args.num_experts_per_tok = 2
+    args.moe_capacity_factor = 1.0  # tight: drops occur and must agree
+    torch.manual_seed(0)
+    model = Model(args)
+    batch = _batch()
+    logits = model(batch[:, :-1])
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, 67), batch[:, 1:].reshape(-1)
+    ) + 0.01 * model.aux_loss
+    loss.backward()
+    ref_logits = logits.detach()
+
+    os.environ["TEST_MOE_CAP"] = "1.0"
+    try:
+        ctx = mp.get_context("spawn")
+        q = ctx.SimpleQueue()
+        port = _free_port()
+        procs = [ctx.Process(target=_ep_worker, args=(r, 2, port, q)) for r in range(2)]
+        for p in procs:
+            p.start()
+        res = [q.get(), q.get()]
+        for p in procs:
+            p.join(240)
+            assert p.exitcode == 0
+    finally:
+        os.environ.pop("TEST_MOE_CAP", None)
+    for r in res:
+        assert ref_logits.numpy() == pytest.approx(r["logits"], abs=2e-4), \
+            f"rank {r['rank']} capacity-EP logits diverged"
