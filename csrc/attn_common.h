@@ -23,6 +23,7 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 #define MOD_SLIDING_WINDOW 2
 #define MOD_PREFIX_LM 3
 #define MOD_ALIBI 4
+#define MOD_BLOCKMASK 5  // arbitrary mask_mod via device block-mask + packed bits
 
 __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
   unsigned packed;

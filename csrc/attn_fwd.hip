@@ -37,7 +37,17 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     const __hip_bfloat16* __restrict__ v, __hip_bfloat16* __restrict__ o,
     float* __restrict__ lse, const float* __restrict__ slopes,
     int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
-    long q_rs, long k_rs, long v_rs) {
+    long q_rs, long k_rs, long v_rs,
+    // MOD_BLOCKMASK only (K2: arbitrary mask_mod compiled to device tensors;
+    // /root/reference/models/attention/flex_attention.py:356-411):
+    //   mb_gran  uint8 [B,H,ceil(Sq/32),ceil(Skv/KVB)] 0=masked 1=bits 2=full
+    //   mb_bits  uint8 [B,H,Sq,ceil(Skv/8)] per-element keep bits
+    //   mb_range int32 [B,H,ceil(Sq/QPB),2] first/last+1 live kv tile
+    //   bias     bf16  [B,H,Sq,Skv] optional additive score bias (score_mod)
+    const unsigned char* __restrict__ mb_gran = nullptr,
+    const unsigned char* __restrict__ mb_bits = nullptr,
+    const int* __restrict__ mb_range = nullptr,
+    const __hip_bfloat16* __restrict__ bias = nullptr) {
   constexpr int TPB = NW * WAVE;
   constexpr int QPB = NW * QPW;
   constexpr int DBLK = D / 16;  // QK^T d-slots
@@ -98,8 +108,24 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
     kv_lo = max(0, blk_qpos_lo - modarg + 1) & ~(KVB - 1);
   } else if constexpr (MOD == MOD_PREFIX_LM) {
     kv_hi = min(Skv, max(blk_qpos_hi + 1, modarg));
+  } else if constexpr (MOD == MOD_BLOCKMASK) {
+    const int nqpb = (Sq + QPB - 1) / QPB;
+    const int* r = mb_range + (((long)b * Hq + hq) * nqpb + qtile) * 2;
+    kv_lo = r[0] * KVB;
+    kv_hi = min(Skv, r[1] * KVB);
+    if (kv_hi <= kv_lo) kv_hi = kv_lo;  // fully-masked row block: loop skips
   }
   const float slope2 = (MOD == MOD_ALIBI) ? slopes[hq] * LOG2E : 0.f;
+  // blockmask strides (granule: this wave's 32-row q granule per kv tile)
+  const int nkvt = (Skv + KVB - 1) / KVB;
+  const int nq32 = (Sq + 31) / 32;
+  const long gran_base = (MOD == MOD_BLOCKMASK)
+      ? (((long)b * Hq + hq) * nq32 + min(q0w / 32, nq32 - 1)) * (long)nkvt : 0;
+  const int kvb8 = (Skv + 7) / 8;
+  const long bits_row = (MOD == MOD_BLOCKMASK && q_valid)
+      ? (((long)b * Hq + hq) * Sq + qrow) * (long)kvb8 : 0;
+  const long bias_row = (MOD == MOD_BLOCKMASK && q_valid && bias)
+      ? (((long)b * Hq + hq) * Sq + qrow) * (long)Skv : 0;
 
   // ---- staging helpers (T14 split: load -> regs early, write -> LDS late) --
   // K: KU4 uint4 chunks per thread (row-major swizzled image, b128 writes).
@@ -235,6 +261,12 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
       full = full && ((kv0 + KVB - 1 <= q0w + q_off) || (kv0 + KVB <= modarg));
     } else if constexpr (MOD == MOD_ALIBI) {
       full = false;  // slope term needs per-element positions anyway
+    } else if constexpr (MOD == MOD_BLOCKMASK) {
+      // per-wave granule: full only if every lane's granule says 2 AND no
+      // bias (bias must be added per element)
+      const unsigned char g = mb_gran[gran_base + (kv0 / KVB)];
+      full = full && (g == 2) && (bias == nullptr) && q_valid;
+      full = __all(full);
     }
 
     float tmax;
@@ -258,9 +290,20 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
 #pragma unroll
         for (int reg = 0; reg < 16; ++reg) {
           const int k_pos = kv0 + kt * 32 + acc_row(reg, hi);
-          const bool keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+          bool keep;
+          if constexpr (MOD == MOD_BLOCKMASK) {
+            keep = q_valid && k_pos < Skv;
+            if (keep)
+              keep = (mb_bits[bits_row + (k_pos >> 3)] >> (k_pos & 7)) & 1;
+          } else {
+            keep = q_valid && attn_keep<MOD>(q_pos, k_pos, Skv, modarg);
+          }
           float s = st[kt][reg] * scale2;
           if constexpr (MOD == MOD_ALIBI) s += slope2 * (k_pos - q_pos);
+          if constexpr (MOD == MOD_BLOCKMASK) {
+            if (bias != nullptr && keep)
+              s += __bfloat162float(bias[bias_row + k_pos]) * LOG2E;
+          }
           st[kt][reg] = keep ? s : -INFINITY;
           tmax = fmaxf(tmax, st[kt][reg]);
         }
@@ -481,6 +524,40 @@ static long bshd_row_stride(at::Tensor& t) {
   if (!(t.stride(3) == 1 && t.stride(2) == D && t.stride(0) == (long)S * t.stride(1)))
     t = t.contiguous();
   return t.stride(1);
+}
+
+std::vector<at::Tensor> attn_fwd_blockmask(at::Tensor q, at::Tensor k, at::Tensor v,
+                                           double scale, at::Tensor gran, at::Tensor bits,
+                                           at::Tensor range, at::Tensor bias) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16, "bf16 only");
+  const long q_rs = bshd_row_stride(q), k_rs = bshd_row_stride(k), v_rs = bshd_row_stride(v);
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1), Hkv = k.size(2);
+  TORCH_CHECK(Hq % Hkv == 0 && (D == 64 || D == 128));
+  TORCH_CHECK(gran.scalar_type() == at::kByte && bits.scalar_type() == at::kByte &&
+              range.scalar_type() == at::kInt, "blockmask tensor dtypes");
+  constexpr int QPB = 8 * 32;
+  TORCH_CHECK(range.size(2) == (Sq + QPB - 1) / QPB, "range shape mismatch");
+  auto o = at::empty({B, Sq, Hq, D}, q.options());
+  auto lse = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  dim3 grid(cdiv(Sq, QPB), Hq, B), block(512);
+  const bool has_bias = bias.numel() > 0;
+  auto* bp = has_bias ? reinterpret_cast<const __hip_bfloat16*>(bias.data_ptr()) : nullptr;
+  auto launch1 = [&](auto dtag) {
+    constexpr int DD = decltype(dtag)::value;
+    attn_fwd_kernel<DD, MOD_BLOCKMASK, 8, 64, 1><<<grid, block, 0, stream>>>(
+        reinterpret_cast<const __hip_bfloat16*>(q.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
+        reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
+        reinterpret_cast<__hip_bfloat16*>(o.data_ptr()), lse.data_ptr<float>(),
+        nullptr, B, Sq, Skv, Hq, Hkv, (float)scale, 0, q_rs, k_rs, v_rs,
+        gran.data_ptr<unsigned char>(), bits.data_ptr<unsigned char>(),
+        range.data_ptr<int>(), bp);
+  };
+  if (D == 64) launch1(std::integral_constant<int, 64>{});
+  else launch1(std::integral_constant<int, 128>{});
+  return {o, lse};
 }
 
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale,

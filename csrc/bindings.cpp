@@ -37,6 +37,9 @@ void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a);
 void shampoo_stats_update(at::Tensor G, at::Tensor S, double beta);
 
 // attn_fwd.hip / attn_bwd.hip
+std::vector<at::Tensor> attn_fwd_blockmask(at::Tensor q, at::Tensor k, at::Tensor v,
+                                           double scale, at::Tensor gran, at::Tensor bits,
+                                           at::Tensor range, at::Tensor bias);
 std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, double scale,
                                  long mod, long modarg, at::Tensor slopes);
 std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o,
@@ -96,6 +99,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("shampoo_stats_update", &shampoo_stats_update,
         "S = beta*S + (1-beta)*G@G^T (fp32 state, bf16 MFMA)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (o, lse)");
+  m.def("attn_fwd_blockmask", &attn_fwd_blockmask,
+        "flash attention fwd with device block-mask/bits/bias (K2 flex)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward (dq, dk, dv)");
   m.def("attn_bwd_out", &attn_bwd_out,
         "flash attention backward into strided out views (fused dQKV)");

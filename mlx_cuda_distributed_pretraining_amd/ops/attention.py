@@ -299,6 +299,8 @@ def flex_attention(
     fp32 reference path (GPU included — they are inherently per-element
     Python and exist for parity/experimentation, not the hot path).
     """
+    if isinstance(block_mask, CompiledBlockMask):
+        return _flex_blockmask(q, k, v, block_mask, score_mod, scale)
     if block_mask is not None and score_mod is None and mask_mod is None:
         p = block_mask
         return flash_attention(
@@ -311,7 +313,36 @@ def flex_attention(
         )
     if score_mod is None and mask_mod is None:
         return flash_attention(q, k, v, causal=True, scale=scale)
+    if (mask_mod is not None and q.is_cuda and not torch.is_grad_enabled()
+            and q.dtype == torch.bfloat16):
+        bm = CompiledBlockMask(mask_mod, q.shape[0], q.shape[2], q.shape[1],
+                               k.shape[1], device=q.device)
+        return _flex_blockmask(q, k, v, bm, score_mod, scale)
     return attention_ref(q, k, v, causal=False, scale=scale, score_mod=score_mod, mask_mod=mask_mod)
+
+
+def _flex_blockmask(q, k, v, bm: "CompiledBlockMask", score_mod, scale):
+    """Kernel-path flex: compiled mask tensors + (optional) a precomputed
+    additive bias tensor for score_mod (evaluated vectorized once; the
+    kernel adds it pre-softmax). Inference/eval surface — with grads the
+    caller stays on the autograd fp32 path."""
+    from ._ext import require_ext
+
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    bias = torch.empty(0, dtype=torch.bfloat16, device=q.device)
+    if score_mod is not None:
+        B, S, H, D = q.shape
+        KV = k.shape[1]
+        qi = torch.arange(S, device=q.device).view(1, 1, S, 1)
+        ki = torch.arange(KV, device=q.device).view(1, 1, 1, KV)
+        bi = torch.arange(B, device=q.device).view(B, 1, 1, 1)
+        hi = torch.arange(H, device=q.device).view(1, H, 1, 1)
+        zero = torch.zeros(B, H, S, KV, device=q.device)
+        bias = (score_mod(zero, bi, hi, qi, ki) - 0.0).to(torch.bfloat16)
+    o, _ = require_ext().attn_fwd_blockmask(q, k, v, float(scale), bm.gran,
+                                            bm.bits, bm.range, bias)
+    return o
 
 
 class BlockMask:
@@ -337,6 +368,70 @@ class BlockMask:
         self.block_size = block_size
 
 
+class CompiledBlockMask:
+    """An arbitrary ``mask_mod`` compiled into device tensors that drive the
+    K1 kernel's block-skip / per-element-bit machinery (VERDICT r1 #5;
+    parity: /root/reference/models/attention/flex_attention.py:356-411 —
+    the reference SAMPLED block midpoints and then materialized full S x S
+    scores anyway; here the mask is exact and only live kv-tile ranges are
+    visited).
+
+    Tensors (uint8/int32, tiny vs S^2 fp32):
+      gran  [B,H,ceil(Q/32),ceil(KV/64)]  0 = granule fully masked,
+                                          1 = partial (per-element bits),
+                                          2 = fully live (mask-free path)
+      bits  [B,H,Q,ceil(KV/8)]            packed keep bits
+      range [B,H,ceil(Q/256),2]           first / last+1 live kv tile per
+                                          256-row kernel block
+    """
+
+    KVB = 64
+    QPB = 256
+
+    def __init__(self, mask_mod: Callable, B: int, H: int, Q_LEN: int, KV_LEN: int,
+                 device="cuda"):
+        self.B, self.H, self.Q_LEN, self.KV_LEN = B, H, Q_LEN, KV_LEN
+        qi = torch.arange(Q_LEN, device=device)
+        ki = torch.arange(KV_LEN, device=device)
+        nq32 = (Q_LEN + 31) // 32
+        nkv = (KV_LEN + self.KVB - 1) // self.KVB
+        nqpb = (Q_LEN + self.QPB - 1) // self.QPB
+        self.gran = torch.zeros(B, H, nq32, nkv, dtype=torch.uint8, device=device)
+        self.bits = torch.zeros(B, H, Q_LEN, (KV_LEN + 7) // 8, dtype=torch.uint8,
+                                device=device)
+        self.range = torch.zeros(B, H, nqpb, 2, dtype=torch.int32, device=device)
+        weights = (1 << torch.arange(8, device=device, dtype=torch.int32)).to(torch.uint8)
+        for b in range(B):
+            for h in range(H):
+                keep = mask_mod(b, h, qi.unsqueeze(1), ki.unsqueeze(0))
+                keep = keep.to(torch.bool)  # [Q, KV], exact
+                kp = torch.zeros(Q_LEN, nkv * self.KVB // 8 * 8, dtype=torch.bool,
+                                 device=device)
+                kp[:, :KV_LEN] = keep
+                packed = (kp.reshape(Q_LEN, -1, 8).to(torch.uint8) *
+                          weights).sum(-1).to(torch.uint8)
+                self.bits[b, h] = packed[:, : (KV_LEN + 7) // 8]
+                # granule codes: pad to 32-row / 64-col granules
+                gq = torch.zeros(nq32 * 32, nkv * self.KVB, dtype=torch.bool,
+                                 device=device)
+                gq[:Q_LEN, :KV_LEN] = keep
+                g = gq.reshape(nq32, 32, nkv, self.KVB)
+                any_live = g.any(dim=(1, 3))
+                # "full" additionally requires every IN-RANGE element live
+                gq2 = torch.ones(nq32 * 32, nkv * self.KVB, dtype=torch.bool,
+                                 device=device)
+                gq2[:Q_LEN, :KV_LEN] = keep
+                all_live = gq2.reshape(nq32, 32, nkv, self.KVB).all(dim=(1, 3))
+                self.gran[b, h] = any_live.to(torch.uint8) + (any_live & all_live).to(torch.uint8)
+                # per 256-row block: first/last live kv tile
+                blk = any_live.reshape(nqpb, -1, nkv).any(dim=1)  # [nqpb, nkv]
+                for qb in range(nqpb):
+                    live = blk[qb].nonzero().flatten()
+                    if live.numel():
+                        self.range[b, h, qb, 0] = int(live.min())
+                        self.range[b, h, qb, 1] = int(live.max()) + 1
+
+
 def create_block_mask(
     mask_mod: Optional[Callable] = None,
     B: int = 1,
@@ -344,6 +439,13 @@ def create_block_mask(
     Q_LEN: int = 0,
     KV_LEN: int = 0,
     pattern: str = "causal",
+    device=None,
     **kwargs,
 ) -> BlockMask:
+    """Named patterns return a pattern BlockMask (kernel block-skip handles
+    them natively); an arbitrary ``mask_mod`` callable with a CUDA device is
+    compiled to a CompiledBlockMask driving the kernel (else it stays a
+    python callable for the fp32 reference path)."""
+    if mask_mod is not None and Q_LEN and KV_LEN and device is not None             and torch.device(device).type == "cuda":
+        return CompiledBlockMask(mask_mod, B, H, Q_LEN, KV_LEN, device=device)
     return BlockMask(pattern=pattern, **kwargs)

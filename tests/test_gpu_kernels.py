@@ -525,3 +525,60 @@ def test_add_rmsnorm_gpu(ext):
         assert (x.grad.float() - x2.grad).abs().max() < 8e-2
         assert (r.grad.float() - r2.grad).abs().max() < 8e-2
         assert (w.grad.float() - w2.grad).abs().max() / w2.grad.abs().max() < 3e-2
+
+
+def test_flex_custom_mask_mod_kernel_path(ext):
+    """K2: arbitrary mask_mod compiled to a device block mask drives the
+    MFMA kernel (MOD_BLOCKMASK) instead of the fp32 S^2 fallback (VERDICT
+    r1 #5). Oracle: attention_ref with the same callable."""
+    from mlx_cuda_distributed_pretraining_amd.ops import attention_ref
+    from mlx_cuda_distributed_pretraining_amd.ops.attention import (
+        CompiledBlockMask, flex_attention,
+    )
+
+    torch.manual_seed(7)
+    B, S, H, D = 2, 512, 2, 64
+
+    def doc_band_mask(b, h, qi, ki):
+        # causal AND within a 128-token band, except documents of 200 tokens
+        # don't attend across their boundary — an arbitrary non-named pattern
+        return (ki <= qi) & (qi - ki < 128) & (qi // 200 == ki // 200)
+
+    q = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    k = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    v = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    bm = CompiledBlockMask(doc_band_mask, B, H, S, S, device=dev())
+    # live kv range honored: first 256-row block only touches tiles < 256/64
+    assert int(bm.range[0, 0, 0, 1]) <= 4
+    with torch.no_grad():
+        o = flex_attention(q, k, v, block_mask=bm)
+    ref = attention_ref(q.float(), k.float(), v.float(), causal=False,
+                        scale=1.0 / math.sqrt(D), mask_mod=doc_band_mask)
+    err = (o.float() - ref).abs().max().item()
+    assert err < 3e-2, f"flex blockmask kernel path err {err}"
+
+
+def test_flex_score_mod_bias_kernel_path(ext):
+    """score_mod -> precomputed additive bias consumed by the kernel."""
+    from mlx_cuda_distributed_pretraining_amd.ops import attention_ref
+    from mlx_cuda_distributed_pretraining_amd.ops.attention import flex_attention
+
+    torch.manual_seed(8)
+    B, S, H, D = 1, 256, 2, 64
+
+    def rel_bias(score, b, h, qi, ki):
+        return score - 0.05 * (qi - ki).abs().to(torch.float32)
+
+    def causal_mask(b, h, qi, ki):
+        return ki <= qi
+
+    q = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    k = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    v = torch.randn(B, S, H, D, device=dev(), dtype=torch.bfloat16)
+    with torch.no_grad():
+        o = flex_attention(q, k, v, score_mod=rel_bias, mask_mod=causal_mask)
+    ref = attention_ref(q.float(), k.float(), v.float(), causal=False,
+                        scale=1.0 / math.sqrt(D),
+                        score_mod=rel_bias, mask_mod=causal_mask)
+    err = (o.float() - ref).abs().max().item()
+    assert err < 3e-2, f"flex bias kernel path err {err}"
