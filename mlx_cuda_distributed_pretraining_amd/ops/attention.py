@@ -58,7 +58,9 @@ def attention_ref(
     q: [B, S, Hq, D]; k, v: [B, Skv, Hkv, D] (BSHD).
     score_mod(scores[B,H,S,S], b_idx, h_idx, q_idx, kv_idx) -> scores, applied
     to the scaled scores (FlexAttention semantics, vectorized over the full
-    score tensor). mask_mod(b, h, q_idx, kv_idx) -> bool keep-mask.
+    score tensor). mask_mod(b, h, q_idx, kv_idx) -> bool keep-mask. Index
+    tensors arrive BROADCASTABLE (q [S,1] x kv [1,Skv]); write mods as plain
+    expressions over them.
     """
     B, S, Hq, D = q.shape
     Hkv = k.shape[2]
@@ -90,14 +92,19 @@ def attention_ref(
         bias = alibi_slopes.float().view(1, Hq, 1, 1) * (kv_idx - q_idx).float().view(1, 1, S, Skv)
         scores = scores + bias
         keep = keep & (kv_idx <= q_idx)
+    # mods receive broadcastable index tensors (q [S,1], kv [1,Skv]) so a
+    # plain expression like (ki <= qi) evaluates as the outer grid — the same
+    # convention CompiledBlockMask compiles with (vectorized FlexAttention
+    # semantics).
     if score_mod is not None:
-        b_idx = torch.arange(B, device=q.device)
-        h_idx = torch.arange(Hq, device=q.device)
-        scores = score_mod(scores, b_idx, h_idx, q_idx.view(-1), kv_idx.view(-1))
+        b_idx = torch.arange(B, device=q.device).view(B, 1, 1, 1)
+        h_idx = torch.arange(Hq, device=q.device).view(1, Hq, 1, 1)
+        scores = score_mod(scores, b_idx, h_idx, q_idx.view(1, 1, S, 1),
+                           kv_idx.view(1, 1, 1, Skv))
     if mask_mod is not None:
         b_idx = torch.arange(B, device=q.device)
         h_idx = torch.arange(Hq, device=q.device)
-        keep = keep & mask_mod(b_idx, h_idx, q_idx.view(-1), kv_idx.view(-1))
+        keep = keep & mask_mod(b_idx, h_idx, q_idx, kv_idx)
     scores = scores.masked_fill(~keep.view(1, 1, S, Skv), float("-inf"))
     lse = torch.logsumexp(scores, dim=-1)
     p = torch.softmax(scores, dim=-1)
