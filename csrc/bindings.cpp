@@ -16,6 +16,9 @@ at::Tensor swiglu_fwd(at::Tensor gu);
 std::vector<at::Tensor> swiglu_fwd_amax(at::Tensor gu, bool with_amax);
 at::Tensor swiglu_bwd(at::Tensor gu, at::Tensor dy);
 // cross_entropy.hip
+std::vector<at::Tensor> ce_vp_stats(at::Tensor logits, at::Tensor targets, long v0, long ignore_index);
+at::Tensor ce_vp_sumexp(at::Tensor logits, at::Tensor m_group);
+at::Tensor ce_vp_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor scale, long v0, long ignore_index);
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets, long ignore_index);
 at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Tensor scale,
                   long ignore_index);
@@ -89,6 +92,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_fwd_amax", &swiglu_fwd_amax, "swiglu fwd emitting |out| amax bits");
   m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");
   m.def("ce_fwd", &ce_fwd, "fused cross-entropy forward (loss_sum, ntok, lse)");
+  m.def("ce_vp_stats", &ce_vp_stats, "vocab-parallel CE pass 1 (m_local, tgt_local)");
+  m.def("ce_vp_sumexp", &ce_vp_sumexp, "vocab-parallel CE pass 2 (sum exp)");
+  m.def("ce_vp_bwd", &ce_vp_bwd, "vocab-parallel CE backward (dlogits)");
   m.def("ce_bwd", &ce_bwd, "fused cross-entropy backward (dlogits)");
   m.def("sumsq", &sumsq, "sum of squares -> f32 scalar");
   m.def("adamw_step", &adamw_step, "fused AdamW over flat buffers");

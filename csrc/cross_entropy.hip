@@ -86,6 +86,91 @@ __global__ void ce_bwd_kernel(const T* __restrict__ logits, const long* __restri
   }
 }
 
+// ---- vocab-parallel CE (TP lm-head shards; parallel/tp.py) ----
+// Pass 1: per-row local max + the target logit if it lives in this shard.
+template <int BLOCK>
+__global__ void ce_vp_stats_kernel(const __hip_bfloat16* __restrict__ logits,
+                                   const long* __restrict__ targets,
+                                   float* __restrict__ m_out, float* __restrict__ tgt_out,
+                                   long rows, int V, long v0, long ignore_index) {
+  __shared__ float scratch[BLOCK / WAVE];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const __hip_bfloat16* lr = logits + row * (long)V;
+    float m = -INFINITY;
+    const int VV = V / 8;
+    const uint4* lv = reinterpret_cast<const uint4*>(lr);
+    for (int i = threadIdx.x; i < VV; i += BLOCK) {
+      U4 u; u.u = lv[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) m = fmaxf(m, bf16_bits_to_f32(u.s[j]));
+    }
+    for (int i = VV * 8 + threadIdx.x; i < V; i += BLOCK) m = fmaxf(m, to_f32(lr[i]));
+    m = block_reduce_max<BLOCK>(m, scratch);
+    if (threadIdx.x == 0) {
+      m_out[row] = m;
+      const long t = targets[row];
+      const long tl = t - v0;
+      tgt_out[row] = (t != ignore_index && tl >= 0 && tl < V) ? to_f32(lr[tl]) : 0.f;
+    }
+  }
+}
+
+// Pass 2 (after the group max all-reduce): sum exp(l - m_group).
+template <int BLOCK>
+__global__ void ce_vp_sumexp_kernel(const __hip_bfloat16* __restrict__ logits,
+                                    const float* __restrict__ m_group,
+                                    float* __restrict__ se_out, long rows, int V) {
+  __shared__ float scratch[BLOCK / WAVE];
+  for (long row = blockIdx.x; row < rows; row += gridDim.x) {
+    const __hip_bfloat16* lr = logits + row * (long)V;
+    const float m = m_group[row];
+    float s = 0.f;
+    const int VV = V / 8;
+    const uint4* lv = reinterpret_cast<const uint4*>(lr);
+    for (int i = threadIdx.x; i < VV; i += BLOCK) {
+      U4 u; u.u = lv[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s += __expf(bf16_bits_to_f32(u.s[j]) - m);
+    }
+    for (int i = VV * 8 + threadIdx.x; i < V; i += BLOCK) s += __expf(to_f32(lr[i]) - m);
+    s = block_reduce_sum<BLOCK>(s, scratch);
+    if (threadIdx.x == 0) se_out[row] = s;
+  }
+}
+
+// Backward: dlogits = (exp(l - lse_group) - onehot(t - v0)) * (*scale), rows
+// with ignored targets get 0 (scale = upstream_grad / ntok, device scalar).
+__global__ void ce_vp_bwd_kernel(const __hip_bfloat16* __restrict__ logits,
+                                 const long* __restrict__ targets,
+                                 const float* __restrict__ lse,
+                                 const float* __restrict__ scale,
+                                 __hip_bfloat16* __restrict__ dlogits,
+                                 long rows, int V, long v0, long ignore_index) {
+  const float sc = *scale;
+  const int VV = V / 8;
+  for (long idx = blockIdx.x * (long)blockDim.x + threadIdx.x; idx < rows * (long)VV;
+       idx += gridDim.x * (long)blockDim.x) {
+    const long row = idx / VV;
+    const int col = (int)(idx % VV) * 8;
+    const long t = targets[row];
+    const float l = lse[row];
+    const __hip_bfloat16* lr = logits + row * (long)V + col;
+    __hip_bfloat16* dr = dlogits + row * (long)V + col;
+    if (t == ignore_index) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) from_f32(&dr[j], 0.f);
+      continue;
+    }
+    const long tl = t - v0;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float p = __expf(to_f32(lr[j]) - l);
+      if (col + j == (int)tl) p -= 1.f;
+      from_f32(&dr[j], p * sc);
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets, long ignore_index) {
@@ -136,4 +221,49 @@ at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse, at::Ten
     }
   });
   return dlogits;
+}
+
+std::vector<at::Tensor> ce_vp_stats(at::Tensor logits, at::Tensor targets, long v0,
+                                    long ignore_index) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous() && logits.dim() == 2 &&
+              logits.scalar_type() == at::kBFloat16);
+  const long rows = logits.size(0);
+  const int V = logits.size(1);
+  TORCH_CHECK(V % 8 == 0, "ce_vp: vocab shard must be a multiple of 8");
+  auto m = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto tgt = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  const int grid = (int)std::min<long>(rows, 2048);
+  ce_vp_stats_kernel<256><<<grid, 256, 0, stream>>>(
+      reinterpret_cast<const __hip_bfloat16*>(logits.data_ptr()),
+      targets.data_ptr<long>(), m.data_ptr<float>(), tgt.data_ptr<float>(),
+      rows, V, v0, ignore_index);
+  return {m, tgt};
+}
+
+at::Tensor ce_vp_sumexp(at::Tensor logits, at::Tensor m_group) {
+  const long rows = logits.size(0);
+  const int V = logits.size(1);
+  auto se = at::empty({rows}, logits.options().dtype(at::kFloat));
+  auto stream = at::cuda::getCurrentHIPStream();
+  const int grid = (int)std::min<long>(rows, 2048);
+  ce_vp_sumexp_kernel<256><<<grid, 256, 0, stream>>>(
+      reinterpret_cast<const __hip_bfloat16*>(logits.data_ptr()),
+      m_group.data_ptr<float>(), se.data_ptr<float>(), rows, V);
+  return se;
+}
+
+at::Tensor ce_vp_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                     at::Tensor scale, long v0, long ignore_index) {
+  const long rows = logits.size(0);
+  const int V = logits.size(1);
+  auto dl = at::empty_like(logits);
+  auto stream = at::cuda::getCurrentHIPStream();
+  const long work = rows * (V / 8);
+  const int grid = (int)std::min<long>((work + 255) / 256, 4096);
+  ce_vp_bwd_kernel<<<grid, 256, 0, stream>>>(
+      reinterpret_cast<const __hip_bfloat16*>(logits.data_ptr()),
+      targets.data_ptr<long>(), lse.data_ptr<float>(), scale.data_ptr<float>(),
+      reinterpret_cast<__hip_bfloat16*>(dl.data_ptr()), rows, V, v0, ignore_index);
+  return dl;
 }

@@ -582,3 +582,49 @@ def test_flex_score_mod_bias_kernel_path(ext):
                         score_mod=rel_bias, mask_mod=causal_mask)
     err = (o.float() - ref).abs().max().item()
     assert err < 3e-2, f"flex bias kernel path err {err}"
+
+
+def test_vocab_parallel_ce_kernels(ext):
+    """ce_vp_* kernels: world-1 full-vocab path == fused_cross_entropy;
+    shard semantics (v0 offset, out-of-shard targets) checked by combining
+    two shard calls by hand (ROADMAP r1 #15 / VERDICT #10)."""
+    import torch.distributed as tdist
+
+    from mlx_cuda_distributed_pretraining_amd.ops._ext import require_ext
+    from mlx_cuda_distributed_pretraining_amd.ops.cross_entropy import fused_cross_entropy
+    from mlx_cuda_distributed_pretraining_amd.parallel.tp import (
+        vocab_parallel_cross_entropy,
+    )
+
+    e = require_ext()
+    torch.manual_seed(9)
+    N, V = 512, 1024
+    logits = torch.randn(N, V, device=dev(), dtype=torch.bfloat16, requires_grad=True)
+    tg = torch.randint(0, V, (N,), device=dev())
+    tg[::7] = -100
+    if not tdist.is_initialized():  # world-1: group reductions are no-ops
+        loss, ntok = vocab_parallel_cross_entropy(logits, tg, 0, ignore_index=-100)
+        l2 = logits.detach().clone().requires_grad_(True)
+        ref, ntok_ref = fused_cross_entropy(l2, tg, -100)
+        assert int(ntok) == int(ntok_ref)
+        assert abs(float(loss) - float(ref)) < 2e-3
+        loss.backward()
+        ref.backward()
+        gerr = (logits.grad.float() - l2.grad.float()).abs().max().item()
+        assert gerr < 1e-4, f"vp-ce grad err {gerr}"
+
+    # manual 2-shard combination equals full CE (shard algebra incl.
+    # out-of-shard targets)
+    lf = logits.detach()
+    half = V // 2
+    m0, t0 = e.ce_vp_stats(lf[:, :half].contiguous(), tg, 0, -100)
+    m1, t1 = e.ce_vp_stats(lf[:, half:].contiguous(), tg, half, -100)
+    m = torch.maximum(m0, m1)
+    se = (e.ce_vp_sumexp(lf[:, :half].contiguous(), m) +
+          e.ce_vp_sumexp(lf[:, half:].contiguous(), m))
+    lse = m + torch.log(se)
+    mask = tg != -100
+    loss_sh = (torch.where(mask, lse - (t0 + t1), torch.zeros_like(lse)).sum()
+               / mask.sum())
+    ref2, _ = fused_cross_entropy(lf.clone().requires_grad_(True), tg, -100)
+    assert abs(float(loss_sh) - float(ref2)) < 2e-3
