@@ -79,13 +79,14 @@ __device__ __forceinline__ bf16x8 img_frag(const __hip_bfloat16* img, int r0,
 
 // ---------------------------------------------------------------------------
 // C[M,N] = alpha * X[M,K] @ Y[N,K]^T + beta * E[M,N]
-template <bool HAS_E, typename TO = __hip_bfloat16>
+template <bool HAS_E, typename TO = __hip_bfloat16, bool PIPE3 = false>
 __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
     const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ Y,
     const TO* __restrict__ E, TO* __restrict__ C,
     int M, int N, int K, float alpha, float beta) {
   constexpr int TILE = BMN * BK;          // elements per image
-  __shared__ __hip_bfloat16 smem[2 * 2 * TILE];
+  constexpr int SLOTS = PIPE3 ? 3 : 2;
+  __shared__ __hip_bfloat16 smem[SLOTS * 2 * TILE];
 
   const int tm = blockIdx.x, tn = blockIdx.y;
   const int m0 = tm * BMN, n0 = tn * BMN;
@@ -101,16 +102,41 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x16{};
 
   const int ksteps = K / BK;
-  // prologue: stage tile 0 into buf 0
+  // prologue: stage tile 0 (and, 3-slot form, tile 1)
   glds_tile(X + (long)m0 * K, K, smem);
   glds_tile(Y + (long)n0 * K, K, smem + TILE);
-  __syncthreads();
+  if constexpr (PIPE3) {
+    if (ksteps > 1) {
+      glds_tile(X + (long)m0 * K + BK, K, smem + 2 * TILE);
+      glds_tile(Y + (long)n0 * K + BK, K, smem + 3 * TILE);
+    }
+  } else {
+    __syncthreads();
+  }
 
   for (int kt = 0; kt < ksteps; ++kt) {
-    const int buf = kt & 1;
-    if (kt + 1 < ksteps) {  // issue next-tile staging FIRST (T3 min-2-phase)
-      glds_tile(X + (long)m0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE);
-      glds_tile(Y + (long)n0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE + TILE);
+    const int buf = PIPE3 ? kt % 3 : (kt & 1);
+    if constexpr (PIPE3) {
+      // T3/T4 counted-vmcnt 3-slot ring: wait for tile kt's 8 glds (leave
+      // tile kt+1's 8 in flight), raw barrier (no vmcnt(0) drain), THEN
+      // issue tile kt+2 into the slot whose readers finished before the
+      // PREVIOUS barrier. Each wave's own pieces are covered by its own
+      // vmcnt; the barrier publishes them to the other waves.
+      if (kt + 1 < ksteps)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");  // tile kt landed
+      else
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // tail: nothing behind
+      __builtin_amdgcn_s_barrier();
+      if (kt + 2 < ksteps) {
+        const int ns = (kt + 2) % 3;
+        glds_tile(X + (long)m0 * K + (kt + 2) * BK, K, smem + ns * 2 * TILE);
+        glds_tile(Y + (long)n0 * K + (kt + 2) * BK, K, smem + ns * 2 * TILE + TILE);
+      }
+    } else {
+      if (kt + 1 < ksteps) {  // issue next-tile staging FIRST (T3 min-2-phase)
+        glds_tile(X + (long)m0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE);
+        glds_tile(Y + (long)n0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE + TILE);
+      }
     }
     const __hip_bfloat16* ax = smem + buf * 2 * TILE;
     const __hip_bfloat16* by = ax + TILE;
@@ -129,8 +155,11 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af[i], bf[j], acc[i][j], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
-    __syncthreads();  // drains the in-flight glds (vmcnt(0)) + buffer swap
+    if constexpr (!PIPE3) {
+      __syncthreads();  // drains the in-flight glds (vmcnt(0)) + buffer swap
+    }
   }
+  if constexpr (PIPE3) __syncthreads();  // epilogue joins before C stores
 
   // epilogue: C = alpha*acc + beta*E, element (m,n) at lane n=lq, reg row
 #pragma unroll
@@ -287,12 +316,25 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
   auto* yp = reinterpret_cast<const __hip_bfloat16*>(Y.data_ptr());
   auto* ep = has_e ? reinterpret_cast<const __hip_bfloat16*>(E.data_ptr()) : nullptr;
   auto* cp = reinterpret_cast<__hip_bfloat16*>(C.data_ptr());
-  if (has_e)
-    muon_gemm_nt_kernel<true><<<grid, block, 0, stream>>>(xp, yp, ep, cp, M, N, K,
-                                                          (float)alpha, (float)beta);
-  else
-    muon_gemm_nt_kernel<false><<<grid, block, 0, stream>>>(xp, yp, ep, cp, M, N, K,
-                                                           (float)alpha, (float)beta);
+  static const bool pipe3 = []() {
+    const char* e = getenv("MCDP_MUON_PIPE3");
+    return e && atoi(e) != 0;
+  }();
+  if (has_e) {
+    if (pipe3)
+      muon_gemm_nt_kernel<true, __hip_bfloat16, true><<<grid, block, 0, stream>>>(
+          xp, yp, ep, cp, M, N, K, (float)alpha, (float)beta);
+    else
+      muon_gemm_nt_kernel<true><<<grid, block, 0, stream>>>(xp, yp, ep, cp, M, N, K,
+                                                            (float)alpha, (float)beta);
+  } else {
+    if (pipe3)
+      muon_gemm_nt_kernel<false, __hip_bfloat16, true><<<grid, block, 0, stream>>>(
+          xp, yp, ep, cp, M, N, K, (float)alpha, (float)beta);
+    else
+      muon_gemm_nt_kernel<false><<<grid, block, 0, stream>>>(xp, yp, ep, cp, M, N, K,
+                                                             (float)alpha, (float)beta);
+  }
 }
 
 // K9 Shampoo statistics EMA on the same MFMA NT kernel, fp32 state:
