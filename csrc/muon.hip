@@ -79,11 +79,17 @@ __device__ __forceinline__ bf16x8 img_frag(const __hip_bfloat16* img, int r0,
 
 // ---------------------------------------------------------------------------
 // C[M,N] = alpha * X[M,K] @ Y[N,K]^T + beta * E[M,N]
-template <bool HAS_E, typename TO = __hip_bfloat16, bool PIPE3 = false>
+// SPLIT: blockIdx.z = K-slice; partial tiles land as fp32 slabs in WS
+// [slices, M, N] and muon_combine_kernel applies alpha/beta/E. Chosen when
+// the (M/128)x(N/128) grid under-fills the 256 CUs (the NS shapes at
+// m <= 1024 run 64 blocks -> 25% occupancy without it; the guide's M=256
+// GEMM lesson: pick SPLITK so blocks ~ 0.5-1x the CU count).
+template <bool HAS_E, typename TO = __hip_bfloat16, bool PIPE3 = false, bool SPLIT = false>
 __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
     const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ Y,
     const TO* __restrict__ E, TO* __restrict__ C,
-    int M, int N, int K, float alpha, float beta) {
+    int M, int N, int K, float alpha, float beta,
+    float* __restrict__ WS = nullptr, int ksteps_per_slice = 0) {
   constexpr int TILE = BMN * BK;          // elements per image
   constexpr int SLOTS = PIPE3 ? 3 : 2;
   __shared__ __hip_bfloat16 smem[SLOTS * 2 * TILE];
@@ -101,21 +107,28 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x16{};
 
-  const int ksteps = K / BK;
+  int k0step = 0, ksteps = K / BK;
+  if constexpr (SPLIT) {
+    k0step = blockIdx.z * ksteps_per_slice;
+    ksteps = min(ksteps, k0step + ksteps_per_slice);
+    if (k0step >= ksteps) return;
+  }
+  const long kb0 = (long)k0step * BK;
   // prologue: stage tile 0 (and, 3-slot form, tile 1)
-  glds_tile(X + (long)m0 * K, K, smem);
-  glds_tile(Y + (long)n0 * K, K, smem + TILE);
+  glds_tile(X + (long)m0 * K + kb0, K, smem);
+  glds_tile(Y + (long)n0 * K + kb0, K, smem + TILE);
   if constexpr (PIPE3) {
-    if (ksteps > 1) {
-      glds_tile(X + (long)m0 * K + BK, K, smem + 2 * TILE);
-      glds_tile(Y + (long)n0 * K + BK, K, smem + 3 * TILE);
+    if (ksteps - k0step > 1) {
+      glds_tile(X + (long)m0 * K + kb0 + BK, K, smem + 2 * TILE);
+      glds_tile(Y + (long)n0 * K + kb0 + BK, K, smem + 3 * TILE);
     }
   } else {
     __syncthreads();
   }
 
-  for (int kt = 0; kt < ksteps; ++kt) {
-    const int buf = PIPE3 ? kt % 3 : (kt & 1);
+  for (int kt = k0step; kt < ksteps; ++kt) {
+    const int rel = kt - k0step;
+    const int buf = PIPE3 ? rel % 3 : (rel & 1);
     if constexpr (PIPE3) {
       // T3/T4 counted-vmcnt 3-slot ring: wait for tile kt's 8 glds (leave
       // tile kt+1's 8 in flight), raw barrier (no vmcnt(0) drain), THEN
@@ -128,14 +141,14 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // tail: nothing behind
       __builtin_amdgcn_s_barrier();
       if (kt + 2 < ksteps) {
-        const int ns = (kt + 2) % 3;
+        const int ns = (rel + 2) % 3;
         glds_tile(X + (long)m0 * K + (kt + 2) * BK, K, smem + ns * 2 * TILE);
         glds_tile(Y + (long)n0 * K + (kt + 2) * BK, K, smem + ns * 2 * TILE + TILE);
       }
     } else {
       if (kt + 1 < ksteps) {  // issue next-tile staging FIRST (T3 min-2-phase)
-        glds_tile(X + (long)m0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE);
-        glds_tile(Y + (long)n0 * K + (kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE + TILE);
+        glds_tile(X + (long)m0 * K + (long)(kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE);
+        glds_tile(Y + (long)n0 * K + (long)(kt + 1) * BK, K, smem + (buf ^ 1) * 2 * TILE + TILE);
       }
     }
     const __hip_bfloat16* ax = smem + buf * 2 * TILE;
@@ -170,10 +183,29 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
       for (int reg = 0; reg < 16; ++reg) {
         const int m = m0 + wr + i * 32 + acc_row(reg, hi);
         const int n = n0 + wc + j * 32 + lq;
-        float v = alpha * acc[i][j][reg];
-        if constexpr (HAS_E) v += beta * to_f32(E[(long)m * N + n]);
-        from_f32(&C[(long)m * N + n], v);
+        if constexpr (SPLIT) {
+          WS[((long)blockIdx.z * M + m) * N + n] = acc[i][j][reg];
+        } else {
+          float v = alpha * acc[i][j][reg];
+          if constexpr (HAS_E) v += beta * to_f32(E[(long)m * N + n]);
+          from_f32(&C[(long)m * N + n], v);
+        }
       }
+}
+
+// combine the split-K slabs: C = alpha * sum_s WS[s] + beta * E
+template <bool HAS_E, typename TO>
+__global__ void muon_combine_kernel(const float* __restrict__ WS, const TO* __restrict__ E,
+                                    TO* __restrict__ C, long mn, int slices,
+                                    float alpha, float beta) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < mn;
+       i += gridDim.x * (long)blockDim.x) {
+    float v = 0.f;
+    for (int s = 0; s < slices; ++s) v += WS[(long)s * mn + i];
+    v *= alpha;
+    if constexpr (HAS_E) v += beta * to_f32(E[i]);
+    from_f32(&C[i], v);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -320,6 +352,37 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
     const char* e = getenv("MCDP_MUON_PIPE3");
     return e && atoi(e) != 0;
   }();
+  // split-K when the tile grid under-fills the chip (env override: 0 = off)
+  const int nblocks = (M / BMN) * (N / BMN);
+  int splitk = 1;
+  static const int sk_env = []() {
+    const char* e = getenv("MCDP_MUON_SPLITK");
+    return e ? atoi(e) : -1;
+  }();
+  if (sk_env >= 0) splitk = sk_env > 0 ? sk_env : 1;
+  else if (nblocks < 192) {
+    splitk = 2;
+    while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) % 1 == 0 &&
+           (K / BK) / (splitk * 2) >= 2)
+      splitk *= 2;
+  }
+  if (splitk > 1) {
+    const int ksteps = K / BK;
+    const int kps = (ksteps + splitk - 1) / splitk;
+    auto ws = at::empty({(long)splitk * M * N}, X.options().dtype(at::kFloat));
+    dim3 gs(M / BMN, N / BMN, splitk);
+    muon_gemm_nt_kernel<false, __hip_bfloat16, false, true><<<gs, block, 0, stream>>>(
+        xp, yp, nullptr, nullptr, M, N, K, 1.f, 0.f, ws.data_ptr<float>(), kps);
+    const long mn = (long)M * N;
+    const int cg = (int)std::min<long>((mn + 1023) / 1024, 2048);
+    if (has_e)
+      muon_combine_kernel<true, __hip_bfloat16><<<cg, 1024, 0, stream>>>(
+          ws.data_ptr<float>(), ep, cp, mn, splitk, (float)alpha, (float)beta);
+    else
+      muon_combine_kernel<false, __hip_bfloat16><<<cg, 1024, 0, stream>>>(
+          ws.data_ptr<float>(), ep, cp, mn, splitk, (float)alpha, (float)beta);
+    return;
+  }
   if (has_e) {
     if (pipe3)
       muon_gemm_nt_kernel<true, __hip_bfloat16, true><<<grid, block, 0, stream>>>(
