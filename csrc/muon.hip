@@ -210,10 +210,12 @@ __global__ void muon_combine_kernel(const float* __restrict__ WS, const TO* __re
 
 // ---------------------------------------------------------------------------
 // C[M,N] = Bm[M,K] @ X[K,N] + a * X2[M,N]   (X2 = the same X when M == K)
+template <bool SPLIT = false>
 __global__ __launch_bounds__(TPB) void muon_gemm_nn_ax_kernel(
     const __hip_bfloat16* __restrict__ Bm, const __hip_bfloat16* __restrict__ X,
     const __hip_bfloat16* __restrict__ X2, __hip_bfloat16* __restrict__ C,
-    int M, int N, int K, float a) {
+    int M, int N, int K, float a,
+    float* __restrict__ WS = nullptr, int ksteps_per_slice = 0) {
   constexpr int TILE = BMN * BK;   // A image [128 m][64 k]
   constexpr int VROW = 64;         // transposed X image rows [128 n][64 k]
   __shared__ __hip_bfloat16 smem[2 * (TILE + BMN * VROW)];
@@ -267,14 +269,19 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nn_ax_kernel(
 #pragma unroll
     for (int j = 0; j < 2; ++j) acc[i][j] = f32x16{};
 
-  const int ksteps = K / BK;
-  glds_tile(Bm + (long)m0 * K, K, smem);
-  xstage_load(0);
+  int k0step = 0, ksteps = K / BK;
+  if constexpr (SPLIT) {
+    k0step = blockIdx.z * ksteps_per_slice;
+    ksteps = min(ksteps, k0step + ksteps_per_slice);
+    if (k0step >= ksteps) return;
+  }
+  glds_tile(Bm + (long)m0 * K + (long)k0step * BK, K, smem);
+  xstage_load(k0step * BK);
   xstage_write(smem + TILE);
   __syncthreads();
 
-  for (int kt = 0; kt < ksteps; ++kt) {
-    const int buf = kt & 1;
+  for (int kt = k0step; kt < ksteps; ++kt) {
+    const int buf = (kt - k0step) & 1;
     __hip_bfloat16* cur = smem + buf * (TILE + BMN * VROW);
     __hip_bfloat16* nxt = smem + (buf ^ 1) * (TILE + BMN * VROW);
     const bool has_next = kt + 1 < ksteps;
@@ -317,8 +324,12 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nn_ax_kernel(
       for (int reg = 0; reg < 16; ++reg) {
         const int m = m0 + wr + i * 32 + acc_row(reg, hi);
         const int n = n0 + wc + j * 32 + lq;
-        const float v = acc[i][j][reg] + a * __bfloat162float(X2[(long)m * N + n]);
-        C[(long)m * N + n] = __float2bfloat16(v);
+        if constexpr (SPLIT) {
+          WS[((long)blockIdx.z * M + m) * N + n] = acc[i][j][reg];
+        } else {
+          const float v = acc[i][j][reg] + a * __bfloat162float(X2[(long)m * N + n]);
+          C[(long)m * N + n] = __float2bfloat16(v);
+        }
       }
 }
 
@@ -430,9 +441,28 @@ void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a) {
               "muon_gemm_nn_ax: pad to 128/128/64 multiples");
   auto stream = at::cuda::getCurrentHIPStream();
   dim3 grid(M / BMN, N / BMN), block(TPB);
-  muon_gemm_nn_ax_kernel<<<grid, block, 0, stream>>>(
-      reinterpret_cast<const __hip_bfloat16*>(Bm.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(X.data_ptr()),
-      reinterpret_cast<const __hip_bfloat16*>(X.data_ptr()),
-      reinterpret_cast<__hip_bfloat16*>(C.data_ptr()), M, N, K, (float)a);
+  auto* bp = reinterpret_cast<const __hip_bfloat16*>(Bm.data_ptr());
+  auto* xp = reinterpret_cast<const __hip_bfloat16*>(X.data_ptr());
+  auto* cp = reinterpret_cast<__hip_bfloat16*>(C.data_ptr());
+  const int nblocks = (M / BMN) * (N / BMN);
+  int splitk = 1;
+  if (nblocks < 192) {
+    splitk = 2;
+    while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) / (splitk * 2) >= 2)
+      splitk *= 2;
+  }
+  if (splitk > 1) {
+    const int kps = (K / BK + splitk - 1) / splitk;
+    auto ws = at::empty({(long)splitk * M * N}, X.options().dtype(at::kFloat));
+    dim3 gs(M / BMN, N / BMN, splitk);
+    muon_gemm_nn_ax_kernel<true><<<gs, block, 0, stream>>>(
+        bp, xp, xp, cp, M, N, K, (float)a, ws.data_ptr<float>(), kps);
+    const long mn = (long)M * N;
+    const int cg = (int)std::min<long>((mn + 1023) / 1024, 2048);
+    // combine with the +a*X epilogue: reuse muon_combine with E = X, beta = a
+    muon_combine_kernel<true, __hip_bfloat16><<<cg, 1024, 0, stream>>>(
+        ws.data_ptr<float>(), xp, cp, mn, splitk, 1.f, (float)a);
+    return;
+  }
+  muon_gemm_nn_ax_kernel<false><<<grid, block, 0, stream>>>(bp, xp, xp, cp, M, N, K, (float)a);
 }
