@@ -111,7 +111,15 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nt_kernel(
   if constexpr (SPLIT) {
     k0step = blockIdx.z * ksteps_per_slice;
     ksteps = min(ksteps, k0step + ksteps_per_slice);
-    if (k0step >= ksteps) return;
+    if (k0step >= ksteps) {
+      // empty tail slice: the combine sums EVERY slab, so write zeros
+      // (returning left recycled-allocation garbage — order-dependent)
+      for (int r = threadIdx.x; r < BMN; r += TPB) {
+        float* row = WS + ((long)blockIdx.z * M + m0 + r) * N + n0;
+        for (int c = 0; c < BMN; ++c) row[c] = 0.f;
+      }
+      return;
+    }
   }
   const long kb0 = (long)k0step * BK;
   // prologue: stage tile 0 (and, 3-slot form, tile 1)
@@ -273,7 +281,13 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nn_ax_kernel(
   if constexpr (SPLIT) {
     k0step = blockIdx.z * ksteps_per_slice;
     ksteps = min(ksteps, k0step + ksteps_per_slice);
-    if (k0step >= ksteps) return;
+    if (k0step >= ksteps) {
+      for (int r = threadIdx.x; r < BMN; r += TPB) {
+        float* row = WS + ((long)blockIdx.z * M + m0 + r) * N + n0;
+        for (int c = 0; c < BMN; ++c) row[c] = 0.f;
+      }
+      return;
+    }
   }
   glds_tile(Bm + (long)m0 * K + (long)k0step * BK, K, smem);
   xstage_load(k0step * BK);
@@ -371,7 +385,7 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
     return e ? atoi(e) : -1;
   }();
   if (sk_env >= 0) splitk = sk_env > 0 ? sk_env : 1;
-  else if (nblocks < 192) {
+  else if (nblocks < 192 && (K / BK) >= 2) {
     splitk = 2;
     while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) % 1 == 0 &&
            (K / BK) / (splitk * 2) >= 2)
@@ -446,7 +460,7 @@ void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a) {
   auto* cp = reinterpret_cast<__hip_bfloat16*>(C.data_ptr());
   const int nblocks = (M / BMN) * (N / BMN);
   int splitk = 1;
-  if (nblocks < 192) {
+  if (nblocks < 192 && (K / BK) >= 2) {
     splitk = 2;
     while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) / (splitk * 2) >= 2)
       splitk *= 2;
