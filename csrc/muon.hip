@@ -385,7 +385,7 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
     return e ? atoi(e) : -1;
   }();
   if (sk_env >= 0) splitk = sk_env > 0 ? sk_env : 1;
-  else if (nblocks < 192 && (K / BK) >= 2) {
+  else if (nblocks < 384 && (K / BK) >= 2) {  // < 1.5x CUs: 2 blocks/CU co-residency needs >= 384
     splitk = 2;
     while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) % 1 == 0 &&
            (K / BK) / (splitk * 2) >= 2)
@@ -460,7 +460,7 @@ void muon_gemm_nn_ax(at::Tensor Bm, at::Tensor X, at::Tensor C, double a) {
   auto* cp = reinterpret_cast<__hip_bfloat16*>(C.data_ptr());
   const int nblocks = (M / BMN) * (N / BMN);
   int splitk = 1;
-  if (nblocks < 192 && (K / BK) >= 2) {
+  if (nblocks < 384 && (K / BK) >= 2) {
     splitk = 2;
     while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) / (splitk * 2) >= 2)
       splitk *= 2;
