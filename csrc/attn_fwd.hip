@@ -6,8 +6,13 @@
 // Structure (gfx950), following the guide's 8-wave attention ladder:
 //   - NW waves/block, each wave owns 32 q rows (8 waves = 256 q rows share
 //     every K/V tile -> staging traffic amortized 8x),
-//   - KVB-row K/V tiles in DOUBLE-BUFFERED LDS, ONE barrier per tile:
-//     tile t+1 is written into buf^1 while buf is being consumed,
+//   - causal CU load balance: tile_map() rotates the q-tile index across a
+//     CU's successive blocks and (head-major layouts) pins each (b,h)'s
+//     blocks to one XCD (attn_common.h; fwd 433 -> 569 TF at the 1B shape),
+//   - cross-tile software pipeline: PV lags QK^T by one tile in a 3-slot
+//     LDS ring, so tile j's exp/psum/pack VALU shares a basic block (and
+//     the issue gaps) with tile j-1's PV MFMAs; branchless softmax tail,
+//   - KVB-row K/V tiles, ONE barrier per tile,
 //   - XOR-swizzled images (guide §6 Guideline 4: row-major D=128 bf16 read
 //     column-wise by a lane group is an up-to-16-way bank conflict, and +1
 //     padding does NOT fix it): K row-major [KVB][D] with element column
@@ -18,7 +23,13 @@
 //   - swapped QK^T (attn_common.h) keeps the online softmax lane-local,
 //     exponentials in base-2 with log2(e) folded into the scale,
 //   - interior causal tiles take a mask-free fast path (wave-uniform branch),
-//   - s_setprio(1) around the MFMA clusters (guide T5),
+//   - T5 STATIC form only: one priority raise for the younger wave half
+//     (per-cluster setprio flips are scheduling fences and were keeping the
+//     softmax VALU out of the MFMA gaps; the sched_group_barrier interleave
+//     corrupts numerics and stays compiled-but-disabled — VAR bit 1),
+//   - MOD_BLOCKMASK (K2): arbitrary mask_mod via device granule codes +
+//     packed keep-bits + per-block live kv ranges; optional additive bias
+//     stream for score_mod (ops/attention.py CompiledBlockMask),
 //   - causal/sliding-window tiles skipped at block level; GQA reads the
 //     shared KV head directly; BSHD layout, strided views accepted.
 #include <torch/extension.h>
