@@ -387,8 +387,7 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
   if (sk_env >= 0) splitk = sk_env > 0 ? sk_env : 1;
   else if (nblocks < 384 && (K / BK) >= 2) {  // < 1.5x CUs: 2 blocks/CU co-residency needs >= 384
     splitk = 2;
-    while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) % 1 == 0 &&
-           (K / BK) / (splitk * 2) >= 2)
+    while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) / (splitk * 2) >= 2)
       splitk *= 2;
   }
   if (splitk > 1) {

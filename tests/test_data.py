@@ -80,3 +80,27 @@ def test_bpe_tokenizer_training(tmp_path):
     ids = tok.tokenize("the quick brown fox")
     assert len(ids) > 0
     assert "quick" in tok.detokenize(ids)
+
+
+def test_synthetic_vocab_capped_to_tokenizer_vocab():
+    """Synthetic ids >= the model vocab are an OOB embedding gather (a GPU
+    memory fault with no traceback — the 7B debugging trail): DataManager
+    caps and warns."""
+    import warnings
+
+    from mlx_cuda_distributed_pretraining_amd.data.dataset import DataManager
+
+    class Tok:
+        vocab_size = 100
+        PAD_TOKEN = 0
+
+    cfg = type("D", (), {})()
+    cfg.preprocessing = {"max_context_size": 8}
+    cfg.synthetic = True
+    cfg.synthetic_vocab_size = 32000
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        dm = DataManager(cfg, Tok(), batch_size=2)
+        assert any("capping" in str(x.message) for x in w)
+    b = dm.generate_batch(0)
+    assert int(b.max()) < 100
