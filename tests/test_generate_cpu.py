@@ -90,3 +90,53 @@ def test_speculative_decoding_matches_greedy():
             got = speculative_generate_tokens(target, d, prompt,
                                               max_tokens=24, k=kk)
             assert got == want, (kk, d is target, got, want)
+
+
+def test_rejection_step_law():
+    """Speculative-sampling math (inference/speculative.py _rejection_step):
+    the combined law (draft proposal -> accept/residual-resample) must be
+    exactly the target distribution p, for an arbitrary draft q."""
+    from mlx_cuda_distributed_pretraining_amd.inference.speculative import (
+        _rejection_step)
+
+    torch.manual_seed(0)
+    V = 8
+    p = torch.rand(V); p = p / p.sum()
+    q = torch.rand(V) ** 2; q = q / q.sum()
+    # p == q: always accepted
+    for tok in range(V):
+        assert _rejection_step(p, p, tok, u=0.999999) is None
+    # disjoint: always rejected, resample lands in p's support
+    p2 = torch.tensor([0.5, 0.5, 0, 0, 0, 0, 0, 0.0])
+    q2 = torch.tensor([0, 0, 0.5, 0.5, 0, 0, 0, 0.0])
+    for tok in (2, 3):
+        r = _rejection_step(p2, q2, tok, u=0.5)
+        assert r in (0, 1)
+    # empirical law: proposal ~ q then rejection step => sample ~ p
+    n = 20000
+    counts = torch.zeros(V)
+    toks = torch.multinomial(q, n, replacement=True)
+    for t in toks.tolist():
+        r = _rejection_step(p, q, t)
+        counts[t if r is None else r] += 1
+    emp = counts / n
+    assert (emp - p).abs().max().item() < 0.02, (emp, p)
+
+
+def test_speculative_decoding_sampled_runs():
+    """temperature > 0 spec decoding: runs, respects vocab/stop/max_tokens;
+    draft == target accepts aggressively (acceptance prob 1 when p == q)."""
+    from mlx_cuda_distributed_pretraining_amd.inference.speculative import (
+        speculative_generate_tokens)
+
+    torch.manual_seed(1)
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+    args = ModelArgs(hidden_size=32, intermediate_size=64, num_layers=2,
+                     num_heads=2, num_kv_heads=2, vocab_size=61,
+                     max_position_embeddings=128)
+    model = Model(args).eval()
+    prompt = [1, 5, 9]
+    out = speculative_generate_tokens(model, model, prompt, max_tokens=12,
+                                      k=3, temperature=0.8)
+    assert 0 < len(out) <= 12
+    assert all(0 <= t < model.args.vocab_size for t in out)
