@@ -357,7 +357,13 @@ class Trainer:
                 import torch.distributed as dist
 
                 dist.all_reduce(ntok_g, group=self.tp_pg)  # THIS replica's tokens
-            return loss_sum / ntok_g, ntok_g.long()
+            loss = loss_sum / ntok_g
+            # MoE under SP: aux is computed on the gathered tokens
+            # (identical per rank, pre-divided by ep_world — parallel/tp.py)
+            aux = getattr(self.model, "aux_loss", None)
+            if aux is not None and self.model.training:
+                loss = loss + self.model_args.router_aux_loss_coef * aux
+            return loss, ntok_g.long()
         loss, ntok = fused_cross_entropy(
             logits.reshape(-1, logits.shape[-1]),
             targets.reshape(-1),

@@ -291,7 +291,13 @@ class MoE(nn.Module):
         # contributions for every token and the g all-reduce sums the
         # partials (no all-to-all needed in this form)
         ep = getattr(self, "_tp", False)
-        if ep:
+        sp = getattr(self, "_sp", False)
+        shard_shape = x.shape
+        if sp:
+            from ..parallel.tp import gather_sp
+
+            x = gather_sp(x)       # S-shard -> full tokens (exact autograd)
+        elif ep:
             x = copy_to_tp(x)
         shape = x.shape
         xf = x.reshape(-1, shape[-1])
@@ -308,10 +314,13 @@ class MoE(nn.Module):
             frac = counts / (n * self.top_k)
         self.aux_loss = self.num_experts * (frac * probs.mean(0)).sum()
         if ep:
-            # computed identically on every EP rank, and the trainer SUMS the
-            # router grads over the group -> pre-divide so the aux term
-            # arrives unscaled (parallel/tp.py ep_allreduce_router_grads)
-            self.aux_loss = self.aux_loss / self._ep_world
+            # every EP rank computes the FULL aux on the replicated router,
+            # and the trainer then SUMS router grads over the group — so the
+            # GRADIENT must be pre-divided by the EP degree, while the
+            # reported VALUE stays the full aux (detached remainder):
+            # value == single-process, grad arrives unscaled after the sum.
+            a = self.aux_loss / self._ep_world
+            self.aux_loss = a + (self.aux_loss - a).detach()
         n_local = self.w_gate_up.shape[0]
         e0 = self._ep_rank * n_local if ep else 0
         if self.capacity_factor > 0:
@@ -327,6 +336,10 @@ class MoE(nn.Module):
                 h = swiglu(toks @ self.w_gate_up[el].t())
                 y = (h @ self.w_down[el].t()) * gates[rows, slot].unsqueeze(-1)
                 out.index_add_(0, rows, y.to(out.dtype))
+        if sp:
+            from ..parallel.tp import scatter_sp
+
+            return scatter_sp(out.reshape(shape)).reshape(shard_shape)
         if ep:
             return reduce_from_tp(out).reshape(shape)
         return out.reshape(shape)

@@ -352,9 +352,13 @@ def _shard_experts(moe, rank: int, world: int, sequence_parallel: bool) -> None:
     rank only backprops its local experts' gate terms) -> the trainer sums
     router grads over the group (ep_allreduce_router_grads). The aux
     load-balance loss is computed identically on every rank, so MoE.forward
-    pre-divides it by the EP degree to survive that sum unscaled."""
-    if sequence_parallel:
-        raise NotImplementedError("SP + MoE expert parallelism: ROADMAP")
+    pre-divides it by the EP degree to survive that sum unscaled.
+
+    Under sequence parallelism the MoE gathers the S-shard first
+    (gather_sp) and scatters the reduced output back (scatter_sp) — exact,
+    at the cost of momentarily holding the full activation for the MoE
+    block (token-level dispatch without the gather is the all-to-all
+    follow-up, ROADMAP)."""
     E = moe.num_experts
     if E % world:
         raise ValueError(f"EP degree {world} must divide num_local_experts={E}")
@@ -368,6 +372,10 @@ def _shard_experts(moe, rank: int, world: int, sequence_parallel: bool) -> None:
     moe.router.weight._ep_router = True  # grads summed over the group
     moe._ep_rank, moe._ep_world = rank, world
     moe._tp = True
+    # SP: the block hands the MLP an S-SHARD; MoE gathers it (all-gather fwd
+    # / reduce-scatter bwd), runs the replicated-activation EP body, and
+    # reduce-scatters the partial-sum output back to the shard.
+    moe._sp = sequence_parallel
 
 
 def ep_allreduce_router_grads(model) -> None:
