@@ -1178,3 +1178,68 @@ def test_sp_moe_expert_parallel_matches_single_process():
     # identical inputs is bitwise exact: /tmp-level probe in the round-2
     # log). Dense SP (continuous) matches at 5e-5 above.
     assert res[0]["losses"] == pytest.approx(ref, abs=1e-2), (res[0]["losses"], ref)
+
+
+def test_reshard_tp_state_dict_roundtrip():
+    """shard(full, tp=B) composed with merge == identity, for B in {2, 4},
+    on a model with GQA + untied head (ROADMAP r1 #13: re-shard to a
+    DIFFERENT tp size)."""
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+    from tools.merge_tp_checkpoint import merge_tp_state_dicts
+    from tools.reshard_tp_checkpoint import shard_tp_state_dict
+
+    torch.manual_seed(3)
+    args = ModelArgs(hidden_size=32, intermediate_size=64, num_layers=2,
+                     num_heads=4, num_kv_heads=4, vocab_size=64,
+                     max_position_embeddings=64, tie_word_embeddings=False)
+    full = Model(args).state_dict()
+    for B in (2, 4):
+        shards = [shard_tp_state_dict(full, r, B, args.num_heads, args.num_kv_heads,
+                                      args.head_dim, args.intermediate_size)
+                  for r in range(B)]
+        back = merge_tp_state_dicts(shards, args.num_heads, args.num_kv_heads,
+                                    args.head_dim, args.intermediate_size)
+        assert set(back) == set(full)
+        for k in full:
+            assert torch.equal(back[k], full[k]), k
+
+    # A -> merge -> B -> merge == A -> merge  (reshard path equivalence)
+    a_shards = [shard_tp_state_dict(full, r, 2, args.num_heads, args.num_kv_heads,
+                                    args.head_dim, args.intermediate_size)
+                for r in range(2)]
+    merged_a = merge_tp_state_dicts(a_shards, args.num_heads, args.num_kv_heads,
+                                    args.head_dim, args.intermediate_size)
+    b_shards = [shard_tp_state_dict(merged_a, r, 4, args.num_heads, args.num_kv_heads,
+                                    args.head_dim, args.intermediate_size)
+                for r in range(4)]
+    merged_b = merge_tp_state_dicts(b_shards, args.num_heads, args.num_kv_heads,
+                                    args.head_dim, args.intermediate_size)
+    for k in full:
+        assert torch.equal(merged_b[k], full[k]), k
+
+
+def test_reshard_sharded_forward_matches_full():
+    """Weights sharded by shard_tp_state_dict load into a live TP-sharded
+    model (apply_tensor_parallel) without mismatch: the slicing rules agree
+    with the runtime sharding."""
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+    from mlx_cuda_distributed_pretraining_amd.parallel.tp import apply_tensor_parallel
+    from tools.reshard_tp_checkpoint import shard_tp_state_dict
+
+    torch.manual_seed(4)
+    args = ModelArgs(hidden_size=32, intermediate_size=64, num_layers=1,
+                     num_heads=4, num_kv_heads=2, vocab_size=64,
+                     max_position_embeddings=64, tie_word_embeddings=False)
+    ref = Model(args)
+    full = ref.state_dict()
+    for r in range(2):
+        m = Model(args)
+        apply_tensor_parallel(m, r, 2)
+        # the runtime TP always vocab-shards the (untied) lm head
+        sd = shard_tp_state_dict(full, r, 2, args.num_heads, args.num_kv_heads,
+                                 args.head_dim, args.intermediate_size,
+                                 vocab_parallel=True)
+        missing, unexpected = m.load_state_dict(sd, strict=False)
+        assert not unexpected, unexpected
+        for k, v in m.state_dict().items():
+            assert v.shape == sd[k].shape, (k, v.shape, sd[k].shape)
