@@ -372,6 +372,8 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
 
   // sub-tiles per barrier: dV runs two (32 MFMAs/barrier, 220-238 VGPR);
   // dK holds V fragments too and spills at NP=2, so it stays at one.
+  // dK at NP=2 re-checked in round 2 (post-NU staging): 256 VGPR with 7
+  // spills + 32 B scratch AND occupancy halves (2 blocks/CU -> 1) — stays 1.
   constexpr int NP = WANT_DK ? 1 : 2;
 
   // T14 staging of two 32-row q tiles: threads 0..255 own Q chunks
