@@ -345,8 +345,19 @@ def _flex_blockmask(q, k, v, bm: "CompiledBlockMask", score_mod, scale):
         ki = torch.arange(KV, device=q.device).view(1, 1, 1, KV)
         bi = torch.arange(B, device=q.device).view(B, 1, 1, 1)
         hi = torch.arange(H, device=q.device).view(1, H, 1, 1)
+        # the kernel consumes score_mod as an ADDITIVE bias (f(s) = s + f(0));
+        # probe additivity on a small slice and fall back to the fp32
+        # composition for non-additive mods (e.g. multiplicative scaling)
+        sp, kp = min(S, 8), min(KV, 8)
+        probe0 = torch.zeros(B, H, sp, kp, device=q.device)
+        f0 = score_mod(probe0, bi, hi, qi[:, :, :sp], ki[:, :, :, :kp]).float()
+        f1 = score_mod(probe0 + 1.0, bi, hi, qi[:, :, :sp], ki[:, :, :, :kp]).float()
+        if not torch.allclose(f1 - f0, torch.ones_like(f0), atol=1e-4):
+            return attention_ref(q, k, v, causal=False, scale=scale,
+                                 score_mod=score_mod,
+                                 mask_mod=getattr(bm, "_mask_mod", None))
         zero = torch.zeros(B, H, S, KV, device=q.device)
-        bias = (score_mod(zero, bi, hi, qi, ki) - 0.0).to(torch.bfloat16)
+        bias = score_mod(zero, bi, hi, qi, ki).to(torch.bfloat16)
     o, _ = require_ext().attn_fwd_blockmask(q, k, v, float(scale), bm.gran,
                                             bm.bits, bm.range, bias)
     return o
@@ -398,6 +409,7 @@ class CompiledBlockMask:
     def __init__(self, mask_mod: Callable, B: int, H: int, Q_LEN: int, KV_LEN: int,
                  device="cuda"):
         self.B, self.H, self.Q_LEN, self.KV_LEN = B, H, Q_LEN, KV_LEN
+        self._mask_mod = mask_mod  # kept for the non-additive score_mod fallback
         qi = torch.arange(Q_LEN, device=device)
         ki = torch.arange(KV_LEN, device=device)
         nq32 = (Q_LEN + 31) // 32
