@@ -162,6 +162,16 @@ def _all_gather_seq(x: torch.Tensor) -> torch.Tensor:
 def _reduce_scatter_seq(x: torch.Tensor) -> torch.Tensor:
     rank, world = _sp_rank_world()
     x = x.contiguous()
+    # RCCL path: a real reduce_scatter moves 1/world the bytes of
+    # all-reduce+slice (ROADMAP r1 #13). gloo has no reduce_scatter_tensor,
+    # and the S dim must split evenly — fall back otherwise (identical
+    # semantics, CPU tests run the fallback).
+    backend = dist.get_backend(_TP_GROUP) if _TP_GROUP is not None else dist.get_backend()
+    if str(backend) == "nccl" and x.shape[1] % world == 0:
+        xt = x.movedim(1, 0).contiguous()  # scatter dim first
+        out = torch.empty_like(xt[: xt.shape[0] // world])
+        dist.reduce_scatter_tensor(out, xt, group=_TP_GROUP)
+        return out.movedim(0, 1).contiguous()
     dist.all_reduce(x, group=_TP_GROUP)
     return _sp_slice(x, rank, world)
 
