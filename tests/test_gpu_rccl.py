@@ -65,11 +65,17 @@ def test_rccl_backend_world1_ddp_zero1_step():
             losses.append(float(loss))
         assert all(torch.isfinite(torch.tensor(losses)))
         assert losses[-1] < losses[0], f"loss did not fall on RCCL path: {losses}"
-        # a real collective on the nccl backend
+        # real collectives on the nccl backend (all-reduce + the
+        # reduce_scatter_tensor the SP path uses on RCCL)
         t = torch.ones(8, device=dev)
         dist.all_reduce(t)
         torch.cuda.synchronize()
         assert t.sum().item() == 8.0
+        src = torch.arange(8.0, device=dev)
+        out = torch.empty(8, device=dev)  # world 1: scatter = identity
+        dist.reduce_scatter_tensor(out, src)
+        torch.cuda.synchronize()
+        assert torch.equal(out, src)
     finally:
         if created and dist.is_initialized():
             dist.destroy_process_group()
