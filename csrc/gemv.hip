@@ -11,6 +11,13 @@
 
 namespace {
 
+// nontemporal builtin needs a native vector type (uint4 is a class)
+typedef unsigned int u32x4_nt __attribute__((ext_vector_type(4)));
+__device__ __forceinline__ uint4 nt_load_u4(const void* p) {
+  u32x4_nt t = __builtin_nontemporal_load(reinterpret_cast<const u32x4_nt*>(p));
+  return *reinterpret_cast<const uint4*>(&t);
+}
+
 // block = 256 (4 waves). Each wave owns one output row per pass; lanes cover
 // 512 elements per pass (64 lanes x 8). Rows grid-strided.
 __global__ __launch_bounds__(256) void gemv_bf16_kernel(
@@ -33,9 +40,26 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
     const __hip_bfloat16* wr = W + (long)row * K;
     float acc = 0.f;
     int k = lane * 8;
+    // deep-unrolled NON-TEMPORAL weight stream (guide: decode weights are
+    // read once per token per CU — keep >=4 loads in flight, nt policy;
+    // a single in-flight uint4 leaves the loop latency-bound at ~25% BW)
+    for (; k + 8 * WAVE * 3 + 8 <= K; k += WAVE * 8 * 4) {
+      U4 wv[4], xv[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        wv[u].u = nt_load_u4(wr + k + u * WAVE * 8);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        xv[u].u = *reinterpret_cast<const uint4*>(x_lds + k + u * WAVE * 8);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc += bf16_bits_to_f32(wv[u].s[j]) * bf16_bits_to_f32(xv[u].s[j]);
+    }
     for (; k + 8 <= K; k += WAVE * 8) {
       U4 wv, xv;
-      wv.u = *reinterpret_cast<const uint4*>(wr + k);
+      wv.u = nt_load_u4(wr + k);
       xv.u = *reinterpret_cast<const uint4*>(x_lds + k);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -118,9 +142,27 @@ __global__ __launch_bounds__(256) void gemv_ex_kernel(
     float acc[BB];
 #pragma unroll
     for (int b = 0; b < BB; ++b) acc[b] = 0.f;
-    for (int k = lane * 8; k + 8 <= K; k += WAVE * 8) {
+    int k = lane * 8;
+    constexpr int UNR2 = (BB > 2) ? 2 : 4;  // keep several nt loads in flight
+    for (; k + 8 * WAVE * (UNR2 - 1) + 8 <= K; k += WAVE * 8 * UNR2) {
+      U4 wv[UNR2];
+#pragma unroll
+      for (int u = 0; u < UNR2; ++u)
+        wv[u].u = nt_load_u4(wr + k + u * WAVE * 8);
+#pragma unroll
+      for (int u = 0; u < UNR2; ++u)
+#pragma unroll
+        for (int b = 0; b < BB; ++b) {
+          U4 xv;
+          xv.u = *reinterpret_cast<const uint4*>(x_lds + (long)b * K + k + u * WAVE * 8);
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            acc[b] += bf16_bits_to_f32(wv[u].s[j]) * bf16_bits_to_f32(xv.s[j]);
+        }
+    }
+    for (; k + 8 <= K; k += WAVE * 8) {
       U4 wv;
-      wv.u = *reinterpret_cast<const uint4*>(wr + k);
+      wv.u = nt_load_u4(wr + k);
 #pragma unroll
       for (int b = 0; b < BB; ++b) {
         U4 xv;
