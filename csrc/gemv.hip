@@ -137,26 +137,18 @@ __global__ __launch_bounds__(256) void gemv_ex_kernel(
   __syncthreads();
 
   const int wid = tid / WAVE, lane = tid % WAVE;
-  // TWO output rows per wave: doubles the independent nt weight loads in
-  // flight (the stream is latency-bound at one row per wave)
-  for (int row = blockIdx.x * 8 + wid * 2; row < N; row += gridDim.x * 8) {
-    const bool two = row + 1 < N;
-    const __hip_bfloat16* wr0 = W + (long)row * K;
-    const __hip_bfloat16* wr1 = W + (long)(two ? row + 1 : row) * K;
-    float acc[2][BB];
+  for (int row = blockIdx.x * 4 + wid; row < N; row += gridDim.x * 4) {
+    const __hip_bfloat16* wr = W + (long)row * K;
+    float acc[BB];
 #pragma unroll
-    for (int r = 0; r < 2; ++r)
-#pragma unroll
-      for (int b = 0; b < BB; ++b) acc[r][b] = 0.f;
+    for (int b = 0; b < BB; ++b) acc[b] = 0.f;
     int k = lane * 8;
-    constexpr int UNR2 = (BB > 2) ? 1 : 2;  // 2 rows x UNR2 nt loads in flight
+    constexpr int UNR2 = (BB > 2) ? 2 : 4;  // keep several nt loads in flight
     for (; k + 8 * WAVE * (UNR2 - 1) + 8 <= K; k += WAVE * 8 * UNR2) {
-      U4 wv0[UNR2], wv1[UNR2];
+      U4 wv[UNR2];
 #pragma unroll
-      for (int u = 0; u < UNR2; ++u) {
-        wv0[u].u = nt_load_u4(wr0 + k + u * WAVE * 8);
-        wv1[u].u = nt_load_u4(wr1 + k + u * WAVE * 8);
-      }
+      for (int u = 0; u < UNR2; ++u)
+        wv[u].u = nt_load_u4(wr + k + u * WAVE * 8);
 #pragma unroll
       for (int u = 0; u < UNR2; ++u)
 #pragma unroll
@@ -164,39 +156,29 @@ __global__ __launch_bounds__(256) void gemv_ex_kernel(
           U4 xv;
           xv.u = *reinterpret_cast<const uint4*>(x_lds + (long)b * K + k + u * WAVE * 8);
 #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            acc[0][b] += bf16_bits_to_f32(wv0[u].s[j]) * bf16_bits_to_f32(xv.s[j]);
-            acc[1][b] += bf16_bits_to_f32(wv1[u].s[j]) * bf16_bits_to_f32(xv.s[j]);
-          }
+          for (int j = 0; j < 8; ++j)
+            acc[b] += bf16_bits_to_f32(wv[u].s[j]) * bf16_bits_to_f32(xv.s[j]);
         }
     }
     for (; k + 8 <= K; k += WAVE * 8) {
-      U4 wv0, wv1;
-      wv0.u = nt_load_u4(wr0 + k);
-      wv1.u = nt_load_u4(wr1 + k);
+      U4 wv;
+      wv.u = nt_load_u4(wr + k);
 #pragma unroll
       for (int b = 0; b < BB; ++b) {
         U4 xv;
         xv.u = *reinterpret_cast<const uint4*>(x_lds + (long)b * K + k);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          acc[0][b] += bf16_bits_to_f32(wv0.s[j]) * bf16_bits_to_f32(xv.s[j]);
-          acc[1][b] += bf16_bits_to_f32(wv1.s[j]) * bf16_bits_to_f32(xv.s[j]);
-        }
+        for (int j = 0; j < 8; ++j)
+          acc[b] += bf16_bits_to_f32(wv.s[j]) * bf16_bits_to_f32(xv.s[j]);
       }
     }
 #pragma unroll
     for (int b = 0; b < BB; ++b) {
       if (b >= B) break;
-      float a0 = wave_reduce_sum(acc[0][b]);
-      float a1 = wave_reduce_sum(acc[1][b]);
+      float a = wave_reduce_sum(acc[b]);
       if (lane == 0) {
-        if (res != nullptr) a0 += to_f32(res[(long)b * N + row]);
-        from_f32(&y[(long)b * N + row], a0);
-        if (two) {
-          if (res != nullptr) a1 += to_f32(res[(long)b * N + row + 1]);
-          from_f32(&y[(long)b * N + row + 1], a1);
-        }
+        if (res != nullptr) a += to_f32(res[(long)b * N + row]);
+        from_f32(&y[(long)b * N + row], a);
       }
     }
   }
