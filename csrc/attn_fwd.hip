@@ -465,7 +465,8 @@ void launch_fwd_cfg(dim3 grid, dim3 block, hipStream_t stream,
                     int B, int Sq, int Skv, int Hq, int Hkv, float scale, int modarg,
                     long q_rs, long k_rs, long v_rs) {
   const char* e = getenv("MCDP_ATTN_KVB");
-  const int kvb = e ? atoi(e) : 64;
+  int kvb = e ? atoi(e) : 64;
+  if (kvb == 128 && D != 64) kvb = 64;  // KVB=128 is D=64-only (VGPR budget)
   const int nw = fwd_qpb() / QPW;
   static const int var = []() {
     const char* e = getenv("MCDP_ATTN_FWD_VAR");
@@ -495,6 +496,10 @@ void launch_fwd_cfg(dim3 grid, dim3 block, hipStream_t stream,
   else if (nw == 4 && kvb == 64) LAUNCH(4, 64);
   else if (nw == 4 && kvb == 32) LAUNCH(4, 32);
   else if (nw == 16 && kvb == 64) LAUNCH(16, 64);
+  else if (nw == 8 && kvb == 128) {
+    if constexpr (D == 64) LAUNCH(8, 128);
+    else TORCH_CHECK(false, "attn_fwd: KVB=128 is D=64-only (LDS/VGPR budget)");
+  }
   else TORCH_CHECK(false, "attn_fwd: unsupported NW/KVB ", nw, "/", kvb);
 #undef LAUNCH
 }
