@@ -66,7 +66,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_fwd_kernel(
   constexpr int KT = KVB / 32;  // 32-row k sub-tiles
   // XOR swizzle masks (element units; <<3 = 8-element/16-byte granules)
   constexpr int KSWZ = (D >= 128) ? 15 : 7;   // K image rows (row bytes = 2D)
-  constexpr int VROW = 64;                    // V^T image row (elements; swzt())
+  constexpr int VROW = (KVB > 64) ? KVB : 64; // V^T image row: col index is the KV row (< KVB)
   constexpr int TILE = KVB * D + D * VROW;    // elements per buffer
 
   // 3-slot ring: with PV lagging QK^T by one tile (cross-tile pipeline),
