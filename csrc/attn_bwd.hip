@@ -31,7 +31,7 @@ namespace {
 // kv tile per dQ iteration: 64 rows staged per barrier (two 32-row halves
 // processed sequentially -> 48 MFMAs between barriers instead of 24; viable
 // at 32 KB/buffer now that the transposed K image is gone)
-constexpr int KVB = 64;
+constexpr int KVB_FILE = 64;
 constexpr float LOG2E = 1.4426950408889634f;
 
 // waves per block (8 or 16): 16-wave blocks halve staging traffic and
@@ -104,6 +104,9 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
   constexpr int QPB = 32 * NW;
   constexpr int DBLK = D / 16;
   constexpr int DCOL = D / 32;
+  // D=64: 128-row kv tiles double the MFMAs per barrier at halved per-tile
+  // register cost (mirrors the fwd D=64 KVB bump)
+  constexpr int KVB = (D == 64) ? 128 : KVB_FILE;
   // K rm + V rm only: the dQ-accumulate B-fragments (K, k-strided) are read
   // straight from the row-major K image with ds_read_b64_tr_b16 (tr16_frag),
   // so no transposed K image is staged.
@@ -209,10 +212,10 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dq_kernel(
     const bool has_next = kv0 + KVB < kv_hi;
     if (has_next) stage_load(kv0 + KVB);
 
-    // two 32-row halves per staged tile (s_tr only uses low row bits, so the
-    // half-image at rows 32..63 is the same layout at a +32*D offset)
+    // 32-row sub-tiles of the staged kv tile (s_tr only uses low row bits,
+    // so each sub-image at rows 32h.. is the same layout at a +32h*D offset)
 #pragma unroll
-    for (int hf = 0; hf < 2; ++hf) {
+    for (int hf = 0; hf < KVB / 32; ++hf) {
       const int kvh = kv0 + hf * 32;
       if (kvh >= kv_hi) break;  // block-uniform (kv_hi is block-level)
       const __hip_bfloat16* k_lds = smem + buf * TILE + hf * 32 * D;
