@@ -28,9 +28,9 @@
 
 namespace {
 
-// kv tile per dQ iteration: 64 rows staged per barrier (two 32-row halves
-// processed sequentially -> 48 MFMAs between barriers instead of 24; viable
-// at 32 KB/buffer now that the transposed K image is gone)
+// kv tile per dQ iteration: 64 rows staged per barrier at D=128 (two 32-row
+// sub-tiles -> 48 MFMAs/barrier), 128 rows at D=64 (four sub-tiles, 96
+// MFMAs/barrier at 166 VGPR) — the dQ kernel overrides this per D.
 constexpr int KVB_FILE = 64;
 constexpr float LOG2E = 1.4426950408889634f;
 
@@ -377,7 +377,7 @@ __global__ __launch_bounds__(NW* WAVE) void attn_bwd_dkv_kernel(
   // dK holds V fragments too and spills at NP=2, so it stays at one.
   // dK at NP=2 re-checked in round 2 (post-NU staging): 256 VGPR with 7
   // spills + 32 B scratch AND occupancy halves (2 blocks/CU -> 1) — stays 1.
-  constexpr int NP = WANT_DK ? 1 : 2;
+  constexpr int NP = WANT_DK ? ((D == 64) ? 2 : 1) : 2;  // dK fits NP=2 at D=64
 
   // T14 staging of two 32-row q tiles: threads 0..255 own Q chunks
   // (2 rows x 8 d), threads 256..511 own dO chunks; each thread carries one
