@@ -347,6 +347,153 @@ __global__ __launch_bounds__(TPB) void muon_gemm_nn_ax_kernel(
       }
 }
 
+
+// ---------------------------------------------------------------------------
+// 256x256 quadrant-phased NT GEMM: C[M,N] = alpha * X[M,K] @ Y[N,K]^T.
+// A step toward the guide's 8-phase template sized for what is provably
+// race-free without the template's region-level buffer reuse (a 3-tile ring
+// does not fit 160 KiB at this tile size):
+//   - BK=64, 8 waves (2Mx4N, 512 thr), 128x64 per wave, 64 MFMAs
+//     (16x16x32 bf16) per wave per K-tile in FOUR quadrant phases of
+//     {12 ds_read_b128 | stage 1 half-tile of tile kt+1 | barrier |
+//      lgkmcnt | 16 MFMA | barrier};
+//   - 2 K-tile LDS buffers (128 KiB): tile kt+1 is staged into buffer
+//     (kt+1)&1, whose readers finished before tile kt's first barrier;
+//   - st_16x32 XOR swizzle (byte bit5 ^= bit9) on the glds SOURCE address
+//     and the ds_read address; LDS destination stays lane-linear
+//     (guide §5.4 rule 21);
+//   - the full counted-vmcnt 8-phase schedule (2 tiles in flight) is the
+//     known next rung; it needs the template's region-reuse proof.
+// Requires M,N % 256 == 0, K % 64 == 0 (the wrapper pads); SPLIT writes
+// fp32 slabs for muon_combine_kernel.
+template <bool SPLIT>
+__global__ __launch_bounds__(512) void muon_gemm_nt8_kernel(
+    const __hip_bfloat16* __restrict__ X, const __hip_bfloat16* __restrict__ Y,
+    __hip_bfloat16* __restrict__ C, int M, int N, int K,
+    float alpha, float* __restrict__ WS, int ksteps_per_slice) {
+  typedef __attribute__((ext_vector_type(4))) float f32x4;
+  extern __shared__ __attribute__((aligned(16))) __hip_bfloat16 smem8[];
+  constexpr int HT = 128 * 64;  // elements per half-tile image ([128][64])
+  const int tm = blockIdx.x, tn = blockIdx.y;
+  const int m0 = tm * 256, n0 = tn * 256;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int wr = (wave >> 2) * 128;  // 2 M-wave rows
+  const int wc = (wave & 3) * 64;    // 4 N-wave cols
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  int k0step = 0, ktiles = K / 64;
+  if constexpr (SPLIT) {
+    k0step = blockIdx.z * ksteps_per_slice;
+    ktiles = min(ktiles, k0step + ksteps_per_slice);
+    if (k0step >= ktiles) {
+      for (int r = tid; r < 256; r += 512) {
+        float* row = WS + ((long)blockIdx.z * M + m0 + r) * N + n0;
+        for (int c = 0; c < 256; ++c) row[c] = 0.f;
+      }
+      return;
+    }
+  }
+
+  // stage one [128][64] half-tile by glds: each wave moves 2 KiB (2 x 1 KiB)
+  auto stage_half = [&](const __hip_bfloat16* src_rows, int kt, __hip_bfloat16* dst) {
+#pragma unroll
+    for (int g = 0; g < 2; ++g) {
+      const int off16 = (wave * 2 + g) * 64 + lane;      // 16-B unit in image
+      const unsigned pb = (unsigned)off16 * 16u;          // physical byte
+      const unsigned lb = pb ^ (((pb >> 9) & 1u) << 5);   // logical byte
+      const int row = (int)(lb >> 7);
+      const int col = (int)(lb & 127) >> 1;               // element
+      const __hip_bfloat16* gp = src_rows + (long)row * K + kt * 64 + col;
+      __hip_bfloat16* lp = dst + (long)(wave * 2 + g) * 512;  // wave-uniform
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gp,
+          (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+    }
+  };
+  // half h of tile t: {0:A rows 0-127, 1:A 128-255, 2:Y 0-127, 3:Y 128-255}
+  auto stage_phase = [&](int h, int t) {
+    if (t >= ktiles) return;
+    const __hip_bfloat16* base = (h < 2) ? X + (long)(m0 + (h & 1) * 128) * K
+                                         : Y + (long)(n0 + (h & 1) * 128) * K;
+    stage_half(base, t, smem8 + ((t & 1) * 4 + h) * HT);
+  };
+  // fragment read (A-style for both operands: row-per-lane, 16 contiguous B)
+  auto frag = [&](const __hip_bfloat16* img, int r, int chunk) -> bf16x8 {
+    const unsigned lbyte = (unsigned)r * 128u + (unsigned)chunk * 16u;
+    const unsigned pbyte = lbyte ^ (((lbyte >> 9) & 1u) << 5);
+    Bf16x8U u;
+    *reinterpret_cast<uint4*>(u.s) = *reinterpret_cast<const uint4*>(
+        reinterpret_cast<const char*>(img) + pbyte);
+    return u.v;
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{};
+
+  // prologue: stage tile k0 fully; barrier includes the vmcnt(0) drain
+  for (int h = 0; h < 4; ++h) stage_phase(h, k0step);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int kt = k0step; kt < ktiles; ++kt) {
+    const __hip_bfloat16* Ah = smem8 + ((kt & 1) * 4 + (wr ? 1 : 0)) * HT;
+    const __hip_bfloat16* Bh = smem8 + ((kt & 1) * 4 + 2 + (wc >> 7)) * HT;
+    const int brow = wc & 127;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      // quadrant q: m-frags 2q, 2q+1 x 4 n-frags x K=64 (2 k-steps)
+      bf16x8 af[2][2], bfr[4][2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          af[mi][kk] = frag(Ah, (q * 2 + mi) * 16 + l15, kk * 4 + l4);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          bfr[ni][kk] = frag(Bh, brow + ni * 16 + l15, kk * 4 + l4);
+      stage_phase(q, kt + 1);  // buffer (kt+1)&1: readers done last tile
+      // no per-quadrant barrier needed: reads target the stable buffer,
+      // writes the other one; hipcc places counted lgkm waits for the
+      // ds_read->MFMA edges itself
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[q * 2 + mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mi][kk], bfr[ni][kk], acc[q * 2 + mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // tile kt+1's 8 glds (per wave) must land before its first read
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue: C/D 16x16 layout: col = lane&15, row = (lane>>4)*4 + reg
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wr + i * 16 + l4 * 4 + r;
+        const int n = n0 + wc + j * 16 + l15;
+        if constexpr (SPLIT) {
+          WS[((long)blockIdx.z * M + m) * N + n] = acc[i][j][r];
+        } else {
+          from_f32(&C[(long)m * N + n], alpha * acc[i][j][r]);
+        }
+      }
+}
+
 void check_2d_bf16(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kBFloat16 && t.dim() == 2 &&
                   t.is_contiguous(),
@@ -389,6 +536,50 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
     splitk = 2;
     while (nblocks * splitk * 2 <= 512 && splitk < 8 && (K / BK) / (splitk * 2) >= 2)
       splitk *= 2;
+  }
+  // 256^2 quadrant-phase kernel for aligned shapes (bigger tile halves the
+  // staging traffic per FLOP; env MCDP_MUON_NT8=0 disables)
+  static const bool use_nt8 = []() {
+    const char* e = getenv("MCDP_MUON_NT8");
+    return !e || atoi(e) != 0;
+  }();
+  const bool nt8_ok = use_nt8 && (M % 256 == 0) && (N % 256 == 0);
+  const size_t nt8_lds = 2 * 4 * 128 * 64 * sizeof(__hip_bfloat16);  // 128 KiB
+  static const bool nt8_attr = []() {  // >64 KiB dynamic LDS needs the opt-in
+    hipFuncSetAttribute(reinterpret_cast<const void*>(&muon_gemm_nt8_kernel<false>),
+                        hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    hipFuncSetAttribute(reinterpret_cast<const void*>(&muon_gemm_nt8_kernel<true>),
+                        hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    return true;
+  }();
+  (void)nt8_attr;
+  if (nt8_ok) {
+    const int nblk8 = (M / 256) * (N / 256);
+    int sk8 = 1;
+    if (nblk8 < 192 && (K / BK) >= 2) {
+      sk8 = 2;
+      while (nblk8 * sk8 * 2 <= 512 && sk8 < 16 && (K / BK) / (sk8 * 2) >= 2)
+        sk8 *= 2;
+    }
+    dim3 g8(M / 256, N / 256, sk8), b8(512);
+    if (sk8 > 1 || has_e) {
+      const int kps = (K / BK + sk8 - 1) / sk8;
+      auto ws = at::empty({(long)sk8 * M * N}, X.options().dtype(at::kFloat));
+      muon_gemm_nt8_kernel<true><<<g8, b8, nt8_lds, stream>>>(
+          xp, yp, nullptr, M, N, K, 1.f, ws.data_ptr<float>(), kps);
+      const long mn = (long)M * N;
+      const int cg = (int)std::min<long>((mn + 1023) / 1024, 2048);
+      if (has_e)
+        muon_combine_kernel<true, __hip_bfloat16><<<cg, 1024, 0, stream>>>(
+            ws.data_ptr<float>(), ep, cp, mn, sk8, (float)alpha, (float)beta);
+      else
+        muon_combine_kernel<false, __hip_bfloat16><<<cg, 1024, 0, stream>>>(
+            ws.data_ptr<float>(), ep, cp, mn, sk8, (float)alpha, (float)beta);
+    } else {
+      muon_gemm_nt8_kernel<false><<<g8, b8, nt8_lds, stream>>>(
+          xp, yp, cp, M, N, K, (float)alpha, nullptr, K / BK);
+    }
+    return;
   }
   if (splitk > 1) {
     const int ksteps = K / BK;
