@@ -539,9 +539,14 @@ void muon_gemm_nt(at::Tensor X, at::Tensor Y, at::Tensor C, double alpha,
   }
   // 256^2 quadrant-phase kernel for aligned shapes (bigger tile halves the
   // staging traffic per FLOP; env MCDP_MUON_NT8=0 disables)
+  // default OFF: correct but measured SLOWER than the 128^2 split-K kernel
+  // (0.56-0.97x vs 0.75-1.04x of hipBLASLt, r2_nt8.log) — the 256^2 tile
+  // only pays with the full counted-vmcnt region-reuse schedule (the
+  // guide's 8-phase template), which needs its own race proof; this kernel
+  // is the scaffold for that rung.
   static const bool use_nt8 = []() {
     const char* e = getenv("MCDP_MUON_NT8");
-    return !e || atoi(e) != 0;
+    return e && atoi(e) != 0;
   }();
   const bool nt8_ok = use_nt8 && (M % 256 == 0) && (N % 256 == 0);
   const size_t nt8_lds = 2 * 4 * 128 * 64 * sizeof(__hip_bfloat16);  // 128 KiB
