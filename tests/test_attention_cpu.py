@@ -140,3 +140,43 @@ def test_alibi_attention():
                     s[i, j] += slopes[h] * (j - i)
         rep_out[0, :, h] = torch.softmax(s, -1) @ vv
     assert torch.allclose(got, rep_out.to(got.dtype), atol=1e-5)
+
+
+def test_compiled_block_mask_tensors_match_bruteforce():
+    """CompiledBlockMask's granule codes / packed bits / kv ranges vs a
+    brute-force evaluation of the callable (CPU tensors; the GPU kernel
+    consuming them is covered in test_gpu_kernels)."""
+    import torch
+
+    from mlx_cuda_distributed_pretraining_amd.ops.attention import CompiledBlockMask
+
+    def mod(b, h, qi, ki):
+        return (ki <= qi) & (qi - ki < 40) & ((ki % 97) != 3)
+
+    B, H, S = 1, 2, 300
+    bm = CompiledBlockMask(mod, B, H, S, S, device="cpu")
+    qi = torch.arange(S).unsqueeze(1)
+    ki = torch.arange(S).unsqueeze(0)
+    keep = mod(0, 1, qi, ki)
+    # bits
+    for q in (0, 31, 64, 299):
+        got = [(int(bm.bits[0, 1, q, k // 8]) >> (k % 8)) & 1 for k in range(S)]
+        assert got == keep[q].int().tolist(), f"bits mismatch at q={q}"
+    # granule codes: 0 = none live, 2 = all live (within bounds)
+    nq, nkv = (S + 31) // 32, (S + 63) // 64
+    for qg in range(nq):
+        for kg in range(nkv):
+            blk = keep[qg * 32:(qg + 1) * 32, kg * 64:(kg + 1) * 64]
+            code = int(bm.gran[0, 1, qg, kg])
+            assert (code > 0) == bool(blk.any()), (qg, kg)
+            if code == 2:
+                assert bool(blk.all()), (qg, kg)
+    # kv ranges cover every live tile of each 256-row block
+    nqpb = (S + 255) // 256
+    for qb in range(nqpb):
+        live = keep[qb * 256:(qb + 1) * 256]
+        live_tiles = [kg for kg in range(nkv)
+                      if live[:, kg * 64:(kg + 1) * 64].any()]
+        lo, hi = int(bm.range[0, 1, qb, 0]), int(bm.range[0, 1, qb, 1])
+        if live_tiles:
+            assert lo <= live_tiles[0] and hi >= live_tiles[-1] + 1
