@@ -442,10 +442,13 @@ class CompiledBlockMask:
                 gq2[:Q_LEN, :KV_LEN] = keep
                 all_live = gq2.reshape(nq32, 32, nkv, self.KVB).all(dim=(1, 3))
                 self.gran[b, h] = any_live.to(torch.uint8) + (any_live & all_live).to(torch.uint8)
-                # per 256-row block: first/last live kv tile
-                blk = any_live.reshape(nqpb, -1, nkv).any(dim=1)  # [nqpb, nkv]
+                # per 256-row block: first/last live kv tile. Slice the
+                # 8 granule rows explicitly — a reshape(nqpb, -1) misassigns
+                # rows whenever ceil(Q/32) is not a multiple of 8 (caught by
+                # tests/test_attention_cpu.py::..._bruteforce at Q=300).
                 for qb in range(nqpb):
-                    live = blk[qb].nonzero().flatten()
+                    rows = any_live[qb * 8:(qb + 1) * 8]
+                    live = rows.any(dim=0).nonzero().flatten()
                     if live.numel():
                         self.range[b, h, qb, 0] = int(live.min())
                         self.range[b, h, qb, 1] = int(live.max()) + 1
