@@ -197,3 +197,36 @@ def test_remote_shard_http_streaming(tmp_path):
             assert all(torch.equal(a, b) for a, b in zip(batches, batches2))
         finally:
             srv.shutdown()
+
+
+def test_remote_shard_redownload_after_eviction(tmp_path):
+    """An evicted cached shard is transparently re-fetched on the next epoch
+    (the DiskSpaceManager budget governs remote caches too)."""
+    import http.server
+    import socketserver
+    import threading
+
+    from mlx_cuda_distributed_pretraining_amd.data.streaming import RemoteShard
+
+    d = _make_shards(tmp_path, n_shards=1, docs_per_shard=3)
+
+    class Handler(http.server.SimpleHTTPRequestHandler):
+        def __init__(self, *a, **kw):
+            super().__init__(*a, directory=str(d), **kw)
+
+        def log_message(self, *a):
+            pass
+
+    with socketserver.TCPServer(("127.0.0.1", 0), Handler) as srv:
+        port = srv.server_address[1]
+        threading.Thread(target=srv.serve_forever, daemon=True).start()
+        try:
+            cache = tmp_path / "cache2"
+            sh = RemoteShard(f"http://127.0.0.1:{port}/shard-0000.jsonl", cache)
+            p1 = sh.fetch()
+            assert p1.exists() and p1.stat().st_size > 0
+            p1.unlink()  # evicted
+            p2 = sh.fetch()
+            assert p2.exists() and p2.stat().st_size > 0
+        finally:
+            srv.shutdown()

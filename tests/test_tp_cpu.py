@@ -1243,3 +1243,54 @@ def test_reshard_sharded_forward_matches_full():
         assert not unexpected, unexpected
         for k, v in m.state_dict().items():
             assert v.shape == sd[k].shape, (k, v.shape, sd[k].shape)
+
+
+def test_reshard_checkpoint_end_to_end(tmp_path):
+    """File-level reshard: write tp=2 shard files + state json, reshard to
+    tp=1 (merge) and back to tp=2; shards match the originals."""
+    import json
+
+    from safetensors.torch import load_file, save_file
+
+    from mlx_cuda_distributed_pretraining_amd.core.config import Config
+    from mlx_cuda_distributed_pretraining_amd.models.llama import Model, ModelArgs
+    from tools.reshard_tp_checkpoint import reshard_checkpoint, shard_tp_state_dict
+
+    torch.manual_seed(6)
+    args = ModelArgs(hidden_size=32, intermediate_size=64, num_layers=1,
+                     num_heads=4, num_kv_heads=4, vocab_size=64,
+                     max_position_embeddings=64, tie_word_embeddings=False)
+    full = Model(args).state_dict()
+    run = tmp_path / "run"
+    (run / "checkpoints").mkdir(parents=True)
+    cfg = Config.from_dict({
+        "name": "reshard-e2e", "overwrite": True,
+        "data": {"synthetic": True, "synthetic_vocab_size": 64},
+        "model": {"dimensions": {"hidden_size": 32, "intermediate_size": 64,
+                                 "num_layers": 1},
+                  "misc": {"tie_word_embeddings": False},
+                  "attention": {"num_heads": 4, "num_kv_heads": 4,
+                                "max_position_embeddings": 64}},
+        "training": {"hyperparameters": {"iters": 1, "batch_size": 2}},
+        "logging": {"steps": {}}, "system": {"device": "cpu"}})
+    cfg.save_yaml(run / "config.yaml")
+    base = str(run / "checkpoints" / "step_9")
+    orig = []
+    for r in range(2):
+        sd = shard_tp_state_dict(full, r, 2, 4, 4, args.head_dim, 64)
+        save_file(sd, f"{base}_tp{r}_model.safetensors", metadata={"format": "pt"})
+        orig.append(sd)
+    (run / "checkpoints" / "step_9_state.json").write_text(
+        json.dumps({"step": 9, "tp_world": 2}))
+
+    out = reshard_checkpoint(base, tp=1)   # merge to full
+    assert out == [f"{base}_model.safetensors"]
+    merged = load_file(out[0])
+    for k in full:
+        assert torch.equal(merged[k], full[k]), k
+
+    out2 = reshard_checkpoint(base, tp=2)  # and back
+    for r in range(2):
+        sd = load_file(out2[r])
+        for k in sd:
+            assert torch.equal(sd[k], orig[r][k]), (r, k)
